@@ -1,0 +1,210 @@
+"""The CDNA4 kernel surface.
+
+Dispatch contract:
+- On a GPU box the in-tree HIP extension (``calfkit_amd/ops/_C*.so``) MUST be
+  present — a missing extension raises immediately; there is no silent eager
+  fallback on GPU (the native path is the one that runs).
+- On CPU the pure-torch reference implementations below serve tests and
+  development; every gpu-lane numerics test compares the HIP kernel against
+  the fp32 torch reference of the same op.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Any
+
+import torch
+
+__all__ = [
+    "have_extension",
+    "rmsnorm",
+    "fused_add_rmsnorm",
+    "silu_mul",
+    "rope_qk",
+    "kv_write",
+    "attn_decode",
+    "attn_prefill",
+    "sample",
+    "build_rope_table",
+]
+
+_K: Any = None
+_IMPORT_ERROR: Exception | None = None
+try:
+    import importlib
+
+    _K = importlib.import_module("._C", __package__)
+except ImportError as e:  # extension not built (CPU dev box is fine)
+    _IMPORT_ERROR = e
+
+
+def have_extension() -> bool:
+    return _K is not None
+
+
+def _native(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if _K is None:
+        raise RuntimeError(
+            "calfkit_amd.ops: tensor is on GPU but the gfx950 extension is not "
+            "built — run `python setup.py build_ext --inplace` "
+            f"(import error: {_IMPORT_ERROR})")
+    return True
+
+
+# ------------------------------------------------------------------ rmsnorm
+
+def rmsnorm(out: torch.Tensor, inp: torch.Tensor, weight: torch.Tensor,
+            eps: float = 1e-5) -> None:
+    if _native(inp):
+        _K.rmsnorm(out, inp, weight, eps)
+        return
+    x = inp.float()
+    n = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + eps)
+    out.copy_((n * weight.float()).to(out.dtype))
+
+
+def fused_add_rmsnorm(hidden: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float = 1e-5) -> None:
+    """residual += hidden; hidden = rmsnorm(residual). Both in place."""
+    if _native(hidden):
+        _K.fused_add_rmsnorm(hidden, residual, weight, eps)
+        return
+    s = (residual.float() + hidden.float())
+    residual.copy_(s.to(residual.dtype))
+    s = residual.float()  # norm over the bf16-rounded sum, like the kernel
+    n = s * torch.rsqrt(s.pow(2).mean(-1, keepdim=True) + eps)
+    hidden.copy_((n * weight.float()).to(hidden.dtype))
+
+
+# ----------------------------------------------------------------- silu_mul
+
+def silu_mul(out: torch.Tensor, gateup: torch.Tensor) -> None:
+    if _native(gateup):
+        _K.silu_mul(out, gateup)
+        return
+    f = out.shape[-1]
+    gate = gateup[..., :f].float()
+    up = gateup[..., f:].float()
+    out.copy_((torch.nn.functional.silu(gate) * up).to(out.dtype))
+
+
+# --------------------------------------------------------------------- rope
+
+def build_rope_table(max_pos: int, dh: int, theta: float = 500000.0,
+                     device: str | torch.device = "cpu") -> torch.Tensor:
+    """[max_pos, dh] f32 table laid out [cos(dh/2) | sin(dh/2)] per position."""
+    inv = 1.0 / (theta ** (torch.arange(0, dh, 2, dtype=torch.float64) / dh))
+    pos = torch.arange(max_pos, dtype=torch.float64)
+    ang = torch.outer(pos, inv)  # [max_pos, dh/2]
+    return torch.cat([ang.cos(), ang.sin()], dim=-1).float().to(device)
+
+
+def rope_qk(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
+            positions: torch.Tensor) -> None:
+    """In-place neox rotate-half RoPE. q: [T,Hq,dh], k: [T,Hkv,dh]."""
+    if _native(q):
+        _K.rope_qk(q, k, cos_sin, positions)
+        return
+    dh = q.shape[-1]
+    half = dh // 2
+    cs = cos_sin[positions.long()]  # [T, dh]
+    cos = cs[:, :half].unsqueeze(1)
+    sin = cs[:, half:].unsqueeze(1)
+    for t in (q, k):
+        x0 = t[..., :half].float()
+        x1 = t[..., half:].float()
+        t[..., :half] = (x0 * cos - x1 * sin).to(t.dtype)
+        t[..., half:] = (x1 * cos + x0 * sin).to(t.dtype)
+
+
+# ----------------------------------------------------------------- kv cache
+
+def kv_write(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
+             v: torch.Tensor, slots: torch.Tensor) -> None:
+    """Scatter [T,Hkv,dh] rows into [P,Hkv,PAGE,dh] caches at flat slots."""
+    if _native(k):
+        _K.kv_write(kcache, vcache, k, v, slots)
+        return
+    page = kcache.shape[2]
+    for t in range(k.shape[0]):
+        s = int(slots[t])
+        if s < 0:
+            continue
+        p, off = divmod(s, page)
+        kcache[p, :, off] = k[t]
+        vcache[p, :, off] = v[t]
+
+
+# ---------------------------------------------------------------- attention
+
+def attn_decode(out: torch.Tensor, q: torch.Tensor, kcache: torch.Tensor,
+                vcache: torch.Tensor, page_table: torch.Tensor,
+                seq_lens: torch.Tensor, scale: float) -> None:
+    """Paged GQA decode attention. q/out: [B,Hq,dh]."""
+    if _native(q):
+        _K.attn_decode(out, q, kcache, vcache, page_table, seq_lens, scale)
+        return
+    B, hq, dh = q.shape
+    hkv, page = kcache.shape[1], kcache.shape[2]
+    G = hq // hkv
+    for b in range(B):
+        n = int(seq_lens[b])
+        npages = (n + page - 1) // page
+        pages = page_table[b, :npages].long()
+        # gather [n, hkv, dh]
+        kk = kcache[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)[:n].float()
+        vv = vcache[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)[:n].float()
+        for h in range(hq):
+            kvh = h // G
+            s = (kk[:, kvh] @ (q[b, h].float() * scale)) # [n]
+            p = torch.softmax(s, dim=-1)
+            out[b, h] = (p @ vv[:, kvh]).to(out.dtype)
+
+
+def attn_prefill(out: torch.Tensor, q: torch.Tensor, k: torch.Tensor,
+                 v: torch.Tensor, scale: float, past_len: int = 0) -> None:
+    """Causal GQA prefill. q/out: [S,Hq,dh]; k/v: [past+S,Hkv,dh]."""
+    if _native(q):
+        _K.attn_prefill(out, q, k, v, scale, past_len)
+        return
+    S, hq, dh = q.shape
+    hkv = k.shape[1]
+    G = hq // hkv
+    T = k.shape[0]
+    mask = torch.arange(T, device=q.device)[None, :] > \
+        (past_len + torch.arange(S, device=q.device))[:, None]
+    for h in range(hq):
+        kvh = h // G
+        s = (q[:, h].float() @ k[:, kvh].float().T) * scale  # [S, T]
+        s = s.masked_fill(mask, float("-inf"))
+        p = torch.softmax(s, dim=-1)
+        out[:, h] = (p @ v[:, kvh].float()).to(out.dtype)
+
+
+# ----------------------------------------------------------------- sampling
+
+def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
+           seed_step: torch.Tensor, topk: torch.Tensor | None = None) -> None:
+    """Gumbel-max sampling; temperature 0 = greedy. out: int32 [B]."""
+    if _native(logits):
+        _K.sample(out, logits, temperature,
+                  seed_step, topk if topk is not None else torch.Tensor())
+        return
+    B, V = logits.shape
+    lg = logits.float().clone()
+    if topk is not None:
+        for b in range(B):
+            kk = int(topk[b])
+            if 0 < kk < V:
+                thr = torch.topk(lg[b], kk).values[-1]
+                lg[b][lg[b] < thr] = float("-inf")
+    for b in range(B):
+        t = float(temperature[b])
+        if t <= 0:
+            out[b] = int(lg[b].argmax())
+        else:
+            g = torch.distributions.Gumbel(0.0, 1.0).sample((V,))
+            out[b] = int((lg[b] / t + g).argmax())

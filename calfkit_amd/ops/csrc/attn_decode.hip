@@ -1,0 +1,210 @@
+// Paged-KV GQA decode attention for MI355X (gfx950).
+//
+// Shape: one query token per sequence (the decode step), grouped-query
+// attention against a paged KV cache laid out [num_pages, Hkv, KVPAGE, dh]
+// with KVPAGE == 64 so one (page, head) panel is a contiguous 16 KB run the
+// wave streams line-aligned.
+//
+// Decomposition: one workgroup per (sequence, kv_head); 4 waves stride the
+// sequence's pages.  Within a wave, LANE == TOKEN for the QK phase (each
+// lane owns one key row, so the G per-head scores are lane-local dots — no
+// cross-lane reduction), and LANE == DIM-PAIR for the PV phase (p broadcast
+// by shuffle, V row read coalesced 4 B/lane).  Online softmax per wave with
+// a final cross-wave m/l combine in LDS.
+//
+// The kernel is memory-bound by construction (~16 B of KV per lane-op);
+// the design targets the HBM3E stream, not MFMA (decode reads each KV byte
+// exactly once — matrix cores have nothing to multiply twice).
+
+#include "common.h"
+
+#define KVPAGE 64
+#define HD 128
+#define NWAVES 4
+
+template <int G>
+__global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
+    __hip_bfloat16* __restrict__ out,
+    const __hip_bfloat16* __restrict__ q,
+    const __hip_bfloat16* __restrict__ kcache,
+    const __hip_bfloat16* __restrict__ vcache,
+    const int32_t* __restrict__ page_table,
+    const int32_t* __restrict__ seq_lens,
+    const int hq, const int hkv, const int max_pages, const float scale) {
+  const int seq = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int slen = seq_lens[seq];
+  const int npages = (slen + KVPAGE - 1) / KVPAGE;
+
+  __shared__ float q_lds[G][HD];
+  __shared__ float o_lds[NWAVES][G][HD];
+  __shared__ float ml_lds[NWAVES][G][2];
+
+  for (int i = threadIdx.x; i < G * HD; i += NWAVES * WAVE) {
+    const int g = i / HD, d = i % HD;
+    q_lds[g][d] = bf2f(q[((int64_t)seq * hq + kvh * G + g) * HD + d]) * scale;
+  }
+  __syncthreads();
+
+  float m[G], l[G], acc[G][4];  // lane owns dims [4*((lane)%32)... see below
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    m[g] = -INFINITY;
+    l[g] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[g][j] = 0.f;
+  }
+  // dim mapping for PV: HD=128 dims; 64 lanes x 4 dims each covers 256 — so
+  // only lanes 0..31 own dims (4*lane..4*lane+3); lanes 32..63 mirror them
+  // for the SECOND half of the tokens to double PV throughput.
+  const int dlane = lane & 31;        // dim group
+  const int thalf = lane >> 5;        // 0: tokens 0..31, 1: tokens 32..63
+
+  const int64_t tbl = (int64_t)seq * max_pages;
+  for (int pi = wid; pi < npages; pi += NWAVES) {
+    const int page = page_table[tbl + pi];
+    const int64_t base = (((int64_t)page * hkv + kvh) * KVPAGE) * HD;
+    const int token = pi * KVPAGE + lane;
+    const bool live = token < slen;
+
+    float s[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) s[g] = 0.f;
+    if (live) {
+      const bf16x8* krow = reinterpret_cast<const bf16x8*>(kcache + base + lane * HD);
+#pragma unroll
+      for (int i = 0; i < HD / 8; ++i) {
+        bf16x8 kv8 = krow[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float kf = bf2f(kv8.v[j]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) s[g] = fmaf(q_lds[g][i * 8 + j], kf, s[g]);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int g = 0; g < G; ++g) s[g] = -INFINITY;
+    }
+
+    float p[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const float mt = wave_max(s[g]);
+      const float mn = fmaxf(m[g], mt);
+      const float r = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
+      p[g] = (s[g] == -INFINITY) ? 0.f : __expf(s[g] - mn);
+      l[g] = l[g] * r + wave_sum(p[g]);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[g][j] *= r;
+      m[g] = mn;
+    }
+
+    // PV: lanes 0..31 take tokens 0..31, lanes 32..63 take 32..63; each
+    // half-wave covers all 128 dims (32 lanes x 4 dims).  V row read is
+    // 8 B/lane coalesced within the half's 32 lanes.
+    const int t0 = thalf * 32;
+    const int nt = min(32, slen - pi * KVPAGE - t0);
+    const float2* vbase = reinterpret_cast<const float2*>(vcache + base);
+    for (int t = 0; t < nt; ++t) {
+      const float2 raw = vbase[(t0 + t) * (HD / 4) + dlane];
+      const __hip_bfloat16* vb = reinterpret_cast<const __hip_bfloat16*>(&raw);
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        const float pg = __shfl(p[g], t0 + t, WAVE);
+        acc[g][0] = fmaf(pg, bf2f(vb[0]), acc[g][0]);
+        acc[g][1] = fmaf(pg, bf2f(vb[1]), acc[g][1]);
+        acc[g][2] = fmaf(pg, bf2f(vb[2]), acc[g][2]);
+        acc[g][3] = fmaf(pg, bf2f(vb[3]), acc[g][3]);
+      }
+    }
+  }
+
+  // ---- combine the two half-wave partials: dims are duplicated across
+  // halves, tokens split — a cross-half sum via shfl(lane^32) finishes each
+  // wave's O.  m/l are wave-uniform already (wave_max/wave_sum).
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[g][j] += __shfl_xor(acc[g][j], 32, WAVE);
+  }
+
+  // ---- cross-wave combine in LDS
+  if (lane < 32) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o_lds[wid][g][dlane * 4 + j] = acc[g][j];
+      if (dlane == 0) {
+        ml_lds[wid][g][0] = m[g];
+        ml_lds[wid][g][1] = l[g];
+      }
+    }
+  }
+  __syncthreads();
+
+  // wave 0 reduces and writes the output rows
+  if (wid == 0 && lane < 32) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      float M = -INFINITY;
+#pragma unroll
+      for (int w = 0; w < NWAVES; ++w) M = fmaxf(M, ml_lds[w][g][0]);
+      float L = 0.f, o[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int w = 0; w < NWAVES; ++w) {
+        const float mw = ml_lds[w][g][0];
+        const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+        L += sc * ml_lds[w][g][1];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) o[j] = fmaf(sc, o_lds[w][g][dlane * 4 + j], o[j]);
+      }
+      const float inv = (L > 0.f) ? 1.f / L : 0.f;
+      __hip_bfloat16* orow = out + ((int64_t)seq * hq + kvh * G + g) * HD;
+      bf16x4 pack;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) pack.v[j] = f2bf(o[j] * inv);
+      reinterpret_cast<bf16x4*>(orow)[dlane] = pack;
+    }
+  }
+}
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
+                 torch::Tensor vcache, torch::Tensor page_table,
+                 torch::Tensor seq_lens, double scale) {
+  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous());
+  TORCH_CHECK(page_table.scalar_type() == torch::kInt32);
+  TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
+  const int B = q.size(0), hq = q.size(1), dh = q.size(2);
+  const int hkv = kcache.size(1), page = kcache.size(2);
+  const int max_pages = page_table.size(1);
+  TORCH_CHECK(dh == HD, "decode kernel assumes head_dim == 128");
+  TORCH_CHECK(page == KVPAGE, "decode kernel assumes page_size == 64");
+  const int G = hq / hkv;
+  dim3 grid(B, hkv), block(NWAVES * WAVE);
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+#define LAUNCH(GV)                                                        \
+  hipLaunchKernelGGL(attn_decode4_kernel<GV>, grid, block, 0, s,          \
+                     (__hip_bfloat16*)out.data_ptr(),                     \
+                     (const __hip_bfloat16*)q.data_ptr(),                 \
+                     (const __hip_bfloat16*)kcache.data_ptr(),            \
+                     (const __hip_bfloat16*)vcache.data_ptr(),            \
+                     (const int32_t*)page_table.data_ptr(),               \
+                     (const int32_t*)seq_lens.data_ptr(), hq, hkv,        \
+                     max_pages, (float)scale)
+  switch (G) {
+    case 1: LAUNCH(1); break;
+    case 2: LAUNCH(2); break;
+    case 4: LAUNCH(4); break;
+    case 8: LAUNCH(8); break;
+    default: TORCH_CHECK(false, "unsupported GQA group size ", G);
+  }
+#undef LAUNCH
+  HIP_CHECK_LAST();
+}

@@ -1,0 +1,31 @@
+// Python bindings for the calfkit_amd CDNA4 kernel set (gfx950 only).
+
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor inp, torch::Tensor weight, double eps);
+void fused_add_rmsnorm(torch::Tensor hidden, torch::Tensor residual,
+                       torch::Tensor weight, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor gateup);
+void rope_qk(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
+             torch::Tensor positions);
+void kv_write(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
+              torch::Tensor v, torch::Tensor slots);
+void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
+                 torch::Tensor vcache, torch::Tensor page_table,
+                 torch::Tensor seq_lens, double scale);
+void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                  torch::Tensor v, double scale, int64_t past_len);
+void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
+            torch::Tensor seed_step, torch::Tensor topk);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "residual += hidden; hidden = rmsnorm(residual)");
+  m.def("silu_mul", &silu_mul, "silu(gate) * up from packed [T,2F]");
+  m.def("rope_qk", &rope_qk, "in-place neox RoPE on Q and K");
+  m.def("kv_write", &kv_write, "scatter K/V rows into the paged cache");
+  m.def("attn_decode", &attn_decode, "paged GQA decode attention");
+  m.def("attn_prefill", &attn_prefill, "causal GQA prefill attention (MFMA)");
+  m.def("sample", &sample, "Gumbel-max token sampling");
+}

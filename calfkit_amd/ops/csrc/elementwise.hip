@@ -1,0 +1,271 @@
+// Fused elementwise / normalization kernels for the Llama-family decode and
+// prefill paths on MI355X (gfx950).  All memory-bound: bf16 I/O is vectorized
+// as 16 B/lane (bf16x8) per Guideline 13; reductions are wave-wide shuffles.
+//
+// Kernels:
+//   rmsnorm           out = x * rsqrt(mean(x^2)+eps) * w
+//   fused_add_rmsnorm residual += hidden; hidden = rmsnorm(residual)
+//   silu_mul          out = silu(gate) * up       (gate/up packed [T, 2F])
+//   rope_qk           in-place neox rotate-half rotary on Q and K rows
+//   kv_write          scatter new K/V rows into the paged cache
+//
+// Capability parity note: the reference (calf-ai/calfkit-sdk) has no kernels —
+// these implement the LocalModelClient compute of BASELINE.json's north star.
+
+#include "common.h"
+
+// ------------------------------------------------------------------ rmsnorm
+
+template <int BLOCK>
+__global__ void rmsnorm_kernel(__hip_bfloat16* __restrict__ out,
+                               const __hip_bfloat16* __restrict__ inp,
+                               const __hip_bfloat16* __restrict__ weight,
+                               const int d, const float eps) {
+  __shared__ float red[BLOCK / WAVE];
+  const int64_t row = blockIdx.x;
+  const __hip_bfloat16* x = inp + row * d;
+  __hip_bfloat16* o = out + row * d;
+
+  float ss = 0.f;
+  const int nvec = d / 8;
+  const bf16x8* xv = reinterpret_cast<const bf16x8*>(x);
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    bf16x8 v = xv[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(v.v[j]);
+      ss += f * f;
+    }
+  }
+  ss = block_sum<BLOCK>(ss, red);
+  const float inv = rsqrtf(ss / d + eps);
+
+  const bf16x8* wv = reinterpret_cast<const bf16x8*>(weight);
+  bf16x8* ov = reinterpret_cast<bf16x8*>(o);
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    bf16x8 v = xv[i], w = wv[i], r;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) r.v[j] = f2bf(bf2f(v.v[j]) * inv * bf2f(w.v[j]));
+    ov[i] = r;
+  }
+}
+
+// residual += hidden; hidden = rmsnorm(residual) — one pass over HBM instead
+// of three (the producing add is fused into the consuming norm).
+template <int BLOCK>
+__global__ void fused_add_rmsnorm_kernel(__hip_bfloat16* __restrict__ hidden,
+                                         __hip_bfloat16* __restrict__ residual,
+                                         const __hip_bfloat16* __restrict__ weight,
+                                         const int d, const float eps) {
+  __shared__ float red[BLOCK / WAVE];
+  const int64_t row = blockIdx.x;
+  __hip_bfloat16* h = hidden + row * d;
+  __hip_bfloat16* r = residual + row * d;
+
+  float ss = 0.f;
+  const int nvec = d / 8;
+  bf16x8* hv = reinterpret_cast<bf16x8*>(h);
+  bf16x8* rv = reinterpret_cast<bf16x8*>(r);
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    bf16x8 a = hv[i], b = rv[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(a.v[j]) + bf2f(b.v[j]);
+      b.v[j] = f2bf(f);
+      ss += f * f;
+    }
+    rv[i] = b;  // residual holds the sum
+  }
+  ss = block_sum<BLOCK>(ss, red);
+  const float inv = rsqrtf(ss / d + eps);
+
+  const bf16x8* wv = reinterpret_cast<const bf16x8*>(weight);
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    bf16x8 b = rv[i], w = wv[i], o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o.v[j] = f2bf(bf2f(b.v[j]) * inv * bf2f(w.v[j]));
+    hv[i] = o;
+  }
+}
+
+// ----------------------------------------------------------------- silu_mul
+
+__global__ void silu_mul_kernel(__hip_bfloat16* __restrict__ out,
+                                const __hip_bfloat16* __restrict__ gateup,
+                                const int64_t rows, const int f) {
+  // gateup row layout: [gate (f) | up (f)]
+  const int64_t row = blockIdx.y;
+  const __hip_bfloat16* g = gateup + row * (2 * (int64_t)f);
+  const __hip_bfloat16* u = g + f;
+  __hip_bfloat16* o = out + row * f;
+  const int nvec = f / 8;
+  const bf16x8* gv = reinterpret_cast<const bf16x8*>(g);
+  const bf16x8* uv = reinterpret_cast<const bf16x8*>(u);
+  bf16x8* ov = reinterpret_cast<bf16x8*>(o);
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += gridDim.x * blockDim.x) {
+    bf16x8 a = gv[i], b = uv[i], r;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float x = bf2f(a.v[j]);
+      float s = x / (1.f + __expf(-x));
+      r.v[j] = f2bf(s * bf2f(b.v[j]));
+    }
+    ov[i] = r;
+  }
+}
+
+// ------------------------------------------------------------------- rope
+
+// Neox rotate-half RoPE, in place, on packed [T, H, dh] Q and K.
+// cos_sin: [max_pos, dh] laid out as [cos(dh/2) | sin(dh/2)] per position
+// (host-precomputed f32 table — on-device trig would turn this memory-bound
+// kernel VALU-bound, Appendix B).
+// One wave per (token, head): lane l rotates the pair (l, l+dh/2), dh=2*WAVE.
+__global__ void rope_qk_kernel(__hip_bfloat16* __restrict__ q,
+                               __hip_bfloat16* __restrict__ k,
+                               const float* __restrict__ cos_sin,
+                               const int32_t* __restrict__ positions,
+                               const int hq, const int hk, const int dh) {
+  const int64_t token = blockIdx.x;
+  const int wid = threadIdx.x / WAVE;      // head slot within the block
+  const int lane = threadIdx.x % WAVE;
+  const int nwave = blockDim.x / WAVE;
+  const int half = dh / 2;
+  const int pos = positions[token];
+  const float* cs = cos_sin + (int64_t)pos * dh;
+
+  for (int h = wid; h < hq + hk; h += nwave) {
+    __hip_bfloat16* base = (h < hq) ? q + (token * hq + h) * dh
+                                    : k + (token * hk + (h - hq)) * dh;
+    for (int i = lane; i < half; i += WAVE) {
+      const float c = cs[i], s = cs[half + i];
+      const float x0 = bf2f(base[i]);
+      const float x1 = bf2f(base[half + i]);
+      base[i] = f2bf(x0 * c - x1 * s);
+      base[half + i] = f2bf(x1 * c + x0 * s);
+    }
+  }
+}
+
+// ---------------------------------------------------------------- kv_write
+
+// Paged KV cache layout: [num_pages, Hkv, PAGE, dh] so one (page, head) is a
+// contiguous PAGE*dh*2B run — the decode kernel streams it line-aligned.
+__global__ void kv_write_kernel(__hip_bfloat16* __restrict__ kcache,
+                                __hip_bfloat16* __restrict__ vcache,
+                                const __hip_bfloat16* __restrict__ k,
+                                const __hip_bfloat16* __restrict__ v,
+                                const int32_t* __restrict__ slots,
+                                const int hkv, const int dh, const int page) {
+  const int64_t token = blockIdx.x;
+  const int slot = slots[token];
+  if (slot < 0) return;  // padding
+  const int p = slot / page, off = slot % page;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int nwave = blockDim.x / WAVE;
+  const int nvec = dh / 8;  // bf16x8 chunks per row
+
+  for (int h = wid; h < hkv; h += nwave) {
+    const bf16x8* ks = reinterpret_cast<const bf16x8*>(k + (token * hkv + h) * dh);
+    const bf16x8* vs = reinterpret_cast<const bf16x8*>(v + (token * hkv + h) * dh);
+    int64_t dst = (((int64_t)p * hkv + h) * page + off) * dh;
+    bf16x8* kd = reinterpret_cast<bf16x8*>(kcache + dst);
+    bf16x8* vd = reinterpret_cast<bf16x8*>(vcache + dst);
+    for (int i = lane; i < nvec; i += WAVE) {
+      kd[i] = ks[i];
+      vd[i] = vs[i];
+    }
+  }
+}
+
+// ------------------------------------------------------------- host glue
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor inp, torch::Tensor weight,
+             double eps) {
+  TORCH_CHECK(inp.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(inp.scalar_type() == torch::kBFloat16);
+  const int d = inp.size(-1);
+  TORCH_CHECK(d % 8 == 0, "hidden dim must be a multiple of 8");
+  const int64_t rows = inp.numel() / d;
+  constexpr int BLOCK = 256;
+  hipLaunchKernelGGL(rmsnorm_kernel<BLOCK>, dim3(rows), dim3(BLOCK), 0,
+                     cur_stream(),
+                     (__hip_bfloat16*)out.data_ptr(),
+                     (const __hip_bfloat16*)inp.data_ptr(),
+                     (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps);
+  HIP_CHECK_LAST();
+}
+
+void fused_add_rmsnorm(torch::Tensor hidden, torch::Tensor residual,
+                       torch::Tensor weight, double eps) {
+  TORCH_CHECK(hidden.is_contiguous() && residual.is_contiguous());
+  const int d = hidden.size(-1);
+  TORCH_CHECK(d % 8 == 0);
+  const int64_t rows = hidden.numel() / d;
+  constexpr int BLOCK = 256;
+  hipLaunchKernelGGL(fused_add_rmsnorm_kernel<BLOCK>, dim3(rows), dim3(BLOCK), 0,
+                     cur_stream(),
+                     (__hip_bfloat16*)hidden.data_ptr(),
+                     (__hip_bfloat16*)residual.data_ptr(),
+                     (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps);
+  HIP_CHECK_LAST();
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor gateup) {
+  TORCH_CHECK(out.is_contiguous() && gateup.is_contiguous());
+  const int f = out.size(-1);
+  TORCH_CHECK(gateup.size(-1) == 2 * f && f % 8 == 0);
+  const int64_t rows = out.numel() / f;
+  const int nvec = f / 8;
+  const int block = 256;
+  const int gx = std::min<int64_t>((nvec + block - 1) / block, 64);
+  hipLaunchKernelGGL(silu_mul_kernel, dim3(gx, rows), dim3(block), 0,
+                     cur_stream(),
+                     (__hip_bfloat16*)out.data_ptr(),
+                     (const __hip_bfloat16*)gateup.data_ptr(), rows, f);
+  HIP_CHECK_LAST();
+}
+
+void rope_qk(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
+             torch::Tensor positions) {
+  // q: [T, Hq, dh], k: [T, Hkv, dh], cos_sin: [max_pos, dh] f32, positions i32
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && cos_sin.is_contiguous());
+  TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(positions.scalar_type() == torch::kInt32);
+  const int T = q.size(0), hq = q.size(1), dh = q.size(2);
+  const int hk = k.size(1);
+  TORCH_CHECK(dh == 2 * WAVE, "rope kernel assumes head_dim == 128");
+  const int block = 256;
+  hipLaunchKernelGGL(rope_qk_kernel, dim3(T), dim3(block), 0, cur_stream(),
+                     (__hip_bfloat16*)q.data_ptr(),
+                     (__hip_bfloat16*)k.data_ptr(),
+                     (const float*)cos_sin.data_ptr(),
+                     (const int32_t*)positions.data_ptr(), hq, hk, dh);
+  HIP_CHECK_LAST();
+}
+
+void kv_write(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
+              torch::Tensor v, torch::Tensor slots) {
+  // kcache/vcache: [num_pages, Hkv, PAGE, dh]; k/v: [T, Hkv, dh]; slots i32 [T]
+  TORCH_CHECK(kcache.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32);
+  const int T = k.size(0), hkv = k.size(1), dh = k.size(2);
+  const int page = kcache.size(2);
+  TORCH_CHECK(kcache.size(1) == hkv && kcache.size(3) == dh);
+  hipLaunchKernelGGL(kv_write_kernel, dim3(T), dim3(256), 0, cur_stream(),
+                     (__hip_bfloat16*)kcache.data_ptr(),
+                     (__hip_bfloat16*)vcache.data_ptr(),
+                     (const __hip_bfloat16*)k.data_ptr(),
+                     (const __hip_bfloat16*)v.data_ptr(),
+                     (const int32_t*)slots.data_ptr(), hkv, dh, page);
+  HIP_CHECK_LAST();
+}
