@@ -111,9 +111,11 @@ __global__ void topk_mask_kernel(float* __restrict__ logits, const int V,
   mn = -block_max<BLOCK>(-mn, red);
   __syncthreads();
 
-  // bisection for the threshold t with |{x >= t}| ~= k
+  // bisection for the threshold: invariant count(>= hi) <= k < count(>= lo).
+  // Masking at hi keeps AT MOST k survivors (never more); the row max always
+  // survives (mx >= hi by the invariant), so sampling stays well-defined.
   float lo = mn, hi = mx;
-  for (int it = 0; it < 12; ++it) {
+  for (int it = 0; it < 24; ++it) {
     const float mid = 0.5f * (lo + hi);
     int cnt = 0;
     for (int i = threadIdx.x; i < V; i += BLOCK) cnt += (row[i] >= mid);
@@ -125,7 +127,7 @@ __global__ void topk_mask_kernel(float* __restrict__ logits, const int V,
       hi = mid;
   }
   for (int i = threadIdx.x; i < V; i += BLOCK)
-    if (row[i] < lo) row[i] = -INFINITY;
+    if (row[i] < hi) row[i] = -INFINITY;
 }
 
 #include <torch/extension.h>

@@ -1,0 +1,29 @@
+"""Distributed/parallel plumbing: one process per GPU, RCCL over xGMI.
+
+``torch.distributed`` with backend "nccl" IS RCCL on ROCm; CPU test lanes use
+"gloo".  TP collectives here are shaped for the MI355X node topology: 7
+point-to-point xGMI links × ~153 GB/s per GPU — decode-step
+activations are latency-bound (small), prefill activations bandwidth-bound,
+so the engine keeps decode all-reduces unbucketed on the compute stream and
+prefill all-reduces chunked.
+"""
+
+from .pg import (
+    DistInfo,
+    all_reduce_sum,
+    barrier,
+    get_dist,
+    init_distributed,
+    is_initialized,
+    shard_dim,
+)
+
+__all__ = [
+    "DistInfo",
+    "all_reduce_sum",
+    "barrier",
+    "get_dist",
+    "init_distributed",
+    "is_initialized",
+    "shard_dim",
+]
