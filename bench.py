@@ -1,0 +1,177 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: agent turns/sec (whole node), Llama-3-8B provider.
+
+Contract (driver): ``python bench.py --gpus N --steps K --warmup W``.
+For N>1 the driver launches this under torch.distributed.run with one rank
+per GPU; each rank runs an independent Worker replica (agent mesh + its own
+Llama-3-8B TP=1 engine on its GPU) — calfkit's one-Worker-per-GPU scaling
+story (weak scaling).  A step = TURNS_PER_STEP concurrent agent turns per
+rank through the full mesh pipeline (client -> agent node -> engine ->
+reply); each turn prefills a ~PROMPT_TOKENS synthetic prompt and decodes
+DECODE_TOKENS tokens on random-init weights.
+
+Rank 0 prints exactly ONE JSON line: whole-job agent turns/sec plus p50
+end-to-end turn latency (p50_turn_ms).
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import sys
+import time
+
+import torch
+
+PROMPT_TOKENS = 128
+DECODE_TOKENS = 64
+TURNS_PER_STEP = 32
+
+
+def log(msg: str) -> None:
+    print(msg, file=sys.stderr, flush=True)
+
+
+async def run_rank(args, dist) -> dict:
+    from calfkit_amd import Agent, Client, Worker
+    from calfkit_amd.adapters.memory import (
+        InMemoryBus,
+        InMemoryDirectory,
+        InProcessThreadStore,
+    )
+    from calfkit_amd.engine.local import LocalModelClient
+    from calfkit_amd.models.config import get_config
+    from calfkit_amd.parallel import barrier
+    from calfkit_amd.schemas import ModelSettings
+
+    on_gpu = torch.cuda.is_available()
+    model_name = args.model
+    overrides = {}
+    if args.layers:
+        overrides["layers"] = args.layers
+    if not on_gpu and model_name != "tiny":
+        log("no GPU visible - falling back to tiny config (sanity mode; "
+            "driver runs this on MI355X)")
+        model_name = "tiny"
+    cfg = get_config(model_name, **overrides) if overrides else get_config(model_name)
+
+    provider = LocalModelClient(
+        cfg, max_batch=max(TURNS_PER_STEP, 32),
+        max_model_len=1024, use_graphs=on_gpu)
+    agent = Agent(
+        name="bench_agent",
+        system_prompt="You are a fast benchmark agent.",
+        model_client=provider,
+        model_settings=ModelSettings(max_tokens=DECODE_TOKENS, temperature=0.0),
+    )
+    bus, store, directory = InMemoryBus(), InProcessThreadStore(), InMemoryDirectory()
+    prompt_words = " ".join(f"w{i}" for i in range(PROMPT_TOKENS - 8))
+
+    latencies: list[float] = []
+
+    async def one_turn(client, i: int, record: bool) -> None:
+        t0 = time.perf_counter()
+        r = await client.execute("bench_agent", f"turn {i}: {prompt_words}",
+                                 timeout=600)
+        assert r.text.strip(), "empty agent reply"
+        if record:
+            latencies.append((time.perf_counter() - t0) * 1e3)
+
+    async def one_step(client, step: int, record: bool) -> None:
+        await asyncio.gather(*(one_turn(client, step * TURNS_PER_STEP + i, record)
+                               for i in range(TURNS_PER_STEP)))
+
+    async with Worker(agent, bus=bus, store=store, directory=directory):
+        client = await Client.connect(bus, directory)
+        # engine spin-up (weights + graph capture) happens on first turn
+        for wstep in range(args.warmup):
+            await one_step(client, -1 - wstep, record=False)
+
+        barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for step in range(args.steps):
+            await one_step(client, step, record=True)
+        if on_gpu:
+            torch.cuda.synchronize()
+        barrier()
+        elapsed = time.perf_counter() - t0
+
+    return {"elapsed": elapsed, "latencies": latencies,
+            "engine_stats": provider.engine.stats, "model": cfg.name}
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=4)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--layers", type=int, default=0,
+                    help="override layer count (diagnostics only; the judged "
+                         "number uses the full model)")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    n_gpus = world if world > 1 else args.gpus
+
+    from calfkit_amd.parallel import init_distributed
+
+    dist = init_distributed()
+    out = asyncio.run(run_rank(args, dist))
+
+    elapsed = out["elapsed"]
+    # max elapsed over ranks = whole-job wall time
+    if world > 1:
+        import torch.distributed as td
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=dist.device if dist.backend == "nccl" else "cpu")
+        td.all_reduce(t, op=td.ReduceOp.MAX)
+        elapsed = float(t.item())
+        lat = torch.tensor(
+            [statistics.median(out["latencies"])], dtype=torch.float64,
+            device=t.device)
+        td.all_reduce(lat, op=td.ReduceOp.SUM)
+        p50 = float(lat.item()) / world
+    else:
+        p50 = statistics.median(out["latencies"])
+
+    turns_total = args.steps * TURNS_PER_STEP * n_gpus
+    value = turns_total / elapsed
+    if rank == 0:
+        result = {
+            "metric": "agent turns/sec (whole node), Llama-3-8B provider",
+            "value": round(value, 3),
+            "unit": "turns/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1e3 / args.steps, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "p50_turn_ms": round(p50, 1),
+            "tokens_per_s": round(turns_total * (PROMPT_TOKENS + DECODE_TOKENS)
+                                  / elapsed, 1),
+            "config": {
+                "model": out["model"],
+                "global_batch": TURNS_PER_STEP * n_gpus,
+                "seq_len": PROMPT_TOKENS + DECODE_TOKENS,
+                "prompt_tokens": PROMPT_TOKENS,
+                "decode_tokens": DECODE_TOKENS,
+                "parallelism": f"dp{n_gpus} (1 worker replica/GPU, TP=1)",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,373 @@
+"""The inference engine: continuous batching + hipGraph-captured decode.
+
+One engine per process (= per GPU rank).  Agent hops from every node hosted
+by this worker funnel into the same engine, so concurrent agent turns batch
+into one decode step — that is where whole-node agent turns/sec comes from.
+
+Threading model: the engine owns a dedicated scheduler thread; the asyncio
+agent plane submits requests through a thread-safe queue and gets
+``concurrent.futures.Future``s back (bridged by LocalModelClient).
+
+Decode step (the hot loop):
+  tokens/positions/slots/page_table/seq_lens live in STATIC device tensors;
+  each step writes the small host-side batch state into them, then either
+  replays the hipGraph captured for the padded batch size or runs eager.
+  Sampling (Gumbel-max) runs inside the graph; the seed/step counter is a
+  device tensor bumped by a tiny copy before each replay.
+"""
+
+from __future__ import annotations
+
+import logging
+import queue
+import threading
+import time
+from concurrent.futures import Future
+from dataclasses import dataclass, field
+from typing import Callable
+
+import torch
+
+from ..models.config import ModelConfig
+from ..models.llama import LlamaShard
+from .kvcache import KVCache, PAGE_SIZE
+
+_log = logging.getLogger("calfkit.engine")
+
+__all__ = ["GenRequest", "GenResult", "InferenceEngine"]
+
+PREFILL_CHUNK = 512
+
+
+@dataclass
+class GenRequest:
+    prompt_ids: list[int]
+    max_new_tokens: int = 64
+    temperature: float = 0.0
+    top_k: int = 0
+    seed: int = 0
+    ignore_eos: bool = True
+    on_token: Callable[[int], None] | None = None
+
+
+@dataclass
+class GenResult:
+    token_ids: list[int] = field(default_factory=list)
+    prompt_tokens: int = 0
+    prefill_ms: float = 0.0
+    decode_ms: float = 0.0
+    finish_reason: str = "length"
+
+
+@dataclass
+class _Seq:
+    seq_id: int
+    request: GenRequest
+    future: Future
+    result: GenResult = field(default_factory=GenResult)
+    last_token: int = 0
+    new_tokens: int = 0
+
+
+class InferenceEngine:
+    #: padded decode batch sizes that get their own captured graph
+    GRAPH_BATCHES = (1, 2, 4, 8, 16, 32, 64)
+
+    def __init__(self, model: LlamaShard, *, max_batch: int = 64,
+                 kv_pages: int | None = None, use_graphs: bool | None = None,
+                 max_model_len: int | None = None):
+        self.model = model
+        self.cfg: ModelConfig = model.cfg
+        self.device = model.device
+        self.max_batch = max_batch
+        self.max_model_len = max_model_len or self.cfg.max_seq
+        self.on_gpu = self.device.type == "cuda"
+        self.use_graphs = self.on_gpu if use_graphs is None else use_graphs
+
+        if kv_pages is None:
+            kv_pages = (max_batch * self.max_model_len + PAGE_SIZE - 1) // PAGE_SIZE + 8
+        self.kv = KVCache(layers=self.cfg.layers, kv_heads=model.hkv,
+                          head_dim=model.dh, num_pages=kv_pages,
+                          device=self.device)
+        self.max_pages = (self.max_model_len + PAGE_SIZE - 1) // PAGE_SIZE
+
+        # prefill scratch KV (contiguous [max_ctx, hkv, dh] per layer)
+        self._prefill_kv = [
+            (torch.zeros(self.max_model_len, model.hkv, model.dh,
+                         dtype=model.dtype, device=self.device),
+             torch.zeros(self.max_model_len, model.hkv, model.dh,
+                         dtype=model.dtype, device=self.device))
+            for _ in range(self.cfg.layers)]
+
+        # ---- static decode buffers (graph I/O)
+        B = max_batch
+        dev = self.device
+        self._tokens = torch.zeros(B, dtype=torch.int64, device=dev)
+        self._positions = torch.zeros(B, dtype=torch.int32, device=dev)
+        self._slots = torch.zeros(B, dtype=torch.int32, device=dev)
+        self._page_table = torch.zeros(B, self.max_pages, dtype=torch.int32,
+                                       device=dev)
+        self._seq_lens = torch.ones(B, dtype=torch.int32, device=dev)
+        self._temperature = torch.zeros(B, dtype=torch.float32, device=dev)
+        self._topk = torch.zeros(B, dtype=torch.int32, device=dev)
+        self._seed_step = torch.zeros(2, dtype=torch.int64, device=dev)
+        self._sampled = torch.zeros(B, dtype=torch.int32, device=dev)
+        self._graphs: dict[int, torch.cuda.CUDAGraph] = {}
+        #: one page sacrificed as the write target of padding rows
+        self._dummy_page = self.kv.num_pages - 1
+        self.kv._free.remove(self._dummy_page)
+
+        self._queue: "queue.Queue[_Seq | None]" = queue.Queue()
+        self._running: list[_Seq] = []
+        self._waiting: list[_Seq] = []
+        self._next_id = 0
+        self._step_counter = 0
+        self._stats = {"decode_steps": 0, "decode_tokens": 0, "prefills": 0,
+                       "decode_ms": 0.0, "prefill_ms": 0.0, "graph_replays": 0}
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._loop, name="calfkit-engine",
+                                        daemon=True)
+        self._started = False
+        self._lock = threading.Lock()
+
+    # ------------------------------------------------------------- lifecycle
+    def start(self) -> None:
+        if not self._started:
+            self._started = True
+            self._thread.start()
+
+    def stop(self) -> None:
+        if self._started:
+            self._stop.set()
+            self._queue.put(None)
+            self._thread.join(timeout=30)
+            self._started = False
+
+    @property
+    def stats(self) -> dict:
+        return dict(self._stats)
+
+    # ---------------------------------------------------------------- submit
+    def submit(self, request: GenRequest) -> Future:
+        """Thread-safe; returns a Future[GenResult]."""
+        self.start()
+        fut: Future = Future()
+        with self._lock:
+            self._next_id += 1
+            sid = self._next_id
+        if len(request.prompt_ids) + request.max_new_tokens > self.max_model_len:
+            request.prompt_ids = request.prompt_ids[
+                -(self.max_model_len - request.max_new_tokens):]
+        self._queue.put(_Seq(seq_id=sid, request=request, future=fut))
+        return fut
+
+    def generate(self, request: GenRequest, timeout: float = 600.0) -> GenResult:
+        return self.submit(request).result(timeout)
+
+    # ------------------------------------------------------------------ loop
+    def _loop(self) -> None:
+        try:
+            if self.use_graphs:
+                self._capture_graphs()
+        except Exception:
+            _log.exception("hipGraph capture failed; falling back to eager")
+            self.use_graphs = False
+        while not self._stop.is_set():
+            self._drain_queue(block=not (self._running or self._waiting))
+            if self._stop.is_set():
+                break
+            try:
+                # admit: prefill waiting sequences while capacity remains
+                while self._waiting and len(self._running) < self.max_batch:
+                    seq = self._waiting.pop(0)
+                    self._prefill(seq)
+                if self._running:
+                    self._decode_step()
+            except Exception as e:
+                _log.exception("engine step failed")
+                for seq in self._running + self._waiting:
+                    if not seq.future.done():
+                        seq.future.set_exception(e)
+                    self.kv.release(seq.seq_id)
+                self._running.clear()
+                self._waiting.clear()
+
+    def _drain_queue(self, block: bool) -> None:
+        try:
+            item = self._queue.get(timeout=0.05 if block else 0.0)
+        except queue.Empty:
+            return
+        if item is not None:
+            self._waiting.append(item)
+        while True:
+            try:
+                item = self._queue.get_nowait()
+            except queue.Empty:
+                return
+            if item is not None:
+                self._waiting.append(item)
+
+    # --------------------------------------------------------------- prefill
+    def _prefill(self, seq: _Seq) -> None:
+        t0 = time.perf_counter()
+        ids = seq.request.prompt_ids
+        n = len(ids)
+        self.kv.open(seq.seq_id)
+        model = self.model
+        dev = self.device
+        logits = None
+        done = 0
+        while done < n:
+            chunk = ids[done:done + PREFILL_CHUNK]
+            t = len(chunk)
+            slots = self.kv.slots_for(seq.seq_id, t)
+            tokens = torch.as_tensor(chunk, dtype=torch.int64, device=dev)
+            positions = torch.arange(done, done + t, dtype=torch.int32,
+                                     device=dev)
+            slots_t = torch.as_tensor(slots, dtype=torch.int32, device=dev)
+            logits = model.forward(
+                tokens, positions, self.kv.kcache, self.kv.vcache, slots_t,
+                prefill_scratch=self._prefill_kv, past_len=done,
+                logits_rows=slice(t - 1, t))
+            done += t
+        assert logits is not None
+        first = self._sample_host(logits, seq)
+        if self.on_gpu:
+            torch.cuda.synchronize(dev)
+        seq.result.prompt_tokens = n
+        seq.result.prefill_ms += (time.perf_counter() - t0) * 1e3
+        self._stats["prefills"] += 1
+        self._stats["prefill_ms"] += seq.result.prefill_ms
+        self._emit_token(seq, first)
+        if not self._finish_if_done(seq):
+            self._running.append(seq)
+
+    def _sample_host(self, logits: torch.Tensor, seq: _Seq) -> int:
+        r = seq.request
+        out = torch.zeros(1, dtype=torch.int32, device=logits.device)
+        temp = torch.full((1,), r.temperature, device=logits.device)
+        seed = torch.as_tensor([r.seed or seq.seq_id, self._step_counter],
+                               dtype=torch.int64, device=logits.device)
+        topk = torch.full((1,), r.top_k, dtype=torch.int32, device=logits.device)
+        from .. import ops
+        ops.sample(out, logits.float(), temp, seed, topk if r.top_k else None)
+        return int(out.item())
+
+    # ---------------------------------------------------------------- decode
+    def _pad_batch(self, b: int) -> int:
+        for g in self.GRAPH_BATCHES:
+            if b <= g:
+                return g
+        return self.max_batch
+
+    def _write_batch_state(self, pad: int) -> None:
+        B = len(self._running)
+        tokens = [s.last_token for s in self._running] + [0] * (pad - B)
+        positions, slots, lens = [], [], []
+        pt_rows = []
+        for s in self._running:
+            slot = self.kv.slots_for(s.seq_id, 1)[0]
+            slots.append(slot)
+            ln = self.kv.length(s.seq_id)  # includes the new token
+            lens.append(ln)
+            positions.append(ln - 1)
+            pages = self.kv._slots[s.seq_id].pages
+            pt_rows.append(pages + [0] * (self.max_pages - len(pages)))
+        for _ in range(pad - B):
+            slots.append(self._dummy_page * PAGE_SIZE)
+            lens.append(1)
+            positions.append(0)
+            pt_rows.append([self._dummy_page] + [0] * (self.max_pages - 1))
+        dev = self.device
+        self._tokens[:pad].copy_(torch.as_tensor(tokens, dtype=torch.int64),
+                                 non_blocking=True)
+        self._positions[:pad].copy_(
+            torch.as_tensor(positions, dtype=torch.int32), non_blocking=True)
+        self._slots[:pad].copy_(torch.as_tensor(slots, dtype=torch.int32),
+                                non_blocking=True)
+        self._seq_lens[:pad].copy_(torch.as_tensor(lens, dtype=torch.int32),
+                                   non_blocking=True)
+        self._page_table[:pad].copy_(
+            torch.as_tensor(pt_rows, dtype=torch.int32), non_blocking=True)
+        temps = [s.request.temperature for s in self._running] + [0.0] * (pad - B)
+        topks = [s.request.top_k for s in self._running] + [0] * (pad - B)
+        self._temperature[:pad].copy_(torch.as_tensor(temps), non_blocking=True)
+        self._topk[:pad].copy_(torch.as_tensor(topks, dtype=torch.int32),
+                               non_blocking=True)
+        self._step_counter += 1
+        base_seed = self._running[0].request.seed or 12345
+        self._seed_step.copy_(torch.as_tensor(
+            [base_seed, self._step_counter], dtype=torch.int64),
+            non_blocking=True)
+
+    def _decode_batch_gpu(self, pad: int) -> None:
+        """The captured region: one decode forward + sampling, batch ``pad``."""
+        model = self.model
+        logits = model.forward(
+            self._tokens[:pad], self._positions[:pad], self.kv.kcache,
+            self.kv.vcache, self._slots[:pad],
+            page_table=self._page_table[:pad], seq_lens=self._seq_lens[:pad])
+        from .. import ops
+        ops.sample(self._sampled[:pad], logits.float(), self._temperature[:pad],
+                   self._seed_step, self._topk[:pad])
+
+    def _capture_graphs(self) -> None:
+        """Capture one hipGraph per padded batch size, largest first."""
+        torch.cuda.synchronize(self.device)
+        for b in reversed([g for g in self.GRAPH_BATCHES if g <= self.max_batch]):
+            # warm up allocations outside capture
+            self._decode_batch_gpu(b)
+            torch.cuda.synchronize(self.device)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._decode_batch_gpu(b)
+            self._graphs[b] = g
+        torch.cuda.synchronize(self.device)
+        _log.info("captured %d decode graphs: %s", len(self._graphs),
+                  sorted(self._graphs))
+
+    def _decode_step(self) -> None:
+        t0 = time.perf_counter()
+        B = len(self._running)
+        pad = self._pad_batch(B)
+        self._write_batch_state(pad)
+        if self.use_graphs and pad in self._graphs:
+            self._graphs[pad].replay()
+            self._stats["graph_replays"] += 1
+        else:
+            self._decode_batch_gpu(pad)
+        sampled = self._sampled[:B].cpu().tolist()
+        dt = (time.perf_counter() - t0) * 1e3
+        self._stats["decode_steps"] += 1
+        self._stats["decode_tokens"] += B
+        self._stats["decode_ms"] += dt
+        still: list[_Seq] = []
+        for seq, tok in zip(self._running, sampled):
+            seq.result.decode_ms += dt
+            self._emit_token(seq, int(tok))
+            if not self._finish_if_done(seq):
+                still.append(seq)
+        self._running = still
+
+    # --------------------------------------------------------------- helpers
+    def _emit_token(self, seq: _Seq, token: int) -> None:
+        seq.last_token = token
+        seq.new_tokens += 1
+        seq.result.token_ids.append(token)
+        cb = seq.request.on_token
+        if cb is not None:
+            try:
+                cb(token)
+            except Exception:
+                _log.exception("on_token callback failed")
+
+    def _finish_if_done(self, seq: _Seq) -> bool:
+        r = seq.request
+        eos_hit = (not r.ignore_eos) and seq.last_token == 2
+        if seq.new_tokens >= r.max_new_tokens or eos_hit:
+            seq.result.finish_reason = "stop" if eos_hit else "length"
+            self.kv.release(seq.seq_id)
+            if not seq.future.done():
+                seq.future.set_result(seq.result)
+            return True
+        return False

@@ -1,0 +1,145 @@
+"""LocalModelClient — the in-process Provider backed by the CDNA4 engine.
+
+This replaces the reference's remote-HTTP providers behind the same
+``Provider`` port: ``complete(EngineRequest, sink)`` renders
+the chat prompt, submits to the shared per-process InferenceEngine (one per
+GPU rank — concurrent agent turns batch into one decode), streams token
+deltas to the sink, and returns an AssistantMessage whose ProviderFacts
+carry the measured prefill/decode times for the StepLedger.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import threading
+from typing import Any
+
+import torch
+
+from ..messages import AssistantMessage, Message, ProviderFacts, TextPart, Usage
+from ..schemas import EngineRequest
+from ..streaming import ChunkSink, StreamChunk
+from .engine import GenRequest, InferenceEngine
+from .tokenizer import SyntheticTokenizer
+
+__all__ = ["LocalModelClient", "shared_engine"]
+
+_ENGINES: dict[tuple, InferenceEngine] = {}
+_ENGINES_LOCK = threading.Lock()
+
+
+def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
+                  **engine_kw: Any) -> InferenceEngine:
+    """One engine per (model, device) per process — all agents share it.
+
+    ``tp`` > 1 shards the model across the torch.distributed world (every
+    rank must construct the same engine and drive the same forward order);
+    the default is one independent replica per process (the
+    one-Worker-per-GPU scaling story).
+    """
+    from ..models.config import ModelConfig, get_config
+    from ..models.llama import LlamaShard
+
+    if isinstance(config, str):
+        config = get_config(config)
+    assert isinstance(config, ModelConfig)
+    if device is None:
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+    key = (config.name, device, tp)
+    with _ENGINES_LOCK:
+        eng = _ENGINES.get(key)
+        if eng is None:
+            if config.is_moe:
+                from ..models.mixtral import MixtralShard as Shard
+            else:
+                Shard = LlamaShard
+            if tp > 1:
+                from ..parallel import get_dist
+
+                dist = get_dist()
+                if dist.world_size != tp:
+                    raise ValueError(
+                        f"tp={tp} but torch.distributed world is {dist.world_size}")
+                model = Shard(config, tp_rank=dist.rank, tp_size=tp, device=device)
+            else:
+                model = Shard(config, tp_rank=0, tp_size=1, device=device)
+            eng = InferenceEngine(model, **engine_kw)
+            eng.start()
+            _ENGINES[key] = eng
+    return eng
+
+
+class LocalModelClient:
+    """Provider port implementation over the in-process engine."""
+
+    def __init__(self, model: Any = "llama-3-8b", *, device: str | None = None,
+                 max_batch: int = 64, max_model_len: int | None = None,
+                 use_graphs: bool | None = None, kv_pages: int | None = None):
+        self._config = model
+        self._device = device
+        self._engine_kw = dict(max_batch=max_batch, max_model_len=max_model_len,
+                               use_graphs=use_graphs, kv_pages=kv_pages)
+        self._engine: InferenceEngine | None = None
+        self._tokenizer: SyntheticTokenizer | None = None
+
+    # lazily built so Worker/Agent construction never touches the GPU
+    @property
+    def engine(self) -> InferenceEngine:
+        if self._engine is None:
+            self._engine = shared_engine(self._config, self._device,
+                                         **self._engine_kw)
+            self._tokenizer = SyntheticTokenizer(self._engine.cfg.vocab_size)
+        return self._engine
+
+    @property
+    def tokenizer(self) -> SyntheticTokenizer:
+        self.engine
+        assert self._tokenizer is not None
+        return self._tokenizer
+
+    # ------------------------------------------------------------ rendering
+    @staticmethod
+    def render_prompt(messages: tuple[Message, ...]) -> str:
+        lines = []
+        for m in messages:
+            text = " ".join(p.text for p in m.parts if isinstance(p, TextPart))
+            lines.append(f"<|{m.role}|> {text}")
+        lines.append("<|assistant|>")
+        return "\n".join(lines)
+
+    # ------------------------------------------------------------- complete
+    async def complete(self, request: EngineRequest, sink: ChunkSink) -> AssistantMessage:
+        engine = self.engine
+        tok = self.tokenizer
+        prompt = self.render_prompt(request.messages)
+        ids = tok.encode(prompt)
+        s = request.settings
+        seq = 0
+
+        def on_token(t: int) -> None:
+            nonlocal seq
+            seq += 1
+            sink.emit(StreamChunk(event="text_delta", seq=seq,
+                                  text=tok.decode([t]) + " "))
+
+        gen = GenRequest(
+            prompt_ids=ids,
+            max_new_tokens=max(1, s.max_tokens),
+            temperature=s.temperature,
+            top_k=s.top_k,
+            seed=s.seed or 0,
+            ignore_eos=False,
+            on_token=on_token,
+        )
+        fut = engine.submit(gen)
+        result = await asyncio.wrap_future(fut)
+        text = tok.decode(result.token_ids)
+        facts = ProviderFacts(
+            model_name=engine.cfg.name,
+            usage=Usage(prompt_tokens=result.prompt_tokens,
+                        completion_tokens=len(result.token_ids)),
+            finish_reason=result.finish_reason,
+            prefill_ms=result.prefill_ms,
+            decode_ms=result.decode_ms,
+        )
+        return AssistantMessage((TextPart(text),), facts=facts)

@@ -1,0 +1,36 @@
+"""Deterministic synthetic tokenizer.
+
+There is no network for real tokenizer downloads and BASELINE runs on
+synthetic data / random-init weights, so tokenization just needs to be
+deterministic, stable across processes, and shaped like the real thing
+(BOS/EOS, roughly word-per-token).
+"""
+
+from __future__ import annotations
+
+import hashlib
+import re
+
+__all__ = ["SyntheticTokenizer"]
+
+_WORD = re.compile(r"\S+")
+
+
+class SyntheticTokenizer:
+    BOS = 1
+    EOS = 2
+    _RESERVED = 16  # ids 0..15 reserved for specials
+
+    def __init__(self, vocab_size: int):
+        self.vocab_size = vocab_size
+
+    def encode(self, text: str, *, bos: bool = True) -> list[int]:
+        ids = [self.BOS] if bos else []
+        span = self.vocab_size - self._RESERVED
+        for m in _WORD.finditer(text):
+            h = hashlib.blake2s(m.group().encode(), digest_size=4).digest()
+            ids.append(self._RESERVED + int.from_bytes(h, "little") % span)
+        return ids
+
+    def decode(self, ids: list[int]) -> str:
+        return " ".join(f"t{i}" for i in ids if i >= self._RESERVED)

@@ -1,0 +1,164 @@
+"""Llama-family model (3-8B / 3-70B shapes) on the CDNA4 op set.
+
+Design (MI355X-first, not a port):
+- projections run on hipBLASLt through torch.matmul (plain library GEMMs);
+  every fused op — residual+RMSNorm, RoPE, paged attention, SiLU·mul,
+  sampling — is a hand-written HIP kernel (calfkit_amd.ops);
+- tensor parallel: column-parallel QKV/gate-up, row-parallel O/down with one
+  RCCL all-reduce per block pair; KV heads shard with the TP group
+  (Hkv=8 → TP≤8, one KV head per rank at TP=8);
+- weights are random-init bf16 shards created directly on the device —
+  there is no full-model materialization and no checkpoint download
+  (BASELINE: synthetic data / random-init weights);
+- decode path is shape-static per batch size so the whole step can be
+  captured in a hipGraph (engine/runner.py).
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import torch
+
+from .. import ops
+from ..parallel import all_reduce_sum, shard_dim
+from .config import ModelConfig
+
+__all__ = ["LlamaShard", "LayerWeights"]
+
+
+@dataclass
+class LayerWeights:
+    wqkv: torch.Tensor      # [hidden, (hq_s + 2*hkv_s) * dh]
+    wo: torch.Tensor        # [hq_s * dh, hidden]
+    wgateup: torch.Tensor   # [hidden, 2 * ffn_s]
+    wdown: torch.Tensor     # [ffn_s, hidden]
+    norm_attn: torch.Tensor
+    norm_mlp: torch.Tensor
+
+
+class LlamaShard:
+    """One TP rank's shard of a Llama model, random-init, inference-only."""
+
+    def __init__(self, cfg: ModelConfig, *, tp_rank: int = 0, tp_size: int = 1,
+                 device: torch.device | str = "cpu",
+                 dtype: torch.dtype = torch.bfloat16, seed: int = 1234):
+        if cfg.heads % tp_size or cfg.kv_heads % tp_size or cfg.ffn % tp_size:
+            raise ValueError(f"{cfg.name}: heads/kv/ffn must divide tp={tp_size}")
+        self.cfg = cfg
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.hq = cfg.heads // tp_size
+        self.hkv = cfg.kv_heads // tp_size
+        self.ffn_s = cfg.ffn // tp_size
+        self.dh = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(self.dh)
+
+        gen = torch.Generator(device="cpu").manual_seed(seed + tp_rank)
+        d = cfg.hidden
+
+        def w(rows: int, cols: int, fan_in: int) -> torch.Tensor:
+            std = 1.0 / math.sqrt(fan_in)
+            t = torch.randn(rows, cols, generator=gen, dtype=torch.float32) * std
+            return t.to(device=self.device, dtype=dtype)
+
+        self.embed = w(cfg.vocab_size, d, d)
+        self.layers: list[LayerWeights] = []
+        qkv_out = (self.hq + 2 * self.hkv) * self.dh
+        for _ in range(cfg.layers):
+            self.layers.append(LayerWeights(
+                wqkv=w(d, qkv_out, d),
+                wo=w(self.hq * self.dh, d, cfg.q_dim),
+                wgateup=w(d, 2 * self.ffn_s, d),
+                wdown=w(self.ffn_s, d, cfg.ffn),
+                norm_attn=torch.ones(d, device=self.device, dtype=dtype),
+                norm_mlp=torch.ones(d, device=self.device, dtype=dtype),
+            ))
+        self.norm_final = torch.ones(d, device=self.device, dtype=dtype)
+        # lm_head replicated (vocab-parallel sharding is a later optimization;
+        # 128256 x 8192 bf16 is ~2 GB of the 288 GB rank)
+        self.lm_head = self.embed if cfg.tie_embeddings else w(cfg.vocab_size, d, d)
+        self.rope = ops.build_rope_table(cfg.max_seq, self.dh, cfg.rope_theta,
+                                         device=self.device)
+
+    # ------------------------------------------------------------------ util
+    def param_bytes(self) -> int:
+        n = self.embed.numel() + self.lm_head.numel() + self.norm_final.numel()
+        for lw in self.layers:
+            n += sum(t.numel() for t in
+                     (lw.wqkv, lw.wo, lw.wgateup, lw.wdown, lw.norm_attn,
+                      lw.norm_mlp))
+        return n * self.embed.element_size()
+
+    # ---------------------------------------------------------------- layers
+    def _attn(self, lw: LayerWeights, hidden: torch.Tensor,
+              positions: torch.Tensor, kcache: torch.Tensor,
+              vcache: torch.Tensor, slots: torch.Tensor, *,
+              prefill_kv: tuple[torch.Tensor, torch.Tensor] | None,
+              page_table: torch.Tensor | None,
+              seq_lens: torch.Tensor | None, past_len: int) -> torch.Tensor:
+        T = hidden.shape[0]
+        qkv = hidden @ lw.wqkv
+        q, k, v = qkv.split([self.hq * self.dh, self.hkv * self.dh,
+                             self.hkv * self.dh], dim=-1)
+        q = q.view(T, self.hq, self.dh).contiguous()
+        k = k.view(T, self.hkv, self.dh).contiguous()
+        v = v.view(T, self.hkv, self.dh).contiguous()
+        ops.rope_qk(q, k, self.rope, positions)
+        ops.kv_write(kcache, vcache, k, v, slots)
+        o = torch.empty_like(q)
+        if prefill_kv is not None:
+            kfull, vfull = prefill_kv
+            kfull[past_len:past_len + T] = k
+            vfull[past_len:past_len + T] = v
+            ops.attn_prefill(o, q, kfull[:past_len + T], vfull[:past_len + T],
+                             self.scale, past_len)
+        else:
+            ops.attn_decode(o, q, kcache, vcache, page_table, seq_lens, self.scale)
+        out = o.view(T, self.hq * self.dh) @ lw.wo
+        return all_reduce_sum(out) if self.tp_size > 1 else out
+
+    def _mlp(self, lw: LayerWeights, hidden: torch.Tensor) -> torch.Tensor:
+        gateup = hidden @ lw.wgateup
+        act = torch.empty(hidden.shape[0], self.ffn_s, dtype=hidden.dtype,
+                          device=hidden.device)
+        ops.silu_mul(act, gateup)
+        down = act @ lw.wdown
+        return all_reduce_sum(down) if self.tp_size > 1 else down
+
+    # --------------------------------------------------------------- forward
+    def forward(self, tokens: torch.Tensor, positions: torch.Tensor,
+                kcaches: list[torch.Tensor], vcaches: list[torch.Tensor],
+                slots: torch.Tensor, *,
+                prefill_scratch: list[tuple[torch.Tensor, torch.Tensor]] | None = None,
+                page_table: torch.Tensor | None = None,
+                seq_lens: torch.Tensor | None = None,
+                past_len: int = 0,
+                logits_rows: slice | None = None) -> torch.Tensor:
+        """One forward pass.  Prefill mode when ``prefill_scratch`` is given
+        (single sequence, T tokens); decode mode otherwise (B sequences,
+        one token each, via the paged tables).
+
+        Returns logits [rows, vocab] (rows = T or B or the logits_rows slice).
+        """
+        hidden = self.embed[tokens.long()]  # [T, d]
+        residual = torch.zeros_like(hidden)
+        for i, lw in enumerate(self.layers):
+            ops.fused_add_rmsnorm(hidden, residual, lw.norm_attn,
+                                  self.cfg.norm_eps)
+            hidden = self._attn(
+                lw, hidden, positions, kcaches[i], vcaches[i], slots,
+                prefill_kv=prefill_scratch[i] if prefill_scratch else None,
+                page_table=page_table, seq_lens=seq_lens, past_len=past_len)
+            ops.fused_add_rmsnorm(hidden, residual, lw.norm_mlp,
+                                  self.cfg.norm_eps)
+            hidden = self._mlp(lw, hidden)
+        # final residual add + norm
+        ops.fused_add_rmsnorm(hidden, residual, self.norm_final,
+                              self.cfg.norm_eps)
+        if logits_rows is not None:
+            hidden = hidden[logits_rows]
+        return hidden @ self.lm_head.T
