@@ -37,6 +37,7 @@ _log = logging.getLogger("calfkit.engine")
 __all__ = ["GenRequest", "GenResult", "InferenceEngine"]
 
 PREFILL_CHUNK = 512
+PREFILL_BUDGET = 8192
 
 
 @dataclass
@@ -177,10 +178,24 @@ class InferenceEngine:
             if self._stop.is_set():
                 break
             try:
-                # admit: prefill waiting sequences while capacity remains
+                # admit: pack waiting sequences into ONE varlen prefill
+                # forward (token-budgeted); oversized prompts go chunked solo
                 while self._waiting and len(self._running) < self.max_batch:
-                    seq = self._waiting.pop(0)
-                    self._prefill(seq)
+                    batch: list[_Seq] = []
+                    budget = PREFILL_BUDGET
+                    while (self._waiting and budget > 0
+                           and len(self._running) + len(batch) < self.max_batch):
+                        n = len(self._waiting[0].request.prompt_ids)
+                        if n > PREFILL_BUDGET:
+                            break  # long prompt: chunked solo below
+                        if n > budget:
+                            break
+                        budget -= n
+                        batch.append(self._waiting.pop(0))
+                    if batch:
+                        self._prefill_batch(batch)
+                    elif self._waiting:
+                        self._prefill(self._waiting.pop(0))
                 if self._running:
                     self._decode_step()
             except Exception as e:
@@ -241,6 +256,75 @@ class InferenceEngine:
         self._emit_token(seq, first)
         if not self._finish_if_done(seq):
             self._running.append(seq)
+
+    def _prefill_batch(self, seqs: list["_Seq"]) -> None:
+        """Pack every waiting prompt into ONE varlen prefill forward."""
+        if len(seqs) == 1 and len(seqs[0].request.prompt_ids) > PREFILL_CHUNK:
+            return self._prefill(seqs[0])
+        t0 = time.perf_counter()
+        model = self.model
+        dev = self.device
+        lens = [len(s.request.prompt_ids) for s in seqs]
+        total = sum(lens)
+        ids: list[int] = []
+        positions: list[int] = []
+        slots: list[int] = []
+        starts = [0]
+        for s, n in zip(seqs, lens):
+            self.kv.open(s.seq_id)
+            ids.extend(s.request.prompt_ids)
+            positions.extend(range(n))
+            slots.extend(self.kv.slots_for(s.seq_id, n))
+            starts.append(starts[-1] + n)
+        from .. import ops
+        from ..models.llama import PrefillPlan
+
+        q_starts = torch.as_tensor(starts, dtype=torch.int32).to(dev)
+        plan = PrefillPlan(
+            scratch=self._prefill_kv,
+            block_map=ops.build_prefill_blocks(lens).to(dev),
+            q_starts=q_starts,
+            kv_starts=q_starts,
+            past_lens=torch.zeros(len(seqs), dtype=torch.int32, device=dev),
+            kv_write_start=0,
+            kv_total=total,
+            kv_in_qkv=True)
+        logits = model.forward(
+            torch.as_tensor(ids, dtype=torch.int64).to(dev),
+            torch.as_tensor(positions, dtype=torch.int32).to(dev),
+            self.kv.kcache, self.kv.vcache,
+            torch.as_tensor(slots, dtype=torch.int32).to(dev),
+            prefill_plan=plan,
+            logits_rows=(q_starts[1:].long() - 1))
+        firsts = self._sample_batch_host(logits, seqs)
+        if self.on_gpu:
+            torch.cuda.synchronize(self.device)
+        ms = (time.perf_counter() - t0) * 1e3
+        self._stats["prefills"] += len(seqs)
+        self._stats["prefill_ms"] += ms
+        for s, n, tok in zip(seqs, lens, firsts):
+            s.result.prompt_tokens = n
+            s.result.prefill_ms += ms * n / max(1, total)  # attribute by share
+            self._emit_token(s, tok)
+            if not self._finish_if_done(s):
+                self._running.append(s)
+
+    def _sample_batch_host(self, logits: torch.Tensor,
+                           seqs: list["_Seq"]) -> list[int]:
+        from .. import ops
+
+        B = len(seqs)
+        out = torch.zeros(B, dtype=torch.int32, device=logits.device)
+        temp = torch.as_tensor([s.request.temperature for s in seqs],
+                               dtype=torch.float32).to(logits.device)
+        topk = torch.as_tensor([s.request.top_k for s in seqs],
+                               dtype=torch.int32).to(logits.device)
+        seed = torch.as_tensor(
+            [seqs[0].request.seed or seqs[0].seq_id, self._step_counter],
+            dtype=torch.int64).to(logits.device)
+        ops.sample(out, logits.float(), temp, seed,
+                   topk if any(s.request.top_k for s in seqs) else None)
+        return [int(t) for t in out.cpu()]
 
     def _sample_host(self, logits: torch.Tensor, seq: _Seq) -> int:
         r = seq.request

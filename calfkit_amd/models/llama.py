@@ -25,7 +25,23 @@ from .. import ops
 from ..parallel import all_reduce_sum, shard_dim
 from .config import ModelConfig
 
-__all__ = ["LlamaShard", "LayerWeights"]
+__all__ = ["LlamaShard", "LayerWeights", "PrefillPlan"]
+
+
+@dataclass
+class PrefillPlan:
+    """Packed varlen prefill description (built once per prefill batch)."""
+
+    scratch: list  # [(k_buf, v_buf)] per layer, packed kv write targets
+    block_map: torch.Tensor   # [nblocks, 2] i32 (seq, qblock)
+    q_starts: torch.Tensor    # [B+1] i32 prefix of packed q rows
+    kv_starts: torch.Tensor   # [B+1] i32 prefix of packed kv rows
+    past_lens: torch.Tensor   # [B] i32
+    kv_write_start: int       # where this forward's new kv rows land
+    kv_total: int             # total kv rows the attention reads
+    #: True when every past_len is 0 and kv rows == q rows: attention reads
+    #: K/V straight from the packed qkv GEMM output (no scratch copy at all)
+    kv_in_qkv: bool = False
 
 
 @dataclass
@@ -98,24 +114,36 @@ class LlamaShard:
               positions: torch.Tensor, kcache: torch.Tensor,
               vcache: torch.Tensor, slots: torch.Tensor, *,
               prefill_kv: tuple[torch.Tensor, torch.Tensor] | None,
+              plan: "PrefillPlan | None",
               page_table: torch.Tensor | None,
-              seq_lens: torch.Tensor | None, past_len: int) -> torch.Tensor:
+              seq_lens: torch.Tensor | None) -> torch.Tensor:
         T = hidden.shape[0]
         qkv = hidden @ lw.wqkv
-        q, k, v = qkv.split([self.hq * self.dh, self.hkv * self.dh,
-                             self.hkv * self.dh], dim=-1)
-        q = q.view(T, self.hq, self.dh).contiguous()
-        k = k.view(T, self.hkv, self.dh).contiguous()
-        v = v.view(T, self.hkv, self.dh).contiguous()
-        ops.rope_qk(q, k, self.rope, positions)
-        ops.kv_write(kcache, vcache, k, v, slots)
-        o = torch.empty_like(q)
-        if prefill_kv is not None:
-            kfull, vfull = prefill_kv
-            kfull[past_len:past_len + T] = k
-            vfull[past_len:past_len + T] = v
-            ops.attn_prefill(o, q, kfull[:past_len + T], vfull[:past_len + T],
-                             self.scale, past_len)
+        # fused epilogue: in-place RoPE on the packed q/k heads + paged-cache
+        # scatter — q/k/v stay strided views of the GEMM output (no copies)
+        ops.rope_qkv_write(qkv, kcache, vcache, self.rope, positions, slots,
+                           self.hq, self.hkv)
+        q = qkv[:, :self.hq * self.dh].view(T, self.hq, self.dh)
+        o = torch.empty(T, self.hq, self.dh, dtype=qkv.dtype, device=qkv.device)
+        if plan is not None:
+            if plan.kv_in_qkv:
+                # fresh packed prefill: the new rows ARE the whole kv set
+                kk = qkv[:, self.hq * self.dh:(self.hq + self.hkv) * self.dh]
+                vv = qkv[:, (self.hq + self.hkv) * self.dh:]
+                kk = kk.view(T, self.hkv, self.dh)
+                vv = vv.view(T, self.hkv, self.dh)
+            else:
+                kfull, vfull = prefill_kv
+                w0 = plan.kv_write_start
+                kfull[w0:w0 + T] = qkv[:, self.hq * self.dh:
+                                       (self.hq + self.hkv) * self.dh].view(
+                                           T, self.hkv, self.dh)
+                vfull[w0:w0 + T] = qkv[:, (self.hq + self.hkv) * self.dh:].view(
+                    T, self.hkv, self.dh)
+                kk, vv = kfull[:plan.kv_total], vfull[:plan.kv_total]
+            ops.attn_prefill_varlen(
+                o, q, kk, vv, plan.block_map, plan.q_starts, plan.kv_starts,
+                plan.past_lens, self.scale)
         else:
             ops.attn_decode(o, q, kcache, vcache, page_table, seq_lens, self.scale)
         out = o.view(T, self.hq * self.dh) @ lw.wo
@@ -130,20 +158,45 @@ class LlamaShard:
         return all_reduce_sum(down) if self.tp_size > 1 else down
 
     # --------------------------------------------------------------- forward
+    def make_single_plan(self, scratch, T: int, past_len: int) -> "PrefillPlan":
+        """Plan for one sequence of T new tokens over past_len context."""
+        dev = self.device
+        nb = (T + 63) // 64
+        bm = torch.stack([torch.zeros(nb, dtype=torch.int32),
+                          torch.arange(nb, dtype=torch.int32)], 1).to(dev)
+        return PrefillPlan(
+            scratch=scratch,
+            block_map=bm.contiguous(),
+            q_starts=torch.tensor([0, T], dtype=torch.int32, device=dev),
+            kv_starts=torch.tensor([0, past_len + T], dtype=torch.int32,
+                                   device=dev),
+            past_lens=torch.tensor([past_len], dtype=torch.int32, device=dev),
+            kv_write_start=past_len,
+            kv_total=past_len + T,
+            kv_in_qkv=(past_len == 0))
+
     def forward(self, tokens: torch.Tensor, positions: torch.Tensor,
                 kcaches: list[torch.Tensor], vcaches: list[torch.Tensor],
                 slots: torch.Tensor, *,
                 prefill_scratch: list[tuple[torch.Tensor, torch.Tensor]] | None = None,
+                prefill_plan: "PrefillPlan | None" = None,
                 page_table: torch.Tensor | None = None,
                 seq_lens: torch.Tensor | None = None,
                 past_len: int = 0,
-                logits_rows: slice | None = None) -> torch.Tensor:
-        """One forward pass.  Prefill mode when ``prefill_scratch`` is given
-        (single sequence, T tokens); decode mode otherwise (B sequences,
-        one token each, via the paged tables).
+                logits_rows: "slice | torch.Tensor | None" = None) -> torch.Tensor:
+        """One forward pass.  Prefill mode when a plan (or the single-seq
+        ``prefill_scratch``+``past_len`` convenience form) is given; decode
+        mode otherwise (B sequences, one token each, via the paged tables).
 
-        Returns logits [rows, vocab] (rows = T or B or the logits_rows slice).
+        Returns logits [rows, vocab] (rows = T or B or the logits_rows
+        slice/index).
         """
+        plan = prefill_plan
+        if plan is None and prefill_scratch is not None:
+            plan = self.make_single_plan(prefill_scratch, tokens.shape[0],
+                                         past_len)
+        elif plan is not None:
+            prefill_scratch = plan.scratch
         hidden = self.embed[tokens.long()]  # [T, d]
         residual = torch.zeros_like(hidden)
         for i, lw in enumerate(self.layers):
@@ -152,7 +205,7 @@ class LlamaShard:
             hidden = self._attn(
                 lw, hidden, positions, kcaches[i], vcaches[i], slots,
                 prefill_kv=prefill_scratch[i] if prefill_scratch else None,
-                page_table=page_table, seq_lens=seq_lens, past_len=past_len)
+                plan=plan, page_table=page_table, seq_lens=seq_lens)
             ops.fused_add_rmsnorm(hidden, residual, lw.norm_mlp,
                                   self.cfg.norm_eps)
             hidden = self._mlp(lw, hidden)

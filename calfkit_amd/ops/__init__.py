@@ -22,9 +22,12 @@ __all__ = [
     "fused_add_rmsnorm",
     "silu_mul",
     "rope_qk",
+    "rope_qkv_write",
     "kv_write",
     "attn_decode",
     "attn_prefill",
+    "attn_prefill_varlen",
+    "build_prefill_blocks",
     "sample",
     "build_rope_table",
 ]
@@ -120,6 +123,26 @@ def rope_qk(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
         t[..., half:] = (x1 * cos + x0 * sin).to(t.dtype)
 
 
+def rope_qkv_write(qkv: torch.Tensor, kcache: torch.Tensor,
+                   vcache: torch.Tensor, cos_sin: torch.Tensor,
+                   positions: torch.Tensor, slots: torch.Tensor,
+                   hq: int, hkv: int) -> None:
+    """Fused QKV-projection epilogue: rotate Q/K heads in place inside the
+    packed [T, (hq+2*hkv)*dh] GEMM output and scatter K/V into the paged
+    cache (one launch instead of 3 copies + rope + kv_write)."""
+    if _native(qkv):
+        _K.rope_qkv_write(qkv, kcache, vcache, cos_sin, positions, slots,
+                          hq, hkv)
+        return
+    dh = kcache.shape[-1]
+    T = qkv.shape[0]
+    q = qkv[:, :hq * dh].view(T, hq, dh)
+    k = qkv[:, hq * dh:(hq + hkv) * dh].view(T, hkv, dh)
+    v = qkv[:, (hq + hkv) * dh:].view(T, hkv, dh)
+    rope_qk(q, k, cos_sin, positions)
+    kv_write(kcache, vcache, k, v, slots)
+
+
 # ----------------------------------------------------------------- kv cache
 
 def kv_write(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
@@ -182,6 +205,37 @@ def attn_prefill(out: torch.Tensor, q: torch.Tensor, k: torch.Tensor,
         s = s.masked_fill(mask, float("-inf"))
         p = torch.softmax(s, dim=-1)
         out[:, h] = (p @ v[:, kvh].float()).to(out.dtype)
+
+
+def build_prefill_blocks(seq_lens: list[int], qblk: int = 64) -> torch.Tensor:
+    """Host-side block map for varlen prefill: [(seq, qblock), ...] i32."""
+    rows = []
+    for s, n in enumerate(seq_lens):
+        for qb in range((n + qblk - 1) // qblk):
+            rows.append((s, qb))
+    return torch.tensor(rows, dtype=torch.int32)
+
+
+def attn_prefill_varlen(out: torch.Tensor, q: torch.Tensor, k: torch.Tensor,
+                        v: torch.Tensor, block_map: torch.Tensor,
+                        q_starts: torch.Tensor, kv_starts: torch.Tensor,
+                        past_lens: torch.Tensor, scale: float) -> None:
+    """Packed multi-sequence causal GQA prefill.
+
+    q/out: [Tq, Hq, dh] packed; k/v: [Tkv, Hkv, dh] packed; sequence i's new
+    rows are q[q_starts[i]:q_starts[i+1]] and its keys
+    k[kv_starts[i]:kv_starts[i+1]] (= past_lens[i] context + new rows).
+    """
+    if _native(q):
+        _K.attn_prefill_varlen(out, q, k, v, block_map, q_starts, kv_starts,
+                               past_lens, scale)
+        return
+    B = past_lens.numel()
+    for i in range(B):
+        qs, qe = int(q_starts[i]), int(q_starts[i + 1])
+        ks, ke = int(kv_starts[i]), int(kv_starts[i + 1])
+        attn_prefill(out[qs:qe], q[qs:qe], k[ks:ke], v[ks:ke], scale,
+                     int(past_lens[i]))
 
 
 # ----------------------------------------------------------------- sampling

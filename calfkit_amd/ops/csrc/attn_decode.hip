@@ -30,7 +30,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
     const __hip_bfloat16* __restrict__ vcache,
     const int32_t* __restrict__ page_table,
     const int32_t* __restrict__ seq_lens,
-    const int hq, const int hkv, const int max_pages, const float scale) {
+    const int hq, const int hkv, const int max_pages, const float scale,
+    const int q_stride) {   // row stride of q in elements (strided qkv view)
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
   const int wid = threadIdx.x / WAVE;
@@ -44,7 +45,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
 
   for (int i = threadIdx.x; i < G * HD; i += NWAVES * WAVE) {
     const int g = i / HD, d = i % HD;
-    q_lds[g][d] = bf2f(q[((int64_t)seq * hq + kvh * G + g) * HD + d]) * scale;
+    q_lds[g][d] =
+        bf2f(q[(int64_t)seq * q_stride + (kvh * G + g) * HD + d]) * scale;
   }
   __syncthreads();
 
@@ -104,20 +106,31 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
 
     // PV: lanes 0..31 take tokens 0..31, lanes 32..63 take 32..63; each
     // half-wave covers all 128 dims (32 lanes x 4 dims).  V row read is
-    // 8 B/lane coalesced within the half's 32 lanes.
+    // 8 B/lane coalesced within the half's 32 lanes.  ALL 32 rows prefetch
+    // up front (32 independent loads in flight) — the serial
+    // load->shuffle->fma version was latency-bound at 64.8 us/call
+    // (profiles/r01_bench_llama8b_kernelstats.md).
     const int t0 = thalf * 32;
     const int nt = min(32, slen - pi * KVPAGE - t0);
     const float2* vbase = reinterpret_cast<const float2*>(vcache + base);
-    for (int t = 0; t < nt; ++t) {
-      const float2 raw = vbase[(t0 + t) * (HD / 4) + dlane];
-      const __hip_bfloat16* vb = reinterpret_cast<const __hip_bfloat16*>(&raw);
+    if (nt > 0) {
+      float2 vbuf[32];
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        const float pg = __shfl(p[g], t0 + t, WAVE);
-        acc[g][0] = fmaf(pg, bf2f(vb[0]), acc[g][0]);
-        acc[g][1] = fmaf(pg, bf2f(vb[1]), acc[g][1]);
-        acc[g][2] = fmaf(pg, bf2f(vb[2]), acc[g][2]);
-        acc[g][3] = fmaf(pg, bf2f(vb[3]), acc[g][3]);
+      for (int t = 0; t < 32; ++t)
+        vbuf[t] = vbase[(t0 + min(t, nt - 1)) * (HD / 4) + dlane];
+#pragma unroll
+      for (int t = 0; t < 32; ++t) {
+        if (t >= nt) break;
+        const __hip_bfloat16* vb =
+            reinterpret_cast<const __hip_bfloat16*>(&vbuf[t]);
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float pg = __shfl(p[g], t0 + t, WAVE);
+          acc[g][0] = fmaf(pg, bf2f(vb[0]), acc[g][0]);
+          acc[g][1] = fmaf(pg, bf2f(vb[1]), acc[g][1]);
+          acc[g][2] = fmaf(pg, bf2f(vb[2]), acc[g][2]);
+          acc[g][3] = fmaf(pg, bf2f(vb[3]), acc[g][3]);
+        }
       }
     }
   }
@@ -177,7 +190,10 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
 void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                  torch::Tensor vcache, torch::Tensor page_table,
                  torch::Tensor seq_lens, double scale) {
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  // q may be a strided row view into the packed qkv GEMM output
+  TORCH_CHECK(out.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
+              "q heads must be contiguous");
   TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous());
   TORCH_CHECK(page_table.scalar_type() == torch::kInt32);
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
@@ -197,7 +213,7 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                      (const __hip_bfloat16*)vcache.data_ptr(),            \
                      (const int32_t*)page_table.data_ptr(),               \
                      (const int32_t*)seq_lens.data_ptr(), hq, hkv,        \
-                     max_pages, (float)scale)
+                     max_pages, (float)scale, (int)q.stride(0))
   switch (G) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;

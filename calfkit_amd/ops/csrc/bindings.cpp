@@ -8,6 +8,10 @@ void fused_add_rmsnorm(torch::Tensor hidden, torch::Tensor residual,
 void silu_mul(torch::Tensor out, torch::Tensor gateup);
 void rope_qk(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
              torch::Tensor positions);
+void rope_qkv_write(torch::Tensor qkv, torch::Tensor kcache,
+                    torch::Tensor vcache, torch::Tensor cos_sin,
+                    torch::Tensor positions, torch::Tensor slots,
+                    int64_t hq, int64_t hkv);
 void kv_write(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
               torch::Tensor v, torch::Tensor slots);
 void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
@@ -15,6 +19,10 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                  torch::Tensor seq_lens, double scale);
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   torch::Tensor v, double scale, int64_t past_len);
+void attn_prefill_varlen(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                         torch::Tensor v, torch::Tensor block_map,
+                         torch::Tensor q_starts, torch::Tensor kv_starts,
+                         torch::Tensor past_lens, double scale);
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor seed_step, torch::Tensor topk);
 
@@ -24,8 +32,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "residual += hidden; hidden = rmsnorm(residual)");
   m.def("silu_mul", &silu_mul, "silu(gate) * up from packed [T,2F]");
   m.def("rope_qk", &rope_qk, "in-place neox RoPE on Q and K");
+  m.def("rope_qkv_write", &rope_qkv_write,
+        "fused QKV epilogue: in-place RoPE + paged-cache K/V scatter");
   m.def("kv_write", &kv_write, "scatter K/V rows into the paged cache");
   m.def("attn_decode", &attn_decode, "paged GQA decode attention");
   m.def("attn_prefill", &attn_prefill, "causal GQA prefill attention (MFMA)");
+  m.def("attn_prefill_varlen", &attn_prefill_varlen,
+        "packed varlen causal GQA prefill attention (MFMA)");
   m.def("sample", &sample, "Gumbel-max token sampling");
 }

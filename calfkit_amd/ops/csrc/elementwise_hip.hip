@@ -181,6 +181,60 @@ __global__ void kv_write_kernel(__hip_bfloat16* __restrict__ kcache,
   }
 }
 
+// ---------------------------------------------------------- rope_qkv_write
+
+// Fused epilogue of the QKV projection: rotate Q and K heads in place inside
+// the packed [T, (Hq+2*Hkv)*dh] GEMM output, and scatter the K/V heads into
+// the paged cache — one launch replacing {3x contiguous-copy, rope, kv_write}
+// (the .contiguous() copies were 130 ms + 36 ms copyBuffer in the r01
+// profile, profiles/r01_bench_llama8b_kernelstats.md).
+__global__ void rope_qkv_write_kernel(
+    __hip_bfloat16* __restrict__ qkv,        // [T, (hq+2*hkv)*dh]
+    __hip_bfloat16* __restrict__ kcache,     // [P, hkv, page, dh]
+    __hip_bfloat16* __restrict__ vcache,
+    const float* __restrict__ cos_sin,       // [max_pos, dh]
+    const int32_t* __restrict__ positions,   // [T]
+    const int32_t* __restrict__ slots,       // [T]
+    const int hq, const int hkv, const int dh, const int page) {
+  const int64_t token = blockIdx.x;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int nwave = blockDim.x / WAVE;
+  const int half = dh / 2;
+  const int W = (hq + 2 * hkv) * dh;
+  const int pos = positions[token];
+  const int slot = slots[token];
+  const float* cs = cos_sin + (int64_t)pos * dh;
+  __hip_bfloat16* row = qkv + token * (int64_t)W;
+
+  // jobs 0..hq-1: rope Q head; hq..hq+hkv-1: rope K head + cache write;
+  // hq+hkv..hq+2*hkv-1: V cache write
+  for (int job = wid; job < hq + 2 * hkv; job += nwave) {
+    if (job < hq + hkv) {
+      __hip_bfloat16* base = row + job * dh;
+      for (int i = lane; i < half; i += WAVE) {
+        const float c = cs[i], s = cs[half + i];
+        const float x0 = bf2f(base[i]);
+        const float x1 = bf2f(base[half + i]);
+        base[i] = f2bf(x0 * c - x1 * s);
+        base[half + i] = f2bf(x1 * c + x0 * s);
+      }
+    }
+    if (job >= hq && slot >= 0) {
+      const int h = (job - hq) % hkv;
+      const bool is_k = job < hq + hkv;
+      // with dh == 2*WAVE each lane rotated exactly elements (lane, lane+64)
+      // above and copies exactly those below — per-lane program order makes
+      // the rope-then-copy safe with no barrier
+      const int p = slot / page, off = slot % page;
+      const __hip_bfloat16* src = row + (hq + (is_k ? 0 : hkv) + h) * dh;
+      int64_t dst = (((int64_t)p * hkv + h) * page + off) * dh;
+      __hip_bfloat16* d = (is_k ? kcache : vcache) + dst;
+      for (int i = lane; i < dh; i += WAVE) d[i] = src[i];
+    }
+  }
+}
+
 // ------------------------------------------------------------- host glue
 
 #include <torch/extension.h>
@@ -251,6 +305,29 @@ void rope_qk(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
                      (__hip_bfloat16*)k.data_ptr(),
                      (const float*)cos_sin.data_ptr(),
                      (const int32_t*)positions.data_ptr(), hq, hk, dh);
+  HIP_CHECK_LAST();
+}
+
+void rope_qkv_write(torch::Tensor qkv, torch::Tensor kcache,
+                    torch::Tensor vcache, torch::Tensor cos_sin,
+                    torch::Tensor positions, torch::Tensor slots,
+                    int64_t hq, int64_t hkv) {
+  TORCH_CHECK(qkv.is_contiguous() && kcache.is_contiguous());
+  TORCH_CHECK(positions.scalar_type() == torch::kInt32);
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32);
+  const int T = qkv.size(0);
+  const int page = kcache.size(2);
+  const int dh = kcache.size(3);
+  TORCH_CHECK(dh == 2 * WAVE, "rope_qkv_write assumes head_dim == 128");
+  TORCH_CHECK(qkv.size(1) == (hq + 2 * hkv) * dh, "bad qkv width");
+  hipLaunchKernelGGL(rope_qkv_write_kernel, dim3(T), dim3(256), 0, cur_stream(),
+                     (__hip_bfloat16*)qkv.data_ptr(),
+                     (__hip_bfloat16*)kcache.data_ptr(),
+                     (__hip_bfloat16*)vcache.data_ptr(),
+                     (const float*)cos_sin.data_ptr(),
+                     (const int32_t*)positions.data_ptr(),
+                     (const int32_t*)slots.data_ptr(), (int)hq, (int)hkv, dh,
+                     page);
   HIP_CHECK_LAST();
 }
 

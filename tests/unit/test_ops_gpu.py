@@ -196,3 +196,74 @@ def test_sample_topk_masks_tail():
         ops.sample(out, logits.clone(), temp, seed, topk)
         for b in range(B):
             assert int(out[b]) in top[b].tolist()
+
+
+def test_rope_qkv_write_fused_matches_unfused():
+    T, hq, hkv, dh, page = 9, 8, 2, 128, 64
+    W = (hq + 2 * hkv) * dh
+    qkv = torch.randn(T, W, dtype=torch.bfloat16, device="cuda")
+    qkv_ref = qkv.clone()
+    pos = torch.randint(0, 512, (T,), dtype=torch.int32, device="cuda")
+    slots = torch.arange(10, 10 + T, dtype=torch.int32, device="cuda")
+    table = ops.build_rope_table(1024, dh, device="cuda")
+    kc = torch.zeros(4, hkv, page, dh, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros_like(kc)
+    ops.rope_qkv_write(qkv, kc, vc, table, pos, slots, hq, hkv)
+
+    # unfused reference path
+    q = qkv_ref[:, :hq * dh].view(T, hq, dh).contiguous()
+    k = qkv_ref[:, hq * dh:(hq + hkv) * dh].view(T, hkv, dh).contiguous()
+    v = qkv_ref[:, (hq + hkv) * dh:].view(T, hkv, dh).contiguous()
+    ops.rope_qk(q, k, table, pos)
+    kc2 = torch.zeros_like(kc)
+    vc2 = torch.zeros_like(vc)
+    ops.kv_write(kc2, vc2, k, v, slots)
+    assert torch.allclose(qkv[:, :hq * dh].float(),
+                          q.view(T, -1).float(), atol=1e-2, rtol=1e-2)
+    assert torch.allclose(kc.float(), kc2.float(), atol=1e-2, rtol=1e-2)
+    assert torch.equal(vc, vc2)
+
+
+def test_attn_decode_strided_q_matches_contiguous():
+    B, hq, hkv, dh, page, slen = 4, 32, 8, 128, 64, 100
+    W = (hq + 2 * hkv) * dh
+    qkv = torch.randn(B, W, dtype=torch.bfloat16, device="cuda")
+    q_view = qkv[:, :hq * dh].view(B, hq, dh)       # strided rows
+    q_cont = q_view.contiguous()
+    npool = 2 * B + 2
+    kc = torch.randn(npool, hkv, page, dh, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn_like(kc)
+    pt = torch.arange(npool, dtype=torch.int32, device="cuda")[:B * 2].reshape(B, 2)
+    lens = torch.full((B,), slen, dtype=torch.int32, device="cuda")
+    o1 = torch.empty(B, hq, dh, dtype=torch.bfloat16, device="cuda")
+    o2 = torch.empty_like(o1)
+    scale = 1.0 / math.sqrt(dh)
+    ops.attn_decode(o1, q_view, kc, vc, pt, lens, scale)
+    ops.attn_decode(o2, q_cont, kc, vc, pt, lens, scale)
+    assert torch.equal(o1, o2)
+
+
+def test_attn_prefill_varlen_matches_per_sequence():
+    torch.manual_seed(7)
+    hq, hkv, dh = 8, 2, 128
+    lens = [70, 128, 33, 1]
+    total = sum(lens)
+    q = torch.randn(total, hq, dh, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(total, hkv, dh, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(total, hkv, dh, dtype=torch.bfloat16, device="cuda")
+    starts = [0]
+    for n in lens:
+        starts.append(starts[-1] + n)
+    q_starts = torch.as_tensor(starts, dtype=torch.int32, device="cuda")
+    bm = ops.build_prefill_blocks(lens).to("cuda")
+    past = torch.zeros(len(lens), dtype=torch.int32, device="cuda")
+    out = torch.empty_like(q)
+    scale = 1.0 / math.sqrt(dh)
+    ops.attn_prefill_varlen(out, q, k, v, bm, q_starts, q_starts, past, scale)
+    for i, n in enumerate(lens):
+        s, e = starts[i], starts[i + 1]
+        ref = torch.empty(n, hq, dh, dtype=torch.bfloat16, device="cuda")
+        ops.attn_prefill(ref, q[s:e].contiguous(), k[s:e].contiguous(),
+                         v[s:e].contiguous(), scale, 0)
+        assert torch.allclose(out[s:e].float(), ref.float(), atol=2e-2,
+                              rtol=2e-2), f"seq {i}"
