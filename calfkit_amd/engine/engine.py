@@ -83,7 +83,10 @@ class InferenceEngine:
         self.max_batch = max_batch
         self.max_model_len = max_model_len or self.cfg.max_seq
         self.on_gpu = self.device.type == "cuda"
-        self.use_graphs = self.on_gpu if use_graphs is None else use_graphs
+        if use_graphs is None:
+            # MoE's bucketing has a host sync per layer — not capturable yet
+            use_graphs = self.on_gpu and not self.cfg.is_moe
+        self.use_graphs = use_graphs
 
         if kv_pages is None:
             kv_pages = (max_batch * self.max_model_len + PAGE_SIZE - 1) // PAGE_SIZE + 8

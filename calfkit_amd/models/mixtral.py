@@ -1,0 +1,96 @@
+"""Mixtral-8x7B-family MoE model on the CDNA4 op set.
+
+Attention is identical to Llama (GQA, dh=128, paged KV) and is inherited;
+the MLP becomes a top-2 router + 8 SwiGLU experts.
+
+Expert compute (round-1 shape): tokens are bucketed per expert on-device
+(argsort over the flat (token, expert) assignment), each expert runs one
+dense gate-up/down GEMM pair over its bucket (hipBLASLt), and results
+scatter-add back with the router weights.  This is the grouped-GEMM
+decomposition with the group loop on the host; a single fused grouped-GEMM
+kernel is the planned next step (SURVEY §2.4 row "Grouped GEMM (MoE)").
+
+TP: every expert's ffn shards over the TP group exactly like the dense MLP
+(row/column parallel + one all-reduce); the router is replicated.  Expert
+parallelism with RCCL all-to-all is a post-baseline extension for EP>1.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import torch
+
+from .. import ops
+from ..parallel import all_reduce_sum
+from .config import ModelConfig
+from .llama import LayerWeights, LlamaShard
+
+__all__ = ["MixtralShard", "MoeLayerWeights"]
+
+
+@dataclass
+class MoeLayerWeights(LayerWeights):
+    # wgateup/wdown become stacked per-expert: [E, hidden, 2*ffn_s], [E, ffn_s, hidden]
+    router: torch.Tensor = None  # [hidden, E]
+
+
+class MixtralShard(LlamaShard):
+    def __init__(self, cfg: ModelConfig, **kw):
+        assert cfg.is_moe, "MixtralShard needs an MoE config"
+        super().__init__(cfg, **kw)
+        # replace the dense MLP weights with stacked expert weights + router
+        gen = torch.Generator(device="cpu").manual_seed(4321 + self.tp_rank)
+        d, e = cfg.hidden, cfg.num_experts
+
+        def w(*shape: int, fan_in: int) -> torch.Tensor:
+            std = 1.0 / math.sqrt(fan_in)
+            t = torch.randn(*shape, generator=gen, dtype=torch.float32) * std
+            return t.to(device=self.device, dtype=self.dtype)
+
+        layers: list[MoeLayerWeights] = []
+        for lw in self.layers:
+            layers.append(MoeLayerWeights(
+                wqkv=lw.wqkv, wo=lw.wo,
+                wgateup=w(e, d, 2 * self.ffn_s, fan_in=d),
+                wdown=w(e, self.ffn_s, d, fan_in=cfg.ffn),
+                norm_attn=lw.norm_attn, norm_mlp=lw.norm_mlp,
+                router=w(d, e, fan_in=d),
+            ))
+        self.layers = layers
+        self.topk = cfg.experts_per_token
+
+    def _mlp(self, lw: MoeLayerWeights, hidden: torch.Tensor) -> torch.Tensor:
+        T, d = hidden.shape
+        e = self.cfg.num_experts
+        k = self.topk
+        logits = (hidden @ lw.router).float()              # [T, E]
+        weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
+        weights = weights / weights.sum(-1, keepdim=True)   # renormalize top-k
+
+        flat_expert = experts.reshape(-1)                   # [T*k]
+        flat_token = (torch.arange(T, device=hidden.device)
+                      .repeat_interleave(k))                # [T*k]
+        order = torch.argsort(flat_expert, stable=True)
+        counts = torch.bincount(flat_expert, minlength=e)
+
+        out = torch.zeros_like(hidden)
+        gathered = hidden[flat_token[order]]                # [T*k, d] bucketed
+        w_sorted = weights.reshape(-1)[order]
+        start = 0
+        counts_l = counts.tolist()  # host sync; grouped kernel removes this
+        for ei in range(e):
+            n = counts_l[ei]
+            if n == 0:
+                continue
+            xs = gathered[start:start + n]
+            gu = xs @ lw.wgateup[ei]
+            act = torch.empty(n, self.ffn_s, dtype=xs.dtype, device=xs.device)
+            ops.silu_mul(act, gu)
+            ys = act @ lw.wdown[ei]
+            out.index_add_(0, flat_token[order[start:start + n]],
+                           (ys.float() * w_sorted[start:start + n, None])
+                           .to(out.dtype))
+            start += n
+        return all_reduce_sum(out) if self.tp_size > 1 else out
