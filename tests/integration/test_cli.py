@@ -1,0 +1,68 @@
+"""Integration: the ck CLI — node loading, mesh lifecycle, end-to-end run."""
+
+import asyncio
+import subprocess
+import sys
+import textwrap
+
+import pytest
+
+from calfkit_amd.cli.main import _load_nodes
+
+
+def _example(tmp_path):
+    f = tmp_path / "nodes.py"
+    f.write_text(textwrap.dedent("""
+        from calfkit_amd import Agent, agent_tool
+        from calfkit_amd.testing import EchoProvider
+
+        @agent_tool
+        def get_weather(city: str) -> str:
+            \"\"\"Get the weather.\"\"\"
+            return f"sunny in {city}"
+
+        weather_agent = Agent(name="weather_agent", tools=[get_weather],
+                              model_client=EchoProvider())
+        _private = object()
+    """))
+    return f
+
+
+def test_load_nodes_by_name_and_all(tmp_path):
+    f = _example(tmp_path)
+    nodes = _load_nodes(f"{f}:weather_agent")
+    assert len(nodes) == 1 and nodes[0].name == "weather_agent"
+    nodes = _load_nodes(str(f))
+    assert {getattr(n, "name", "?") for n in nodes} == {"weather_agent",
+                                                         "get_weather"}
+
+
+def test_cli_help_runs():
+    out = subprocess.run(
+        [sys.executable, "-m", "calfkit_amd.cli", "dev", "--help"],
+        capture_output=True, text=True, cwd="/root/repo")
+    assert out.returncode == 0
+    for cmd in ("run", "chat", "mesh", "status", "stop", "down"):
+        assert cmd in out.stdout
+
+
+async def test_worker_via_cli_module_against_broker(tmp_path):
+    """End-to-end: broker in-process, worker from the example file, client."""
+    from calfkit_amd.adapters.mesh import MeshBroker, connect_mesh
+    from calfkit_amd.adapters.memory import InProcessThreadStore
+    from calfkit_amd.client import Client
+    from calfkit_amd.worker import Worker
+
+    f = _example(tmp_path)
+    broker = await MeshBroker().start()
+    try:
+        nodes = _load_nodes(str(f))
+        bus, directory = await connect_mesh(broker.address)
+        async with Worker(*nodes, bus=bus, store=InProcessThreadStore(),
+                          directory=directory):
+            bus_c, dir_c = await connect_mesh(broker.address)
+            client = await Client.connect(bus_c, dir_c)
+            r = await client.execute("weather_agent", "hi there", timeout=10)
+            assert r.text == "echo: hi there"
+    finally:
+        await broker.stop()
