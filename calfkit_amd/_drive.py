@@ -141,6 +141,9 @@ async def drive_step(engine: Engine, surface: StepSurface, policy: TurnPolicy,
                         call_id=call.call_id, tool_name=call.tool_name,
                         content=f"unknown tool {call.tool_name!r}", is_error=True))
 
+            # effects publish regardless of how the rest of the step resolves
+            resolution.effects.extend(effects)
+
             synthesized = rejects + unknown
             if synthesized:
                 resolution.new_messages.append(ToolMessage(tuple(synthesized)))
@@ -148,22 +151,13 @@ async def drive_step(engine: Engine, surface: StepSurface, policy: TurnPolicy,
 
             if pending_tools or pending_peers:
                 # D5/D7 — a pending frame ends the hop
-                resolution.effects.extend(effects)
                 resolution.pending_tool_calls = pending_tools
                 resolution.pending_peer_calls = pending_peers
                 resolution.frame_id = uuid.uuid4().hex
                 return resolution
 
-            if synthesized:
-                # every call degenerated to a synthesized return: step again
-                corrections += 1
-                if corrections > policy.max_corrections:
-                    resolution.final = message
-                    return resolution
-                continue
-
-            # only fire-and-forget dispatches: effects out, turn continues
-            resolution.effects.extend(effects)
+            # synthesized returns and/or fire-and-forget dispatches:
+            # the turn continues with another engine step (bounded)
             corrections += 1
             if corrections > policy.max_corrections:
                 resolution.final = message

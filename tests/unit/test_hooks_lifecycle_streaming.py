@@ -1,0 +1,106 @@
+"""Unit: hook kernel, lifecycle brackets, stream chunk vocabulary."""
+
+import pytest
+
+from calfkit_amd.context import TurnContext
+from calfkit_amd.errors import ErrorReport
+from calfkit_amd.hooks import (
+    AgentHook,
+    BeforeStep,
+    Drop,
+    HookSet,
+    OnToolError,
+    parse_hooks,
+)
+from calfkit_amd.lifecycle import Lifecycle, resource
+from calfkit_amd.messages import TextPart, UserMessage
+from calfkit_amd.streaming import CollectingSink, StreamChunk
+
+
+def ctx():
+    return TurnContext(agent_name="a", thread_id="t")
+
+
+async def test_drop_stops_delivery():
+    class Dropper(AgentHook):
+        async def on_delivery(self, ctx, delivery):
+            return Drop()
+
+    hooks = parse_hooks([Dropper()])
+    assert await hooks.run_on_delivery(ctx(), None) is False
+    assert await parse_hooks([]).run_on_delivery(ctx(), None) is True
+
+
+async def test_before_step_merging():
+    class A(AgentHook):
+        async def before_step(self, ctx):
+            return BeforeStep(inject=[UserMessage((TextPart("a"),))],
+                              extra_instructions="one")
+
+    class B(AgentHook):
+        async def before_step(self, ctx):
+            return BeforeStep(extra_instructions="two")
+
+    merged = await parse_hooks([A(), B()]).run_before_step(ctx())
+    assert len(merged.inject) == 1
+    assert merged.extra_instructions == "one\ntwo"
+
+
+async def test_function_hook_and_tool_error_default():
+    calls = []
+
+    def after_step(ctx, message):
+        calls.append(message)
+
+    hooks = parse_hooks([after_step])
+    await hooks.run_after_step(ctx(), UserMessage((TextPart("m"),)))
+    assert len(calls) == 1
+
+    report = ErrorReport.from_exception("n", "tool", ValueError("kapow"))
+    part = await hooks.run_on_tool_error(ctx(), report, "c1", "t")
+    assert part.is_error and "kapow" in str(part.content)
+
+
+async def test_parse_hooks_rejects_junk():
+    with pytest.raises(TypeError):
+        parse_hooks([42])
+
+
+async def test_lifecycle_resource_bracket_lifo():
+    events = []
+
+    @resource
+    async def db():
+        events.append("open-db")
+        yield "DB"
+        events.append("close-db")
+
+    @resource
+    async def cache():
+        events.append("open-cache")
+        yield "CACHE"
+        events.append("close-cache")
+
+    lc = Lifecycle()
+    lc.on_startup.append(lambda: events.append("on_startup"))
+    lc.after_startup.append(lambda: events.append("after_startup"))
+    lc.on_shutdown.append(lambda: events.append("on_shutdown"))
+    lc.after_shutdown.append(lambda: events.append("after_shutdown"))
+    lc.add_resource(db)
+    lc.add_resource(cache)
+    await lc.startup()
+    assert lc.get("db") == "DB" and lc.get("cache") == "CACHE"
+    await lc.shutdown()
+    assert events == ["on_startup", "open-db", "open-cache", "after_startup",
+                      "on_shutdown", "close-cache", "close-db",
+                      "after_shutdown"]
+
+
+def test_stream_chunk_wire_and_sink():
+    c = StreamChunk(event="text_delta", seq=3, text="hi", node="n",
+                    thread_id="t")
+    assert StreamChunk.from_wire(c.to_wire()) == c
+    sink = CollectingSink()
+    sink.emit(c)
+    sink.emit(StreamChunk(event="step_finished"))
+    assert sink.text == "hi"
