@@ -15,6 +15,9 @@ import time
 
 import torch
 
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(
+    __import__("os").path.abspath(__file__))))
+
 
 def main() -> None:
     ap = argparse.ArgumentParser()

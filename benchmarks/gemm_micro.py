@@ -1,0 +1,88 @@
+#!/usr/bin/env python3
+"""Skinny decode GEMM: calfkit skinny_gemm vs hipBLASLt (F.linear) on the
+llama decode shapes.  Prints per-shape timings + effective weight-stream
+bandwidth (the roofline for this regime) as JSON lines."""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import calfkit_amd.ops as ops  # noqa: E402
+
+SHAPES = [  # (name, M, K, N)
+    ("qkv_8b", 32, 4096, 6144),
+    ("o_8b", 32, 4096, 4096),
+    ("gateup_8b", 32, 4096, 28672),
+    ("down_8b", 32, 14336, 4096),
+    ("lmhead_8b", 32, 4096, 128256),
+    ("qkv_70b_tp8", 32, 8192, 1280),
+    ("gateup_70b_tp8", 32, 8192, 7168),
+]
+
+
+def bench(fn, iters=50):
+    for _ in range(5):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6  # us
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--iters", type=int, default=50)
+    args = ap.parse_args()
+    assert torch.cuda.is_available()
+    torch.manual_seed(0)
+    results = []
+    for name, M, K, N in SHAPES:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / K ** 0.5
+        out_lib = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+        out_sk = torch.empty_like(out_lib)
+        sk = ops.choose_splitk(N, K)
+        work = torch.empty(sk * 64 * N, dtype=torch.float32, device="cuda")
+
+        def lib():
+            torch.nn.functional.linear(x, w, out=out_lib)
+
+        def mine():
+            ops.skinny_gemm(out_sk, x, w, work, sk)
+
+        mine()
+        lib()
+        ref = (x.float() @ w.float().T)
+        err = (out_sk.float() - ref).abs().max().item()
+        rel = err / ref.abs().max().item()
+        us_lib = bench(lib, args.iters)
+        us_sk = bench(mine, args.iters)
+        wbytes = N * K * 2
+        row = {
+            "shape": name, "M": M, "K": K, "N": N, "splitk": sk,
+            "us_hipblaslt": round(us_lib, 1), "us_skinny": round(us_sk, 1),
+            "speedup": round(us_lib / us_sk, 2),
+            "TBps_skinny": round(wbytes / us_sk / 1e6, 2),
+            "TBps_hipblaslt": round(wbytes / us_lib / 1e6, 2),
+            "max_rel_err": round(rel, 5),
+        }
+        results.append(row)
+        print(json.dumps(row), flush=True)
+    geo = 1.0
+    for r in results:
+        geo *= r["speedup"]
+    print(json.dumps({"geomean_speedup": round(geo ** (1 / len(results)), 3)}))
+
+
+if __name__ == "__main__":
+    main()

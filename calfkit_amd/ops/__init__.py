@@ -29,6 +29,9 @@ __all__ = [
     "attn_prefill_varlen",
     "build_prefill_blocks",
     "sample",
+    "skinny_gemm",
+    "skinny_gemm_silu",
+    "choose_splitk",
     "build_rope_table",
 ]
 
@@ -262,3 +265,37 @@ def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
         else:
             g = torch.distributions.Gumbel(0.0, 1.0).sample((V,))
             out[b] = int((lg[b] / t + g).argmax())
+
+
+# -------------------------------------------------------------- skinny GEMM
+
+def choose_splitk(N: int, K: int, target_blocks: int = 384) -> int:
+    """Split-K so (N/64)*splitk lands near 1-2x the 256 CUs, K dividing."""
+    best = 1
+    for sk in (1, 2, 4, 8):
+        if K % (64 * sk * 4) != 0 and K % (32 * sk * 4) != 0:
+            continue
+        if (N // 64) * sk <= target_blocks:
+            best = sk
+    return best
+
+
+def skinny_gemm(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                work: torch.Tensor, splitk: int) -> None:
+    """out[M,N] = x[M,K] @ w[N,K]^T for decode batches (M <= 64)."""
+    if _native(x):
+        _K.skinny_gemm(out, x, w, work, splitk)
+        return
+    out.copy_((x.float() @ w.float().T).to(out.dtype))
+
+
+def skinny_gemm_silu(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                     work: torch.Tensor, splitk: int) -> None:
+    """out[M,F] = silu(x@Wg^T) * (x@Wu^T); w = [gate rows | up rows] [2F,K]."""
+    if _native(x):
+        _K.skinny_gemm_silu(out, x, w, work, splitk)
+        return
+    F = out.shape[-1]
+    y = x.float() @ w.float().T
+    g, u = y[:, :F], y[:, F:]
+    out.copy_((torch.nn.functional.silu(g) * u).to(out.dtype))

@@ -25,6 +25,10 @@ void attn_prefill_varlen(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                          torch::Tensor past_lens, double scale);
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor seed_step, torch::Tensor topk);
+void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                 torch::Tensor work, int64_t splitk);
+void skinny_gemm_silu(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor work, int64_t splitk);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
@@ -40,4 +44,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_varlen", &attn_prefill_varlen,
         "packed varlen causal GQA prefill attention (MFMA)");
   m.def("sample", &sample, "Gumbel-max token sampling");
+  m.def("skinny_gemm", &skinny_gemm,
+        "decode projection GEMM: out = x @ w^T, w stored [N,K]");
+  m.def("skinny_gemm_silu", &skinny_gemm_silu,
+        "fused SiLU-MLP gate: out = silu(x@Wg^T) * (x@Wu^T)");
 }

@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "attn_decode.hip"),
     os.path.join(CSRC, "attn_prefill.hip"),
     os.path.join(CSRC, "sampling.hip"),
+    os.path.join(CSRC, "skinny_gemm.hip"),
 ]
 
 setup(

@@ -267,3 +267,36 @@ def test_attn_prefill_varlen_matches_per_sequence():
                          v[s:e].contiguous(), scale, 0)
         assert torch.allclose(out[s:e].float(), ref.float(), atol=2e-2,
                               rtol=2e-2), f"seq {i}"
+
+
+@pytest.mark.parametrize("M,K,N,sk", [
+    (32, 4096, 6144, 4),
+    (1, 4096, 4096, 4),
+    (17, 2048, 1280, 2),
+    (64, 14336, 4096, 4),
+    (33, 4096, 128, 1),
+])
+def test_skinny_gemm_vs_fp32(M, K, N, sk):
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / K ** 0.5)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    work = torch.empty(sk * 64 * N, dtype=torch.float32, device="cuda")
+    ops.skinny_gemm(out, x, w, work, sk)
+    ref = x.float() @ w.float().T
+    scale = ref.abs().max()
+    assert ((out.float() - ref).abs().max() / scale) < 2e-2, \
+        (out.float() - ref).abs().max()
+
+
+def test_skinny_gemm_silu_fused():
+    M, K, F, sk = 32, 4096, 14336, 1
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = (torch.randn(2 * F, K, dtype=torch.bfloat16, device="cuda") / K ** 0.5)
+    out = torch.empty(M, F, dtype=torch.bfloat16, device="cuda")
+    work = torch.empty(sk * 64 * 2 * F, dtype=torch.float32, device="cuda")
+    ops.skinny_gemm_silu(out, x, w, work, sk)
+    y = x.float() @ w.float().T
+    g, u = y[:, :F], y[:, F:]
+    ref = torch.nn.functional.silu(g) * u
+    scale = ref.abs().max()
+    assert ((out.float() - ref).abs().max() / scale) < 2e-2
