@@ -46,10 +46,13 @@ class PrefillPlan:
 
 @dataclass
 class LayerWeights:
-    wqkv: torch.Tensor      # [hidden, (hq_s + 2*hkv_s) * dh]
-    wo: torch.Tensor        # [hq_s * dh, hidden]
-    wgateup: torch.Tensor   # [hidden, 2 * ffn_s]
-    wdown: torch.Tensor     # [ffn_s, hidden]
+    # nn.Linear layout [out, in]: rows are output features, so the skinny
+    # decode GEMM's B-fragments are contiguous 16 B loads (see
+    # ops/csrc/skinny_gemm.hip) and hipBLASLt gets its preferred TN form.
+    wqkv: torch.Tensor      # [(hq_s + 2*hkv_s) * dh, hidden]
+    wo: torch.Tensor        # [hidden, hq_s * dh]
+    wgateup: torch.Tensor   # [2 * ffn_s, hidden]
+    wdown: torch.Tensor     # [hidden, ffn_s]
     norm_attn: torch.Tensor
     norm_mlp: torch.Tensor
 
@@ -86,19 +89,46 @@ class LlamaShard:
         qkv_out = (self.hq + 2 * self.hkv) * self.dh
         for _ in range(cfg.layers):
             self.layers.append(LayerWeights(
-                wqkv=w(d, qkv_out, d),
-                wo=w(self.hq * self.dh, d, cfg.q_dim),
-                wgateup=w(d, 2 * self.ffn_s, d),
-                wdown=w(self.ffn_s, d, cfg.ffn),
+                wqkv=w(qkv_out, d, d),
+                wo=w(d, self.hq * self.dh, cfg.q_dim),
+                wgateup=w(2 * self.ffn_s, d, d),
+                wdown=w(d, self.ffn_s, cfg.ffn),
                 norm_attn=torch.ones(d, device=self.device, dtype=dtype),
                 norm_mlp=torch.ones(d, device=self.device, dtype=dtype),
             ))
+        # fp32 workspace for the skinny decode GEMMs' split-K slabs
+        self._gemm_work = torch.empty(
+            8 * 64 * max(qkv_out, 2 * self.ffn_s, d),
+            dtype=torch.float32, device=self.device) \
+            if self.device.type == "cuda" else torch.empty(0)
         self.norm_final = torch.ones(d, device=self.device, dtype=dtype)
         # lm_head replicated (vocab-parallel sharding is a later optimization;
         # 128256 x 8192 bf16 is ~2 GB of the 288 GB rank)
         self.lm_head = self.embed if cfg.tie_embeddings else w(cfg.vocab_size, d, d)
         self.rope = ops.build_rope_table(cfg.max_seq, self.dh, cfg.rope_theta,
                                          device=self.device)
+        self._splitk_cache: dict[int, int] = {}
+
+    # ------------------------------------------------------- GEMM dispatch
+    def _use_skinny(self, T: int, n_out: int) -> bool:
+        """Decode-regime projections go to the hand-written kernel where it
+        measured faster than hipBLASLt (benchmarks/gemm_micro.py): small-N
+        weight-streaming shapes.  Large-N shapes stay on the library."""
+        return (self.device.type == "cuda" and T <= 64
+                and n_out % 64 == 0 and n_out <= 8192)
+
+    def _proj(self, x: torch.Tensor, weight: torch.Tensor,
+              decode: bool) -> torch.Tensor:
+        n_out = weight.shape[0]
+        if decode and self._use_skinny(x.shape[0], n_out):
+            sk = self._splitk_cache.get(n_out)
+            if sk is None:
+                sk = ops.choose_splitk(n_out, weight.shape[1])
+                self._splitk_cache[n_out] = sk
+            out = torch.empty(x.shape[0], n_out, dtype=x.dtype, device=x.device)
+            ops.skinny_gemm(out, x, weight, self._gemm_work, sk)
+            return out
+        return torch.nn.functional.linear(x, weight)
 
     # ------------------------------------------------------------------ util
     def param_bytes(self) -> int:
@@ -118,7 +148,7 @@ class LlamaShard:
               page_table: torch.Tensor | None,
               seq_lens: torch.Tensor | None) -> torch.Tensor:
         T = hidden.shape[0]
-        qkv = hidden @ lw.wqkv
+        qkv = self._proj(hidden, lw.wqkv, decode=plan is None)
         # fused epilogue: in-place RoPE on the packed q/k heads + paged-cache
         # scatter — q/k/v stay strided views of the GEMM output (no copies)
         ops.rope_qkv_write(qkv, kcache, vcache, self.rope, positions, slots,
@@ -146,15 +176,22 @@ class LlamaShard:
                 plan.past_lens, self.scale)
         else:
             ops.attn_decode(o, q, kcache, vcache, page_table, seq_lens, self.scale)
-        out = o.view(T, self.hq * self.dh) @ lw.wo
+        out = self._proj(o.view(T, self.hq * self.dh), lw.wo,
+                         decode=plan is None)
         return all_reduce_sum(out) if self.tp_size > 1 else out
 
-    def _mlp(self, lw: LayerWeights, hidden: torch.Tensor) -> torch.Tensor:
-        gateup = hidden @ lw.wgateup
-        act = torch.empty(hidden.shape[0], self.ffn_s, dtype=hidden.dtype,
+    def _mlp(self, lw: LayerWeights, hidden: torch.Tensor, *,
+             decode: bool = False) -> torch.Tensor:
+        T = hidden.shape[0]
+        act = torch.empty(T, self.ffn_s, dtype=hidden.dtype,
                           device=hidden.device)
-        ops.silu_mul(act, gateup)
-        down = act @ lw.wdown
+        if decode and self._use_skinny(T, 2 * self.ffn_s):
+            sk = ops.choose_splitk(2 * self.ffn_s, self.cfg.hidden)
+            ops.skinny_gemm_silu(act, hidden, lw.wgateup, self._gemm_work, sk)
+        else:
+            gateup = torch.nn.functional.linear(hidden, lw.wgateup)
+            ops.silu_mul(act, gateup)
+        down = self._proj(act, lw.wdown, decode=decode)
         return all_reduce_sum(down) if self.tp_size > 1 else down
 
     # --------------------------------------------------------------- forward
@@ -208,10 +245,10 @@ class LlamaShard:
                 plan=plan, page_table=page_table, seq_lens=seq_lens)
             ops.fused_add_rmsnorm(hidden, residual, lw.norm_mlp,
                                   self.cfg.norm_eps)
-            hidden = self._mlp(lw, hidden)
+            hidden = self._mlp(lw, hidden, decode=plan is None)
         # final residual add + norm
         ops.fused_add_rmsnorm(hidden, residual, self.norm_final,
                               self.cfg.norm_eps)
         if logits_rows is not None:
             hidden = hidden[logits_rows]
-        return hidden @ self.lm_head.T
+        return torch.nn.functional.linear(hidden, self.lm_head)

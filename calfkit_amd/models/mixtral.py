@@ -53,19 +53,20 @@ class MixtralShard(LlamaShard):
         for lw in self.layers:
             layers.append(MoeLayerWeights(
                 wqkv=lw.wqkv, wo=lw.wo,
-                wgateup=w(e, d, 2 * self.ffn_s, fan_in=d),
-                wdown=w(e, self.ffn_s, d, fan_in=cfg.ffn),
+                wgateup=w(e, 2 * self.ffn_s, d, fan_in=d),
+                wdown=w(e, d, self.ffn_s, fan_in=cfg.ffn),
                 norm_attn=lw.norm_attn, norm_mlp=lw.norm_mlp,
-                router=w(d, e, fan_in=d),
+                router=w(e, d, fan_in=d),
             ))
         self.layers = layers
         self.topk = cfg.experts_per_token
 
-    def _mlp(self, lw: MoeLayerWeights, hidden: torch.Tensor) -> torch.Tensor:
+    def _mlp(self, lw: MoeLayerWeights, hidden: torch.Tensor, *,
+             decode: bool = False) -> torch.Tensor:
         T, d = hidden.shape
         e = self.cfg.num_experts
         k = self.topk
-        logits = (hidden @ lw.router).float()              # [T, E]
+        logits = torch.nn.functional.linear(hidden, lw.router).float()  # [T,E]
         weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
         weights = weights / weights.sum(-1, keepdim=True)   # renormalize top-k
 
@@ -85,10 +86,10 @@ class MixtralShard(LlamaShard):
             if n == 0:
                 continue
             xs = gathered[start:start + n]
-            gu = xs @ lw.wgateup[ei]
+            gu = torch.nn.functional.linear(xs, lw.wgateup[ei])
             act = torch.empty(n, self.ffn_s, dtype=xs.dtype, device=xs.device)
             ops.silu_mul(act, gu)
-            ys = act @ lw.wdown[ei]
+            ys = torch.nn.functional.linear(act, lw.wdown[ei])
             out.index_add_(0, flat_token[order[start:start + n]],
                            (ys.float() * w_sorted[start:start + n, None])
                            .to(out.dtype))

@@ -36,13 +36,18 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 #define SGWAVES 4  // waves per block (in-block K split)
 
 // MT: padded M (16, 32 or 64).  x is [M,K]; rows >= M read row M-1 (the
-// result rows >= M are discarded by the epilogue).
-template <int MT>
+// result rows >= M are discarded by the epilogue).  KU: k-unroll depth per
+// wave iteration (more independent loads in flight).  W loads are
+// NONTEMPORAL: every weight byte is read exactly once per step — keeping it
+// out of L2 leaves the cache to x and the KV pages.
+template <int MT, int KU>
 __global__ __launch_bounds__(SGWAVES * WAVE) void skinny_gemm_kernel(
     float* __restrict__ work,            // [SPLITK, Mp, N] fp32 slabs
+    __hip_bfloat16* __restrict__ out,    // [M, N] direct target (splitk==1)
     const __hip_bfloat16* __restrict__ x, // [M, K]
     const __hip_bfloat16* __restrict__ w, // [N, K]  (W^T of the math)
-    const int M, const int N, const int K, const int splitk) {
+    const int M, const int N, const int K, const int splitk,
+    const int direct) {  // 1: write bf16 straight to out (needs splitk==1)
   constexpr int MTILES = MT / 16;
   const int n0 = blockIdx.x * BN;
   const int sk = blockIdx.y;
@@ -70,24 +75,29 @@ __global__ __launch_bounds__(SGWAVES * WAVE) void skinny_gemm_kernel(
     arow[mi] = min(mi * 16 + l16, M - 1);
   const int wrow = n0 + l16;  // W^T row for this lane's B fragments
 
-  for (int k = k_begin; k < k_end; k += 64) {
-    // two k-steps of 32 unrolled: 4 A-frags + 8 B-frags in flight
+  for (int k = k_begin; k < k_end; k += KU) {
+    // KU/32 k-steps unrolled: all A/B fragment loads of the span issue
+    // before the MFMAs consume them (the compiler schedules the waits)
+    bf16x8v a[KU / 32][MTILES], b[KU / 32][BN / 16];
 #pragma unroll
-    for (int kk = 0; kk < 64; kk += 32) {
-      bf16x8v a[MTILES];
+    for (int kk = 0; kk < KU; kk += 32) {
 #pragma unroll
       for (int mi = 0; mi < MTILES; ++mi)
-        a[mi] = *reinterpret_cast<const bf16x8v*>(
+        a[kk / 32][mi] = *reinterpret_cast<const bf16x8v*>(
             x + (int64_t)arow[mi] * K + k + kk + l4 * 8);
 #pragma unroll
-      for (int ni = 0; ni < BN / 16; ++ni) {
-        const bf16x8v b = *reinterpret_cast<const bf16x8v*>(
-            w + (int64_t)(wrow + ni * 16) * K + k + kk + l4 * 8);
+      for (int ni = 0; ni < BN / 16; ++ni)
+        b[kk / 32][ni] = __builtin_nontemporal_load(
+            reinterpret_cast<const bf16x8v*>(
+                w + (int64_t)(wrow + ni * 16) * K + k + kk + l4 * 8));
+    }
+#pragma unroll
+    for (int kk = 0; kk < KU; kk += 32)
+#pragma unroll
+      for (int ni = 0; ni < BN / 16; ++ni)
 #pragma unroll
         for (int mi = 0; mi < MTILES; ++mi)
-          acc[mi][ni] = MFMA16(a[mi], b, acc[mi][ni]);
-      }
-    }
+          acc[mi][ni] = MFMA16(a[kk / 32][mi], b[kk / 32][ni], acc[mi][ni]);
   }
 
   // ---- cross-wave reduce through LDS, then one fp32 slab write
@@ -101,6 +111,18 @@ __global__ __launch_bounds__(SGWAVES * WAVE) void skinny_gemm_kernel(
         red[wid][mi * 16 + l4 * 4 + r][ni * 16 + l16] = acc[mi][ni][r];
   __syncthreads();
 
+  if (direct) {
+    // single-slice: write bf16 straight to the output, no slab round trip
+    for (int i = threadIdx.x; i < MT * BN; i += SGWAVES * WAVE) {
+      const int r = i / BN, c = i % BN;
+      if (r >= M) continue;
+      float v = red[0][r][c];
+#pragma unroll
+      for (int wv = 1; wv < SGWAVES; ++wv) v += red[wv][r][c];
+      out[(int64_t)r * N + n0 + c] = f2bf(v);
+    }
+    return;
+  }
   // 256 threads cover MT*BN outputs (MT=64: 16 each)
   float* slab = work + ((int64_t)sk * MT + 0) * N + n0;
   for (int i = threadIdx.x; i < MT * BN; i += SGWAVES * WAVE) {
@@ -153,22 +175,34 @@ __global__ void reduce_slabs_silu_kernel(__hip_bfloat16* __restrict__ out,
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
-static void launch_skinny(torch::Tensor work, torch::Tensor x, torch::Tensor w,
-                          int64_t splitk, int Mp, hipStream_t s) {
+static void launch_skinny(torch::Tensor work, torch::Tensor out,
+                          torch::Tensor x, torch::Tensor w,
+                          int64_t splitk, int Mp, hipStream_t s,
+                          int direct) {
   const int M = x.size(0), K = x.size(1), N = w.size(0);
   dim3 grid(N / BN, splitk), block(SGWAVES * WAVE);
-#define L(MT)                                                              \
-  hipLaunchKernelGGL(skinny_gemm_kernel<MT>, grid, block, 0, s,            \
+  const int wave_k = (K / splitk) / SGWAVES;
+  const int ku = (wave_k % 128 == 0) ? 128 : ((wave_k % 64 == 0) ? 64 : 32);
+#define L(MT, KU)                                                          \
+  hipLaunchKernelGGL((skinny_gemm_kernel<MT, KU>), grid, block, 0, s,      \
                      (float*)work.data_ptr(),                              \
+                     (__hip_bfloat16*)out.data_ptr(),                      \
                      (const __hip_bfloat16*)x.data_ptr(),                  \
                      (const __hip_bfloat16*)w.data_ptr(), M, N, K,         \
-                     (int)splitk)
+                     (int)splitk, direct)
+#define LM(MT)                                                             \
+  do {                                                                     \
+    if (ku == 128) L(MT, 128);                                             \
+    else if (ku == 64) L(MT, 64);                                          \
+    else L(MT, 32);                                                        \
+  } while (0)
   switch (Mp) {
-    case 16: L(16); break;
-    case 32: L(32); break;
-    case 64: L(64); break;
+    case 16: LM(16); break;
+    case 32: LM(32); break;
+    case 64: LM(64); break;
     default: TORCH_CHECK(false, "skinny_gemm: M must pad to 16/32/64, got ", Mp);
   }
+#undef LM
 #undef L
 }
 
@@ -185,14 +219,17 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(N % BN == 0, "N must be a multiple of 64");
   TORCH_CHECK(K % (64 * splitk * SGWAVES) == 0 || K % (32 * splitk * SGWAVES) == 0,
               "K must split evenly: K=", K, " splitk=", splitk);
-  TORCH_CHECK(work.numel() >= splitk * (int64_t)Mp * N, "workspace too small");
+  TORCH_CHECK(splitk == 1 || work.numel() >= splitk * (int64_t)Mp * N,
+              "workspace too small");
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
-  launch_skinny(work, x, w, splitk, Mp, s);
-  const int64_t total = (int64_t)M * N;
-  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
-  hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks), dim3(256), 0, s,
-                     (__hip_bfloat16*)out.data_ptr(),
-                     (const float*)work.data_ptr(), M, N, Mp, (int)splitk);
+  launch_skinny(work, out, x, w, splitk, Mp, s, splitk == 1 ? 1 : 0);
+  if (splitk > 1) {
+    const int64_t total = (int64_t)M * N;
+    const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+    hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks), dim3(256), 0, s,
+                       (__hip_bfloat16*)out.data_ptr(),
+                       (const float*)work.data_ptr(), M, N, Mp, (int)splitk);
+  }
   HIP_CHECK_LAST();
 }
 
@@ -207,7 +244,8 @@ void skinny_gemm_silu(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(out.size(1) == F);
   TORCH_CHECK(work.numel() >= splitk * (int64_t)Mp * N, "workspace too small");
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
-  launch_skinny(work, x, w, splitk, Mp, s);
+  // the silu epilogue mixes gate and up columns, so it always reads slabs
+  launch_skinny(work, out, x, w, splitk, Mp, s, /*direct=*/0);
   const int64_t total = (int64_t)M * F;
   const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(reduce_slabs_silu_kernel, dim3(blocks), dim3(256), 0, s,

@@ -21,17 +21,17 @@ def test_moe_mlp_matches_dense_reference():
     assert out.shape == x.shape
 
     # reference: dense per-token loop over its top-2 experts
-    logits = (x @ lw.router).float()
+    logits = torch.nn.functional.linear(x, lw.router).float()
     w, e = torch.topk(torch.softmax(logits, -1), 2, dim=-1)
     w = w / w.sum(-1, keepdim=True)
     ref = torch.zeros_like(x, dtype=torch.float32)
     for t in range(x.shape[0]):
         for j in range(2):
             ei = int(e[t, j])
-            gu = x[t:t + 1] @ lw.wgateup[ei]
+            gu = torch.nn.functional.linear(x[t:t + 1], lw.wgateup[ei])
             f = gu.shape[-1] // 2
             act = torch.nn.functional.silu(gu[:, :f].float()) * gu[:, f:].float()
-            ys = act.to(x.dtype) @ lw.wdown[ei]
+            ys = torch.nn.functional.linear(act.to(x.dtype), lw.wdown[ei])
             ref[t] += float(w[t, j]) * ys[0].float()
     assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2), \
         (out.float() - ref).abs().max()

@@ -27,23 +27,27 @@ def _shard_from_full(full_sd, cfg, rank, world):
     shard.lm_head.copy_(full_sd["lm_head"])
     shard.norm_final.copy_(full_sd["norm_final"])
     for i, lw in enumerate(shard.layers):
-        wqkv = full_sd[f"l{i}.wqkv"]  # [d, (hq + 2hkv)*dh] full
+        # weights are [out, in]: column-parallel = ROW slices of wqkv/wgateup,
+        # row-parallel = COLUMN slices of wo/wdown
+        wqkv = full_sd[f"l{i}.wqkv"]  # [(hq + 2hkv)*dh, d] full
         hq, hkv = cfg.heads, cfg.kv_heads
-        qf = wqkv[:, : hq * dh]
-        kf = wqkv[:, hq * dh:(hq + hkv) * dh]
-        vf = wqkv[:, (hq + hkv) * dh:]
+        qf = wqkv[: hq * dh]
+        kf = wqkv[hq * dh:(hq + hkv) * dh]
+        vf = wqkv[(hq + hkv) * dh:]
         lw.wqkv.copy_(torch.cat([
-            qf[:, rank * hq_s * dh:(rank + 1) * hq_s * dh],
-            kf[:, rank * hkv_s * dh:(rank + 1) * hkv_s * dh],
-            vf[:, rank * hkv_s * dh:(rank + 1) * hkv_s * dh]], dim=1))
-        lw.wo.copy_(full_sd[f"l{i}.wo"][rank * hq_s * dh:(rank + 1) * hq_s * dh])
-        wg = full_sd[f"l{i}.wgateup"]  # [d, 2*ffn] full: [gate | up]
-        gate = wg[:, : cfg.ffn]
-        up = wg[:, cfg.ffn:]
+            qf[rank * hq_s * dh:(rank + 1) * hq_s * dh],
+            kf[rank * hkv_s * dh:(rank + 1) * hkv_s * dh],
+            vf[rank * hkv_s * dh:(rank + 1) * hkv_s * dh]], dim=0))
+        lw.wo.copy_(
+            full_sd[f"l{i}.wo"][:, rank * hq_s * dh:(rank + 1) * hq_s * dh])
+        wg = full_sd[f"l{i}.wgateup"]  # [2*ffn, d] full: [gate rows | up rows]
+        gate = wg[: cfg.ffn]
+        up = wg[cfg.ffn:]
         lw.wgateup.copy_(torch.cat([
-            gate[:, rank * ffn_s:(rank + 1) * ffn_s],
-            up[:, rank * ffn_s:(rank + 1) * ffn_s]], dim=1))
-        lw.wdown.copy_(full_sd[f"l{i}.wdown"][rank * ffn_s:(rank + 1) * ffn_s])
+            gate[rank * ffn_s:(rank + 1) * ffn_s],
+            up[rank * ffn_s:(rank + 1) * ffn_s]], dim=0))
+        lw.wdown.copy_(
+            full_sd[f"l{i}.wdown"][:, rank * ffn_s:(rank + 1) * ffn_s])
         lw.norm_attn.copy_(full_sd[f"l{i}.norm_attn"])
         lw.norm_mlp.copy_(full_sd[f"l{i}.norm_mlp"])
     return shard
@@ -115,7 +119,7 @@ def test_tp2_matches_tp1():
              for r in range(2)]
     for p in procs:
         p.start()
-    tp_logits = q.get(timeout=300)
+    tp_logits = q.get(timeout=120)
     for p in procs:
         p.join(timeout=120)
         assert p.exitcode == 0
