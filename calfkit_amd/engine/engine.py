@@ -8,12 +8,15 @@ Threading model: the engine owns a dedicated scheduler thread; the asyncio
 agent plane submits requests through a thread-safe queue and gets
 ``concurrent.futures.Future``s back (bridged by LocalModelClient).
 
-Decode step (the hot loop):
-  tokens/positions/slots/page_table/seq_lens live in STATIC device tensors;
-  each step writes the small host-side batch state into them, then either
-  replays the hipGraph captured for the padded batch size or runs eager.
-  Sampling (Gumbel-max) runs inside the graph; the seed/step counter is a
-  device tensor bumped by a tiny copy before each replay.
+Decode runs (the hot loop):
+  decode is data-independent except for the sampled token VALUES — slots,
+  positions and sequence lengths all advance deterministically.  So the
+  engine executes M consecutive decode steps per host visit: one hipGraph
+  per (padded batch, M) contains M chained {forward -> sample -> feed the
+  sampled ids back in on-device} steps over step-indexed static buffers,
+  and the host syncs ONCE per run to read the M x B sampled ids.  Sequences
+  that pass max_new/EOS mid-run overshoot by < M tokens; the extras are
+  discarded and their KV pages truncated.
 """
 
 from __future__ import annotations
@@ -38,6 +41,8 @@ __all__ = ["GenRequest", "GenResult", "InferenceEngine"]
 
 PREFILL_CHUNK = 512
 PREFILL_BUDGET = 8192
+#: decode steps executed per host visit (graphs exist for each power of 2)
+RUN_STEPS = (8, 4, 2, 1)
 
 
 @dataclass
@@ -89,7 +94,10 @@ class InferenceEngine:
         self.use_graphs = use_graphs
 
         if kv_pages is None:
-            kv_pages = (max_batch * self.max_model_len + PAGE_SIZE - 1) // PAGE_SIZE + 8
+            # + max_batch: headroom for multi-step run overshoot (< RUN_STEPS
+            # extra slots per sequence) and the padding dummy page
+            kv_pages = ((max_batch * self.max_model_len + PAGE_SIZE - 1)
+                        // PAGE_SIZE + 8 + max_batch)
         self.kv = KVCache(layers=self.cfg.layers, kv_heads=model.hkv,
                           head_dim=model.dh, num_pages=kv_pages,
                           device=self.device)
@@ -103,20 +111,22 @@ class InferenceEngine:
                          dtype=model.dtype, device=self.device))
             for _ in range(self.cfg.layers)]
 
-        # ---- static decode buffers (graph I/O)
+        # ---- static decode buffers (graph I/O); step-indexed for the
+        # multi-step runs (MS = max run length)
         B = max_batch
+        MS = RUN_STEPS[0]
         dev = self.device
         self._tokens = torch.zeros(B, dtype=torch.int64, device=dev)
-        self._positions = torch.zeros(B, dtype=torch.int32, device=dev)
-        self._slots = torch.zeros(B, dtype=torch.int32, device=dev)
+        self._positions = torch.zeros(MS, B, dtype=torch.int32, device=dev)
+        self._slots = torch.zeros(MS, B, dtype=torch.int32, device=dev)
         self._page_table = torch.zeros(B, self.max_pages, dtype=torch.int32,
                                        device=dev)
-        self._seq_lens = torch.ones(B, dtype=torch.int32, device=dev)
+        self._seq_lens = torch.ones(MS, B, dtype=torch.int32, device=dev)
         self._temperature = torch.zeros(B, dtype=torch.float32, device=dev)
         self._topk = torch.zeros(B, dtype=torch.int32, device=dev)
-        self._seed_step = torch.zeros(2, dtype=torch.int64, device=dev)
-        self._sampled = torch.zeros(B, dtype=torch.int32, device=dev)
-        self._graphs: dict[int, torch.cuda.CUDAGraph] = {}
+        self._seeds = torch.zeros(MS, 2, dtype=torch.int64, device=dev)
+        self._sampled = torch.zeros(MS, B, dtype=torch.int32, device=dev)
+        self._graphs: dict[tuple[int, int], torch.cuda.CUDAGraph] = {}
         #: one page sacrificed as the write target of padding rows
         self._dummy_page = self.kv.num_pages - 1
         self.kv._free.remove(self._dummy_page)
@@ -347,33 +357,46 @@ class InferenceEngine:
                 return g
         return self.max_batch
 
-    def _write_batch_state(self, pad: int) -> None:
+    def _run_len(self) -> int:
+        """Steps per host visit: bounded by the shortest remaining budget
+        (a finishing sequence overshoots < M tokens, which is cheaper than
+        shrinking every run to the laggard, but never run an entire batch
+        past its collective end)."""
+        rem = max(s.request.max_new_tokens - s.new_tokens
+                  for s in self._running)
+        for m in RUN_STEPS:
+            if m <= rem:
+                return m
+        return 1
+
+    def _write_run_state(self, pad: int, M: int) -> None:
         B = len(self._running)
         tokens = [s.last_token for s in self._running] + [0] * (pad - B)
-        positions, slots, lens = [], [], []
+        slots = [[0] * pad for _ in range(M)]
+        positions = [[0] * pad for _ in range(M)]
+        lens = [[1] * pad for _ in range(M)]
         pt_rows = []
-        for s in self._running:
-            slot = self.kv.slots_for(s.seq_id, 1)[0]
-            slots.append(slot)
-            ln = self.kv.length(s.seq_id)  # includes the new token
-            lens.append(ln)
-            positions.append(ln - 1)
+        for b, s in enumerate(self._running):
+            base = self.kv.length(s.seq_id)
+            run_slots = self.kv.slots_for(s.seq_id, M)
+            for i in range(M):
+                slots[i][b] = run_slots[i]
+                positions[i][b] = base + i
+                lens[i][b] = base + i + 1
             pages = self.kv._slots[s.seq_id].pages
             pt_rows.append(pages + [0] * (self.max_pages - len(pages)))
-        for _ in range(pad - B):
-            slots.append(self._dummy_page * PAGE_SIZE)
-            lens.append(1)
-            positions.append(0)
+        for b in range(B, pad):
+            for i in range(M):
+                slots[i][b] = self._dummy_page * PAGE_SIZE + i
             pt_rows.append([self._dummy_page] + [0] * (self.max_pages - 1))
-        dev = self.device
         self._tokens[:pad].copy_(torch.as_tensor(tokens, dtype=torch.int64),
                                  non_blocking=True)
-        self._positions[:pad].copy_(
+        self._positions[:M, :pad].copy_(
             torch.as_tensor(positions, dtype=torch.int32), non_blocking=True)
-        self._slots[:pad].copy_(torch.as_tensor(slots, dtype=torch.int32),
-                                non_blocking=True)
-        self._seq_lens[:pad].copy_(torch.as_tensor(lens, dtype=torch.int32),
-                                   non_blocking=True)
+        self._slots[:M, :pad].copy_(torch.as_tensor(slots, dtype=torch.int32),
+                                    non_blocking=True)
+        self._seq_lens[:M, :pad].copy_(
+            torch.as_tensor(lens, dtype=torch.int32), non_blocking=True)
         self._page_table[:pad].copy_(
             torch.as_tensor(pt_rows, dtype=torch.int32), non_blocking=True)
         temps = [s.request.temperature for s in self._running] + [0.0] * (pad - B)
@@ -381,34 +404,40 @@ class InferenceEngine:
         self._temperature[:pad].copy_(torch.as_tensor(temps), non_blocking=True)
         self._topk[:pad].copy_(torch.as_tensor(topks, dtype=torch.int32),
                                non_blocking=True)
-        self._step_counter += 1
         base_seed = self._running[0].request.seed or 12345
-        self._seed_step.copy_(torch.as_tensor(
-            [base_seed, self._step_counter], dtype=torch.int64),
-            non_blocking=True)
+        seeds = [[base_seed, self._step_counter + i + 1] for i in range(M)]
+        self._seeds[:M].copy_(torch.as_tensor(seeds, dtype=torch.int64),
+                              non_blocking=True)
+        self._step_counter += M
 
-    def _decode_batch_gpu(self, pad: int) -> None:
-        """The captured region: one decode forward + sampling, batch ``pad``."""
-        model = self.model
-        logits = model.forward(
-            self._tokens[:pad], self._positions[:pad], self.kv.kcache,
-            self.kv.vcache, self._slots[:pad],
-            page_table=self._page_table[:pad], seq_lens=self._seq_lens[:pad])
+    def _decode_run_gpu(self, pad: int, M: int) -> None:
+        """The captured region: M chained {forward -> sample} steps."""
         from .. import ops
-        ops.sample(self._sampled[:pad], logits.float(), self._temperature[:pad],
-                   self._seed_step, self._topk[:pad])
+
+        model = self.model
+        tokens = self._tokens[:pad]
+        for i in range(M):
+            logits = model.forward(
+                tokens, self._positions[i, :pad], self.kv.kcache,
+                self.kv.vcache, self._slots[i, :pad],
+                page_table=self._page_table[:pad],
+                seq_lens=self._seq_lens[i, :pad])
+            ops.sample(self._sampled[i, :pad], logits.float(),
+                       self._temperature[:pad], self._seeds[i],
+                       self._topk[:pad])
+            tokens = self._sampled[i, :pad].long()  # on-device feedback
 
     def _capture_graphs(self) -> None:
-        """Capture one hipGraph per padded batch size, largest first."""
+        """One hipGraph per (padded batch, run length), largest first."""
         torch.cuda.synchronize(self.device)
         for b in reversed([g for g in self.GRAPH_BATCHES if g <= self.max_batch]):
-            # warm up allocations outside capture
-            self._decode_batch_gpu(b)
-            torch.cuda.synchronize(self.device)
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self._decode_batch_gpu(b)
-            self._graphs[b] = g
+            for m in RUN_STEPS:
+                self._decode_run_gpu(b, m)  # warm up allocations
+                torch.cuda.synchronize(self.device)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._decode_run_gpu(b, m)
+                self._graphs[(b, m)] = g
         torch.cuda.synchronize(self.device)
         _log.info("captured %d decode graphs: %s", len(self._graphs),
                   sorted(self._graphs))
@@ -417,22 +446,34 @@ class InferenceEngine:
         t0 = time.perf_counter()
         B = len(self._running)
         pad = self._pad_batch(B)
-        self._write_batch_state(pad)
-        if self.use_graphs and pad in self._graphs:
-            self._graphs[pad].replay()
+        M = self._run_len()
+        self._write_run_state(pad, M)
+        if self.use_graphs and (pad, M) in self._graphs:
+            self._graphs[(pad, M)].replay()
             self._stats["graph_replays"] += 1
         else:
-            self._decode_batch_gpu(pad)
-        sampled = self._sampled[:B].cpu().tolist()
+            self._decode_run_gpu(pad, M)
+        sampled = self._sampled[:M, :B].cpu()  # the ONE sync per run
         dt = (time.perf_counter() - t0) * 1e3
-        self._stats["decode_steps"] += 1
-        self._stats["decode_tokens"] += B
+        self._stats["decode_steps"] += M
+        self._stats["decode_tokens"] += M * B
         self._stats["decode_ms"] += dt
         still: list[_Seq] = []
-        for seq, tok in zip(self._running, sampled):
+        for b, seq in enumerate(self._running):
             seq.result.decode_ms += dt
-            self._emit_token(seq, int(tok))
-            if not self._finish_if_done(seq):
+            done = False
+            for i in range(M):
+                self._emit_token(seq, int(sampled[i, b]))
+                done = self._check_done(seq)
+                if done:
+                    overshoot = M - 1 - i
+                    if overshoot > 0:  # drop the blind-run extras
+                        self.kv.truncate(seq.seq_id,
+                                         self.kv.length(seq.seq_id) - overshoot)
+                    break
+            if done:
+                self._finish(seq)
+            else:
                 still.append(seq)
         self._running = still
 
@@ -448,13 +489,24 @@ class InferenceEngine:
             except Exception:
                 _log.exception("on_token callback failed")
 
-    def _finish_if_done(self, seq: _Seq) -> bool:
+    def _check_done(self, seq: _Seq) -> bool:
         r = seq.request
         eos_hit = (not r.ignore_eos) and seq.last_token == 2
-        if seq.new_tokens >= r.max_new_tokens or eos_hit:
-            seq.result.finish_reason = "stop" if eos_hit else "length"
-            self.kv.release(seq.seq_id)
-            if not seq.future.done():
-                seq.future.set_result(seq.result)
+        if eos_hit:
+            seq.result.finish_reason = "stop"
+            return True
+        if seq.new_tokens >= r.max_new_tokens:
+            seq.result.finish_reason = "length"
+            return True
+        return False
+
+    def _finish(self, seq: _Seq) -> None:
+        self.kv.release(seq.seq_id)
+        if not seq.future.done():
+            seq.future.set_result(seq.result)
+
+    def _finish_if_done(self, seq: _Seq) -> bool:
+        if self._check_done(seq):
+            self._finish(seq)
             return True
         return False
