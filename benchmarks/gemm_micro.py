@@ -89,10 +89,6 @@ def main() -> None:
     print(json.dumps({"geomean_speedup": round(geo ** (1 / len(results)), 3)}))
 
 
-if __name__ == "__main__":
-    main()
-
-
 def sweep(args) -> None:
     """Within-probe variant sweep (the guide's A/B rule): every variant on
     every shape in ONE process, interleaved rounds."""
@@ -131,3 +127,7 @@ def sweep(args) -> None:
                           "us_hipblaslt": round(lib_us, 1),
                           "variants": rows, "best": best[0],
                           "best_us": round(best[1], 1)}), flush=True)
+
+
+if __name__ == "__main__":
+    main()
