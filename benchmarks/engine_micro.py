@@ -66,7 +66,7 @@ def main() -> None:
         # raw replay rate: GPU-only cost of one decode step (no host logic)
         raw_ms = None
         pad = engine._pad_batch(args.batch)
-        g = engine._graphs.get(pad)
+        g = engine._graphs.get((pad, 8)) or engine._graphs.get(pad)
         if g is not None:
             torch.cuda.synchronize()
             t0 = time.perf_counter()

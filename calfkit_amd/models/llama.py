@@ -110,17 +110,19 @@ class LlamaShard:
         self._splitk_cache: dict[int, int] = {}
 
     # ------------------------------------------------------- GEMM dispatch
-    def _use_skinny(self, T: int, n_out: int) -> bool:
+    def _use_skinny(self, T: int, n_out: int, k_in: int) -> bool:
         """Decode-regime projections go to the hand-written kernel where it
-        measured faster than hipBLASLt (benchmarks/gemm_micro.py): small-N
-        weight-streaming shapes.  Large-N shapes stay on the library."""
+        measured faster than hipBLASLt (profiles/r01_gemm_sweep.md): the
+        win region is N*K <= ~40M weight elements (qkv/o and all TP-sharded
+        70B projections); the giant tiles (gate-up, down, lm_head) stay on
+        the library until the glds-staged variant lands."""
         return (self.device.type == "cuda" and T <= 64
-                and n_out % 64 == 0 and n_out <= 8192)
+                and n_out % 64 == 0 and n_out * k_in <= 42_000_000)
 
     def _proj(self, x: torch.Tensor, weight: torch.Tensor,
               decode: bool) -> torch.Tensor:
         n_out = weight.shape[0]
-        if decode and self._use_skinny(x.shape[0], n_out):
+        if decode and self._use_skinny(x.shape[0], n_out, weight.shape[1]):
             sk = self._splitk_cache.get(n_out)
             if sk is None:
                 sk = ops.choose_splitk(n_out, weight.shape[1])
@@ -185,7 +187,7 @@ class LlamaShard:
         T = hidden.shape[0]
         act = torch.empty(T, self.ffn_s, dtype=hidden.dtype,
                           device=hidden.device)
-        if decode and self._use_skinny(T, 2 * self.ffn_s):
+        if decode and self._use_skinny(T, 2 * self.ffn_s, self.cfg.hidden):
             sk = ops.choose_splitk(2 * self.ffn_s, self.cfg.hidden)
             ops.skinny_gemm_silu(act, hidden, lw.wgateup, self._gemm_work, sk)
         else:
