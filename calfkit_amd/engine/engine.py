@@ -335,7 +335,7 @@ class InferenceEngine:
         seed = torch.as_tensor(
             [seqs[0].request.seed or seqs[0].seq_id, self._step_counter],
             dtype=torch.int64).to(logits.device)
-        ops.sample(out, logits.float(), temp, seed,
+        ops.sample(out, logits, temp, seed,
                    topk if any(s.request.top_k for s in seqs) else None)
         return [int(t) for t in out.cpu()]
 
@@ -347,7 +347,7 @@ class InferenceEngine:
                                dtype=torch.int64, device=logits.device)
         topk = torch.full((1,), r.top_k, dtype=torch.int32, device=logits.device)
         from .. import ops
-        ops.sample(out, logits.float(), temp, seed, topk if r.top_k else None)
+        ops.sample(out, logits, temp, seed, topk if r.top_k else None)
         return int(out.item())
 
     # ---------------------------------------------------------------- decode
@@ -422,7 +422,7 @@ class InferenceEngine:
                 self.kv.vcache, self._slots[i, :pad],
                 page_table=self._page_table[:pad],
                 seq_lens=self._seq_lens[i, :pad])
-            ops.sample(self._sampled[i, :pad], logits.float(),
+            ops.sample(self._sampled[i, :pad], logits,
                        self._temperature[:pad], self._seeds[i],
                        self._topk[:pad])
             tokens = self._sampled[i, :pad].long()  # on-device feedback
