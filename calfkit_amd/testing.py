@@ -25,7 +25,7 @@ from .schemas import EngineRequest
 from .streaming import ChunkSink, StreamChunk
 
 __all__ = ["FakeProvider", "FunctionModel", "EchoProvider", "FakeMCPSession",
-           "reply", "call_tool"]
+           "ToolCallDirector", "reply", "call_tool"]
 
 
 def reply(text: str) -> AssistantMessage:
@@ -151,3 +151,33 @@ class FakeMCPSession:
 
     async def close(self) -> None:
         self.closed = True
+
+
+class ToolCallDirector:
+    """Wraps a real provider, directing its tool-call behavior.
+
+    The inner provider (e.g. the GPU LocalModelClient) still does ALL its
+    generation work; the director then overrides the *decision* — on the
+    first step of a thread it swaps the text for the scripted tool calls,
+    and after the returns arrive it lets the provider's text stand.  This
+    keeps benches honest (full engine work per step) while random-init
+    weights cannot emit real calls (BASELINE config 3's shape).
+    """
+
+    def __init__(self, inner: Any, calls: Sequence[tuple[str, dict]] = (),
+                 rounds: int = 1):
+        self.inner = inner
+        self.calls = list(calls)
+        self.rounds = rounds
+
+    async def complete(self, request: Any, sink: Any) -> AssistantMessage:
+        message = await self.inner.complete(request, sink)
+        # count this thread's prior assistant steps to find our position
+        steps = sum(1 for m in request.messages if m.role == "assistant")
+        if steps < self.rounds and self.calls:
+            parts = tuple(
+                ToolCallPart(call_id=uuid.uuid4().hex, tool_name=name,
+                             arguments=dict(args))
+                for name, args in self.calls)
+            return AssistantMessage(parts, facts=message.facts)
+        return message
