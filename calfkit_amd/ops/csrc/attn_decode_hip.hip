@@ -67,6 +67,10 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
 
   const int64_t tbl = (int64_t)seq * max_pages;
   for (int pi = wid; pi < npages; pi += NWAVES) {
+    // bar the compiler from hoisting the per-page q_lds re-reads out of
+    // this loop: without it LICM keeps G*128 floats live across pages
+    // (251-747 VGPR spills measured; see the r01/r03 kernel profiles)
+    asm volatile("" ::: "memory");
     const int page = page_table[tbl + pi];
     const int64_t base = (((int64_t)page * hkv + kvh) * KVPAGE) * HD;
     const int token = pi * KVPAGE + lane;
@@ -75,19 +79,33 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
     float s[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) s[g] = 0.f;
-    if (live) {
-      const bf16x8* krow = reinterpret_cast<const bf16x8*>(kcache + base + lane * HD);
+    {
+      // all 16 row loads issue together (one memory round trip), then the
+      // FMA sweep consumes them; q re-read from LDS as float4 broadcasts
+      // (cheap; every lane reads the same address). Masked lanes clamp to
+      // row 0 of the page (valid memory) and overwrite s below.
+      const bf16x8* krow = reinterpret_cast<const bf16x8*>(
+          kcache + base + (live ? lane : 0) * HD);
+      bf16x8 kreg[HD / 8];
+#pragma unroll
+      for (int i = 0; i < HD / 8; ++i) kreg[i] = krow[i];
 #pragma unroll
       for (int i = 0; i < HD / 8; ++i) {
-        bf16x8 kv8 = krow[i];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float kf = bf2f(kv8.v[j]);
+        for (int jj = 0; jj < 2; ++jj) {
 #pragma unroll
-          for (int g = 0; g < G; ++g) s[g] = fmaf(q_lds[g][i * 8 + j], kf, s[g]);
+          for (int g = 0; g < G; ++g) {
+            const float4 qv =
+                *reinterpret_cast<const float4*>(&q_lds[g][i * 8 + jj * 4]);
+            s[g] = fmaf(qv.x, bf2f(kreg[i].v[jj * 4 + 0]), s[g]);
+            s[g] = fmaf(qv.y, bf2f(kreg[i].v[jj * 4 + 1]), s[g]);
+            s[g] = fmaf(qv.z, bf2f(kreg[i].v[jj * 4 + 2]), s[g]);
+            s[g] = fmaf(qv.w, bf2f(kreg[i].v[jj * 4 + 3]), s[g]);
+          }
         }
       }
-    } else {
+    }
+    if (!live) {
 #pragma unroll
       for (int g = 0; g < G; ++g) s[g] = -INFINITY;
     }
@@ -113,26 +131,39 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
     // (profiles/r01_bench_llama8b_kernelstats.md).
     const int t0 = thalf * 32;
     const int nt = min(32, slen - pi * KVPAGE - t0);
-    const float2* vbase = reinterpret_cast<const float2*>(vcache + base);
+    const float2* vbase = reinterpret_cast<const float2*>(vcache + base) + dlane;
     if (nt > 0) {
-      float2 vbuf[32];
-#pragma unroll
-      for (int t = 0; t < 32; ++t)
-        vbuf[t] = vbase[(t0 + min(t, nt - 1)) * (HD / 4) + dlane];
-#pragma unroll
-      for (int t = 0; t < 32; ++t) {
-        if (t >= nt) break;
-        const __hip_bfloat16* vb =
-            reinterpret_cast<const __hip_bfloat16*>(&vbuf[t]);
-#pragma unroll
-        for (int g = 0; g < G; ++g) {
-          const float pg = __shfl(p[g], t0 + t, WAVE);
-          acc[g][0] = fmaf(pg, bf2f(vb[0]), acc[g][0]);
-          acc[g][1] = fmaf(pg, bf2f(vb[1]), acc[g][1]);
-          acc[g][2] = fmaf(pg, bf2f(vb[2]), acc[g][2]);
-          acc[g][3] = fmaf(pg, bf2f(vb[3]), acc[g][3]);
-        }
-      }
+      // 2x8-row double buffer, macro-expanded so every array index is a
+      // compile-time constant (pointer selection would demote to scratch)
+      float2 vA[8], vB[8];
+#define PV_LOAD(BUF, C)                                                     \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j)                             \
+      BUF[j] = vbase[(t0 + min((C) * 8 + j, nt - 1)) * (HD / 4)];
+#define PV_FMA(BUF, C)                                                      \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
+    const int t = (C) * 8 + j;                                              \
+    if (t < nt) {                                                           \
+      const __hip_bfloat16* vb =                                            \
+          reinterpret_cast<const __hip_bfloat16*>(&BUF[j]);                 \
+      _Pragma("unroll") for (int g = 0; g < G; ++g) {                       \
+        const float pg = __shfl(p[g], t0 + t, WAVE);                        \
+        acc[g][0] = fmaf(pg, bf2f(vb[0]), acc[g][0]);                       \
+        acc[g][1] = fmaf(pg, bf2f(vb[1]), acc[g][1]);                       \
+        acc[g][2] = fmaf(pg, bf2f(vb[2]), acc[g][2]);                       \
+        acc[g][3] = fmaf(pg, bf2f(vb[3]), acc[g][3]);                       \
+      }                                                                     \
+    }                                                                       \
+  }
+      PV_LOAD(vA, 0)
+      PV_LOAD(vB, 1)
+      PV_FMA(vA, 0)
+      PV_LOAD(vA, 2)
+      PV_FMA(vB, 1)
+      PV_LOAD(vB, 3)
+      PV_FMA(vA, 2)
+      PV_FMA(vB, 3)
+#undef PV_LOAD
+#undef PV_FMA
     }
   }
 
