@@ -119,14 +119,19 @@ class LlamaShard:
         return (self.device.type == "cuda" and T <= 64
                 and n_out % 64 == 0 and n_out * k_in <= 42_000_000)
 
+    def _splitk_for(self, weight: torch.Tensor) -> int:
+        n_out = weight.shape[0]
+        sk = self._splitk_cache.get(n_out)
+        if sk is None:
+            sk = ops.choose_splitk(n_out, weight.shape[1])
+            self._splitk_cache[n_out] = sk
+        return sk
+
     def _proj(self, x: torch.Tensor, weight: torch.Tensor,
               decode: bool) -> torch.Tensor:
         n_out = weight.shape[0]
         if decode and self._use_skinny(x.shape[0], n_out, weight.shape[1]):
-            sk = self._splitk_cache.get(n_out)
-            if sk is None:
-                sk = ops.choose_splitk(n_out, weight.shape[1])
-                self._splitk_cache[n_out] = sk
+            sk = self._splitk_for(weight)
             out = torch.empty(x.shape[0], n_out, dtype=x.dtype, device=x.device)
             ops.skinny_gemm(out, x, weight, self._gemm_work, sk)
             return out
@@ -150,11 +155,24 @@ class LlamaShard:
               page_table: torch.Tensor | None,
               seq_lens: torch.Tensor | None) -> torch.Tensor:
         T = hidden.shape[0]
-        qkv = self._proj(hidden, lw.wqkv, decode=plan is None)
-        # fused epilogue: in-place RoPE on the packed q/k heads + paged-cache
-        # scatter — q/k/v stay strided views of the GEMM output (no copies)
-        ops.rope_qkv_write(qkv, kcache, vcache, self.rope, positions, slots,
-                           self.hq, self.hkv)
+        decode = plan is None
+        qkv_w = lw.wqkv
+        if decode and self._use_skinny(T, qkv_w.shape[0], qkv_w.shape[1]):
+            # slab-fused path: split-K slabs -> {reduce + RoPE + cache write}
+            # in the consumer kernel (no reduce_slabs pass, no re-read)
+            sk = self._splitk_for(qkv_w)
+            Mp = ops.skinny_gemm_slabs(hidden, qkv_w, self._gemm_work, sk)
+            qkv = torch.empty(T, qkv_w.shape[0], dtype=hidden.dtype,
+                              device=hidden.device)
+            ops.rope_qkv_write_slab(qkv, self._gemm_work, kcache, vcache,
+                                    self.rope, positions, slots, self.hq,
+                                    self.hkv, Mp, sk)
+        else:
+            qkv = self._proj(hidden, qkv_w, decode=decode)
+            # fused epilogue: in-place RoPE on the packed q/k heads +
+            # paged-cache scatter — q/k/v stay strided views (no copies)
+            ops.rope_qkv_write(qkv, kcache, vcache, self.rope, positions,
+                               slots, self.hq, self.hkv)
         q = qkv[:, :self.hq * self.dh].view(T, self.hq, self.dh)
         o = torch.empty(T, self.hq, self.dh, dtype=qkv.dtype, device=qkv.device)
         if plan is not None:
@@ -178,8 +196,15 @@ class LlamaShard:
                 plan.past_lens, self.scale)
         else:
             ops.attn_decode(o, q, kcache, vcache, page_table, seq_lens, self.scale)
-        out = self._proj(o.view(T, self.hq * self.dh), lw.wo,
-                         decode=plan is None)
+        o2 = o.view(T, self.hq * self.dh)
+        if (decode and self.tp_size == 1
+                and self._use_skinny(T, lw.wo.shape[0], lw.wo.shape[1])):
+            # leave the o-projection as slabs; the following residual norm
+            # fuses the split-K reduce
+            sk = self._splitk_for(lw.wo)
+            Mp = ops.skinny_gemm_slabs(o2, lw.wo, self._gemm_work, sk)
+            return ("slabs", Mp, sk)
+        out = self._proj(o2, lw.wo, decode=decode)
         return all_reduce_sum(out) if self.tp_size > 1 else out
 
     def _mlp(self, lw: LayerWeights, hidden: torch.Tensor, *,
@@ -241,12 +266,20 @@ class LlamaShard:
         for i, lw in enumerate(self.layers):
             ops.fused_add_rmsnorm(hidden, residual, lw.norm_attn,
                                   self.cfg.norm_eps)
-            hidden = self._attn(
+            attn_out = self._attn(
                 lw, hidden, positions, kcaches[i], vcaches[i], slots,
                 prefill_kv=prefill_scratch[i] if prefill_scratch else None,
                 plan=plan, page_table=page_table, seq_lens=seq_lens)
-            ops.fused_add_rmsnorm(hidden, residual, lw.norm_mlp,
-                                  self.cfg.norm_eps)
+            if isinstance(attn_out, tuple):  # ("slabs", Mp, sk)
+                _, Mp, sk = attn_out
+                hidden = torch.empty_like(residual)
+                ops.fused_add_rmsnorm_slab(hidden, residual, self._gemm_work,
+                                           lw.norm_mlp, self.cfg.norm_eps,
+                                           Mp, sk)
+            else:
+                hidden = attn_out
+                ops.fused_add_rmsnorm(hidden, residual, lw.norm_mlp,
+                                      self.cfg.norm_eps)
             hidden = self._mlp(lw, hidden, decode=plan is None)
         # final residual add + norm
         ops.fused_add_rmsnorm(hidden, residual, self.norm_final,

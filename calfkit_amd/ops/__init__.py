@@ -30,6 +30,9 @@ __all__ = [
     "build_prefill_blocks",
     "sample",
     "skinny_gemm",
+    "skinny_gemm_slabs",
+    "rope_qkv_write_slab",
+    "fused_add_rmsnorm_slab",
     "skinny_gemm_silu",
     "choose_splitk",
     "build_rope_table",
@@ -310,3 +313,49 @@ def skinny_gemm_silu(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     y = x.float() @ w.float().T
     g, u = y[:, :F], y[:, F:]
     out.copy_((torch.nn.functional.silu(g) * u).to(out.dtype))
+
+
+def skinny_gemm_slabs(x: torch.Tensor, w: torch.Tensor, work: torch.Tensor,
+                      splitk: int) -> int:
+    """Run only the slab stage; returns Mp. Consumer fuses the reduce."""
+    Mp = 16 if x.shape[0] <= 16 else (32 if x.shape[0] <= 32 else 64)
+    if _native(x):
+        _K.skinny_gemm_slabs(x, w, work, splitk)
+        return Mp
+    # CPU reference: one fp32 slab with the full product, rest zero
+    N = w.shape[0]
+    y = x.float() @ w.float().T
+    wk = work[: splitk * Mp * N].view(splitk, Mp, N)
+    wk.zero_()
+    wk[0, : x.shape[0]] = y
+    return Mp
+
+
+def rope_qkv_write_slab(qkv: torch.Tensor, work: torch.Tensor,
+                        kcache: torch.Tensor, vcache: torch.Tensor,
+                        cos_sin: torch.Tensor, positions: torch.Tensor,
+                        slots: torch.Tensor, hq: int, hkv: int,
+                        Mp: int, splitk: int) -> None:
+    """qkv = sum(slabs); then RoPE + paged cache write (one launch)."""
+    if _native(qkv):
+        _K.rope_qkv_write_slab(qkv, work, kcache, vcache, cos_sin, positions,
+                               slots, hq, hkv, Mp, splitk)
+        return
+    T, W = qkv.shape
+    summed = work[: splitk * Mp * W].view(splitk, Mp, W)[:, :T].sum(0)
+    qkv.copy_(summed.to(qkv.dtype))
+    rope_qkv_write(qkv, kcache, vcache, cos_sin, positions, slots, hq, hkv)
+
+
+def fused_add_rmsnorm_slab(hidden: torch.Tensor, residual: torch.Tensor,
+                           work: torch.Tensor, weight: torch.Tensor,
+                           eps: float, Mp: int, splitk: int) -> None:
+    """residual += sum(slabs); hidden = rmsnorm(residual)."""
+    if _native(hidden):
+        _K.fused_add_rmsnorm_slab(hidden, residual, work, weight, eps, Mp,
+                                  splitk)
+        return
+    T, d = hidden.shape
+    summed = work[: splitk * Mp * d].view(splitk, Mp, d)[:, :T].sum(0)
+    hidden.copy_(summed.to(hidden.dtype))
+    fused_add_rmsnorm(hidden, residual, weight, eps)

@@ -12,6 +12,16 @@ void rope_qkv_write(torch::Tensor qkv, torch::Tensor kcache,
                     torch::Tensor vcache, torch::Tensor cos_sin,
                     torch::Tensor positions, torch::Tensor slots,
                     int64_t hq, int64_t hkv);
+void rope_qkv_write_slab(torch::Tensor qkv, torch::Tensor work,
+                         torch::Tensor kcache, torch::Tensor vcache,
+                         torch::Tensor cos_sin, torch::Tensor positions,
+                         torch::Tensor slots, int64_t hq, int64_t hkv,
+                         int64_t Mp, int64_t splitk);
+void fused_add_rmsnorm_slab(torch::Tensor hidden, torch::Tensor residual,
+                            torch::Tensor work, torch::Tensor weight,
+                            double eps, int64_t Mp, int64_t splitk);
+void skinny_gemm_slabs(torch::Tensor x, torch::Tensor w, torch::Tensor work,
+                       int64_t splitk);
 void kv_write(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
               torch::Tensor v, torch::Tensor slots);
 void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
@@ -38,6 +48,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_qk", &rope_qk, "in-place neox RoPE on Q and K");
   m.def("rope_qkv_write", &rope_qkv_write,
         "fused QKV epilogue: in-place RoPE + paged-cache K/V scatter");
+  m.def("rope_qkv_write_slab", &rope_qkv_write_slab,
+        "split-K reduce + RoPE + cache scatter in one launch");
+  m.def("fused_add_rmsnorm_slab", &fused_add_rmsnorm_slab,
+        "split-K reduce + residual add + RMSNorm in one launch");
+  m.def("skinny_gemm_slabs", &skinny_gemm_slabs,
+        "skinny GEMM, slabs only (consumer fuses the reduce)");
   m.def("kv_write", &kv_write, "scatter K/V rows into the paged cache");
   m.def("attn_decode", &attn_decode, "paged GQA decode attention");
   m.def("attn_prefill", &attn_prefill, "causal GQA prefill attention (MFMA)");

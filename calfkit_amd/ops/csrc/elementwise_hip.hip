@@ -235,6 +235,103 @@ __global__ void rope_qkv_write_kernel(
   }
 }
 
+// slab-fused variant: qkv = sum_sk(work) cast bf16, then rope + cache
+// write in the same launch (the split-K reduce of the QKV projection rides
+// the epilogue instead of its own kernel)
+__global__ void rope_qkv_write_slab_kernel(
+    __hip_bfloat16* __restrict__ qkv,        // [T, W] out
+    const float* __restrict__ work,          // [sk, Mp, W] slabs
+    __hip_bfloat16* __restrict__ kcache,
+    __hip_bfloat16* __restrict__ vcache,
+    const float* __restrict__ cos_sin,
+    const int32_t* __restrict__ positions,
+    const int32_t* __restrict__ slots,
+    const int hq, const int hkv, const int dh, const int page,
+    const int Mp, const int splitk) {
+  const int64_t token = blockIdx.x;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int nwave = blockDim.x / WAVE;
+  const int half = dh / 2;
+  const int W = (hq + 2 * hkv) * dh;
+  const int pos = positions[token];
+  const int slot = slots[token];
+  const float* cs = cos_sin + (int64_t)pos * dh;
+  __hip_bfloat16* row = qkv + token * (int64_t)W;
+
+  for (int job = wid; job < hq + 2 * hkv; job += nwave) {
+    const int col0 = job * dh;
+    // reduce this head's slab columns; lane l owns elements (l, l+64)
+    float v0 = 0.f, v1 = 0.f;
+    for (int sk = 0; sk < splitk; ++sk) {
+      const float* slab = work + ((int64_t)sk * Mp + token) * W + col0;
+      v0 += slab[lane];
+      v1 += slab[half + lane];
+    }
+    if (job < hq + hkv) {
+      const float c = cs[lane], sn = cs[half + lane];
+      const float r0 = v0 * c - v1 * sn;
+      const float r1 = v1 * c + v0 * sn;
+      v0 = r0;
+      v1 = r1;
+    }
+    row[col0 + lane] = f2bf(v0);
+    row[col0 + half + lane] = f2bf(v1);
+    if (job >= hq && slot >= 0) {
+      const int h = (job - hq) % hkv;
+      const bool is_k = job < hq + hkv;
+      const int p = slot / page, off = slot % page;
+      int64_t dst = (((int64_t)p * hkv + h) * page + off) * dh;
+      __hip_bfloat16* d = (is_k ? kcache : vcache) + dst;
+      d[lane] = f2bf(v0);
+      d[half + lane] = f2bf(v1);
+    }
+  }
+}
+
+// slab-fused variant of fused_add_rmsnorm: residual += sum_sk(slabs);
+// hidden = rmsnorm(residual) — the split-K reduce of a row-parallel
+// projection (o / down) fused into the consuming norm.
+template <int BLOCK>
+__global__ void fused_add_rmsnorm_slab_kernel(
+    __hip_bfloat16* __restrict__ hidden,     // [T, d] out (the norm)
+    __hip_bfloat16* __restrict__ residual,   // [T, d] in/out
+    const float* __restrict__ work,          // [sk, Mp, d]
+    const __hip_bfloat16* __restrict__ weight,
+    const int d, const float eps, const int Mp, const int splitk) {
+  __shared__ float red[BLOCK / WAVE];
+  const int64_t row = blockIdx.x;
+  __hip_bfloat16* h = hidden + row * d;
+  __hip_bfloat16* r = residual + row * d;
+
+  float ss = 0.f;
+  const int nvec = d / 8;
+  bf16x8* rv = reinterpret_cast<bf16x8*>(r);
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    bf16x8 b = rv[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(b.v[j]);
+      for (int sk = 0; sk < splitk; ++sk)
+        f += work[((int64_t)sk * Mp + row) * d + i * 8 + j];
+      b.v[j] = f2bf(f);
+      float fr = bf2f(b.v[j]);  // norm over the bf16-rounded sum
+      ss += fr * fr;
+    }
+    rv[i] = b;
+  }
+  ss = block_sum<BLOCK>(ss, red);
+  const float inv = rsqrtf(ss / d + eps);
+  const bf16x8* wv = reinterpret_cast<const bf16x8*>(weight);
+  bf16x8* hv = reinterpret_cast<bf16x8*>(h);
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    bf16x8 b = rv[i], w = wv[i], o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o.v[j] = f2bf(bf2f(b.v[j]) * inv * bf2f(w.v[j]));
+    hv[i] = o;
+  }
+}
+
 // ------------------------------------------------------------- host glue
 
 #include <torch/extension.h>
@@ -328,6 +425,47 @@ void rope_qkv_write(torch::Tensor qkv, torch::Tensor kcache,
                      (const int32_t*)positions.data_ptr(),
                      (const int32_t*)slots.data_ptr(), (int)hq, (int)hkv, dh,
                      page);
+  HIP_CHECK_LAST();
+}
+
+void rope_qkv_write_slab(torch::Tensor qkv, torch::Tensor work,
+                         torch::Tensor kcache, torch::Tensor vcache,
+                         torch::Tensor cos_sin, torch::Tensor positions,
+                         torch::Tensor slots, int64_t hq, int64_t hkv,
+                         int64_t Mp, int64_t splitk) {
+  TORCH_CHECK(qkv.is_contiguous() && work.is_contiguous());
+  const int T = qkv.size(0);
+  const int page = kcache.size(2);
+  const int dh = kcache.size(3);
+  TORCH_CHECK(dh == 2 * WAVE);
+  hipLaunchKernelGGL(rope_qkv_write_slab_kernel, dim3(T), dim3(256), 0,
+                     cur_stream(),
+                     (__hip_bfloat16*)qkv.data_ptr(),
+                     (const float*)work.data_ptr(),
+                     (__hip_bfloat16*)kcache.data_ptr(),
+                     (__hip_bfloat16*)vcache.data_ptr(),
+                     (const float*)cos_sin.data_ptr(),
+                     (const int32_t*)positions.data_ptr(),
+                     (const int32_t*)slots.data_ptr(), (int)hq, (int)hkv, dh,
+                     page, (int)Mp, (int)splitk);
+  HIP_CHECK_LAST();
+}
+
+void fused_add_rmsnorm_slab(torch::Tensor hidden, torch::Tensor residual,
+                            torch::Tensor work, torch::Tensor weight,
+                            double eps, int64_t Mp, int64_t splitk) {
+  TORCH_CHECK(hidden.is_contiguous() && residual.is_contiguous());
+  const int d = hidden.size(-1);
+  TORCH_CHECK(d % 8 == 0);
+  const int64_t rows = hidden.numel() / d;
+  constexpr int BLOCK = 256;
+  hipLaunchKernelGGL(fused_add_rmsnorm_slab_kernel<BLOCK>, dim3(rows),
+                     dim3(BLOCK), 0, cur_stream(),
+                     (__hip_bfloat16*)hidden.data_ptr(),
+                     (__hip_bfloat16*)residual.data_ptr(),
+                     (const float*)work.data_ptr(),
+                     (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps,
+                     (int)Mp, (int)splitk);
   HIP_CHECK_LAST();
 }
 

@@ -253,6 +253,19 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   HIP_CHECK_LAST();
 }
 
+void skinny_gemm_slabs(torch::Tensor x, torch::Tensor w, torch::Tensor work,
+                       int64_t splitk) {
+  // slab-only launch: the consumer kernel fuses the split-K reduce
+  const int M = x.size(0), N = w.size(0);
+  const int Mp = pad_m(M);
+  TORCH_CHECK(work.numel() >= splitk * (int64_t)Mp * N, "workspace too small");
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  auto dummy_out = work;  // unused when direct=0
+  launch_skinny(work, x /*placeholder bf16 tensor, unused*/, x, w, splitk, Mp,
+                s, /*direct=*/0);
+  HIP_CHECK_LAST();
+}
+
 void skinny_gemm_silu(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                       torch::Tensor work, int64_t splitk) {
   // fused SiLU-MLP front half: out[M,F] = silu(x@Wg^T) * (x@Wu^T) where

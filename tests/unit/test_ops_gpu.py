@@ -300,3 +300,56 @@ def test_skinny_gemm_silu_fused():
     ref = torch.nn.functional.silu(g) * u
     scale = ref.abs().max()
     assert ((out.float() - ref).abs().max() / scale) < 2e-2
+
+
+def test_slab_fused_qkv_epilogue_matches_unfused():
+    T, hq, hkv, dh, page = 32, 8, 2, 128, 64
+    d = 512
+    W = (hq + 2 * hkv) * dh
+    x = torch.randn(T, d, dtype=torch.bfloat16, device="cuda")
+    wqkv = torch.randn(W, d, dtype=torch.bfloat16, device="cuda") / d ** 0.5
+    pos = torch.randint(0, 256, (T,), dtype=torch.int32, device="cuda")
+    slots = torch.arange(T, dtype=torch.int32, device="cuda")
+    table = ops.build_rope_table(512, dh, device="cuda")
+    sk = 2
+    work = torch.empty(sk * 32 * W, dtype=torch.float32, device="cuda")
+
+    # fused slab path
+    kc1 = torch.zeros(2, hkv, page, dh, dtype=torch.bfloat16, device="cuda")
+    vc1 = torch.zeros_like(kc1)
+    qkv1 = torch.empty(T, W, dtype=torch.bfloat16, device="cuda")
+    Mp = ops.skinny_gemm_slabs(x, wqkv, work, sk)
+    ops.rope_qkv_write_slab(qkv1, work, kc1, vc1, table, pos, slots, hq, hkv,
+                            Mp, sk)
+
+    # unfused reference path
+    kc2 = torch.zeros_like(kc1)
+    vc2 = torch.zeros_like(vc1)
+    qkv2 = (x.float() @ wqkv.float().T).to(torch.bfloat16)
+    ops.rope_qkv_write(qkv2, kc2, vc2, table, pos, slots, hq, hkv)
+    assert torch.allclose(qkv1.float(), qkv2.float(), atol=3e-2, rtol=3e-2), \
+        (qkv1.float() - qkv2.float()).abs().max()
+    assert torch.allclose(kc1.float(), kc2.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(vc1.float(), vc2.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_slab_fused_residual_norm_matches_unfused():
+    T, d, dk = 32, 512, 1024
+    x = torch.randn(T, dk, dtype=torch.bfloat16, device="cuda")
+    wo = torch.randn(d, dk, dtype=torch.bfloat16, device="cuda") / dk ** 0.5
+    weight = torch.randn(d, dtype=torch.bfloat16, device="cuda")
+    resid = torch.randn(T, d, dtype=torch.bfloat16, device="cuda")
+    sk = 4
+    work = torch.empty(sk * 32 * d, dtype=torch.float32, device="cuda")
+
+    r1 = resid.clone()
+    h1 = torch.empty_like(r1)
+    Mp = ops.skinny_gemm_slabs(x, wo, work, sk)
+    ops.fused_add_rmsnorm_slab(h1, r1, work, weight, 1e-5, Mp, sk)
+
+    r2 = resid.clone()
+    h2 = (x.float() @ wo.float().T).to(torch.bfloat16)
+    ops.fused_add_rmsnorm(h2, r2, weight, 1e-5)
+    assert torch.allclose(r1.float(), r2.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(h1.float(), h2.float(), atol=3e-2, rtol=3e-2), \
+        (h1.float() - h2.float()).abs().max()
