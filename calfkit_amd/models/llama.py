@@ -96,11 +96,18 @@ class LlamaShard:
                 norm_attn=torch.ones(d, device=self.device, dtype=dtype),
                 norm_mlp=torch.ones(d, device=self.device, dtype=dtype),
             ))
-        # fp32 workspace for the skinny decode GEMMs' split-K slabs
-        self._gemm_work = torch.empty(
-            8 * 64 * max(qkv_out, 2 * self.ffn_s, d),
-            dtype=torch.float32, device=self.device) \
-            if self.device.type == "cuda" else torch.empty(0)
+        # fp32 workspaces for the skinny decode GEMMs' split-K slabs; the
+        # second one carries down-proj slabs across the layer boundary while
+        # the next layer's qkv GEMM already writes the first
+        work_elems = 8 * 64 * max(qkv_out, 2 * self.ffn_s, d, cfg.vocab_size)
+        if self.device.type == "cuda":
+            self._gemm_work = torch.empty(work_elems, dtype=torch.float32,
+                                          device=self.device)
+            self._gemm_work2 = torch.empty(8 * 64 * d, dtype=torch.float32,
+                                           device=self.device)
+        else:
+            self._gemm_work = torch.empty(0)
+            self._gemm_work2 = torch.empty(0)
         self.norm_final = torch.ones(d, device=self.device, dtype=dtype)
         # lm_head replicated (vocab-parallel sharding is a later optimization;
         # 128256 x 8192 bf16 is ~2 GB of the 288 GB rank)
@@ -111,13 +118,14 @@ class LlamaShard:
 
     # ------------------------------------------------------- GEMM dispatch
     def _use_skinny(self, T: int, n_out: int, k_in: int) -> bool:
-        """Decode-regime projections go to the hand-written kernel where it
-        measured faster than hipBLASLt (profiles/r01_gemm_sweep.md): the
-        win region is N*K <= ~40M weight elements (qkv/o and all TP-sharded
-        70B projections); the giant tiles (gate-up, down, lm_head) stay on
-        the library until the glds-staged variant lands."""
-        return (self.device.type == "cuda" and T <= 64
-                and n_out % 64 == 0 and n_out * k_in <= 42_000_000)
+        """Every decode-regime projection goes to the hand-written kernel:
+        eager hipBLASLt beats it on the giant tiles (profiles/
+        r01_gemm_sweep.md) but hipBLASLt UNDER GRAPH CAPTURE picks worse
+        tiles than skinny on every decode shape measured (53.5 vs 46.9 us
+        gate-up, 33.4 vs 25.3 down — r03 profile), so the decode graph is
+        fully hand-written."""
+        return (self.device.type == "cuda" and T <= 64 and n_out % 64 == 0
+                and (k_in % 128 == 0))
 
     def _splitk_for(self, weight: torch.Tensor) -> int:
         n_out = weight.shape[0]
@@ -208,7 +216,7 @@ class LlamaShard:
         return all_reduce_sum(out) if self.tp_size > 1 else out
 
     def _mlp(self, lw: LayerWeights, hidden: torch.Tensor, *,
-             decode: bool = False) -> torch.Tensor:
+             decode: bool = False):
         T = hidden.shape[0]
         act = torch.empty(T, self.ffn_s, dtype=hidden.dtype,
                           device=hidden.device)
@@ -218,6 +226,12 @@ class LlamaShard:
         else:
             gateup = torch.nn.functional.linear(hidden, lw.wgateup)
             ops.silu_mul(act, gateup)
+        if (decode and self.tp_size == 1
+                and self._use_skinny(T, lw.wdown.shape[0], lw.wdown.shape[1])):
+            # down stays as slabs; the NEXT residual norm fuses the reduce
+            sk = self._splitk_for(lw.wdown)
+            Mp = ops.skinny_gemm_slabs(act, lw.wdown, self._gemm_work2, sk)
+            return ("slabs2", Mp, sk)
         down = self._proj(act, lw.wdown, decode=decode)
         return all_reduce_sum(down) if self.tp_size > 1 else down
 
@@ -264,8 +278,15 @@ class LlamaShard:
         hidden = self.embed[tokens.long()]  # [T, d]
         residual = torch.zeros_like(hidden)
         for i, lw in enumerate(self.layers):
-            ops.fused_add_rmsnorm(hidden, residual, lw.norm_attn,
-                                  self.cfg.norm_eps)
+            if isinstance(hidden, tuple):  # down-proj slabs from layer i-1
+                _, Mp, sk = hidden
+                hidden = torch.empty_like(residual)
+                ops.fused_add_rmsnorm_slab(hidden, residual, self._gemm_work2,
+                                           lw.norm_attn, self.cfg.norm_eps,
+                                           Mp, sk)
+            else:
+                ops.fused_add_rmsnorm(hidden, residual, lw.norm_attn,
+                                      self.cfg.norm_eps)
             attn_out = self._attn(
                 lw, hidden, positions, kcaches[i], vcaches[i], slots,
                 prefill_kv=prefill_scratch[i] if prefill_scratch else None,
@@ -282,8 +303,15 @@ class LlamaShard:
                                       self.cfg.norm_eps)
             hidden = self._mlp(lw, hidden, decode=plan is None)
         # final residual add + norm
-        ops.fused_add_rmsnorm(hidden, residual, self.norm_final,
-                              self.cfg.norm_eps)
+        if isinstance(hidden, tuple):
+            _, Mp, sk = hidden
+            hidden = torch.empty_like(residual)
+            ops.fused_add_rmsnorm_slab(hidden, residual, self._gemm_work2,
+                                       self.norm_final, self.cfg.norm_eps,
+                                       Mp, sk)
+        else:
+            ops.fused_add_rmsnorm(hidden, residual, self.norm_final,
+                                  self.cfg.norm_eps)
         if logits_rows is not None:
             hidden = hidden[logits_rows]
         return torch.nn.functional.linear(hidden, self.lm_head)
