@@ -118,14 +118,14 @@ class LlamaShard:
 
     # ------------------------------------------------------- GEMM dispatch
     def _use_skinny(self, T: int, n_out: int, k_in: int) -> bool:
-        """Every decode-regime projection goes to the hand-written kernel:
-        eager hipBLASLt beats it on the giant tiles (profiles/
-        r01_gemm_sweep.md) but hipBLASLt UNDER GRAPH CAPTURE picks worse
-        tiles than skinny on every decode shape measured (53.5 vs 46.9 us
-        gate-up, 33.4 vs 25.3 down — r03 profile), so the decode graph is
-        fully hand-written."""
+        """Decode projections route to the hand-written kernel in its
+        measured win region (profiles/r01_gemm_sweep.md): N*K <= ~40M weight
+        elements (qkv/o and the TP-sharded 70B shapes).  On the giant tiles
+        hipBLASLt streams at ~6.5 TB/s (roofline) and keeps them; the
+        all-skinny experiment (r6) measured a net regression 70.4 -> 66.6
+        turns/s and was reverted."""
         return (self.device.type == "cuda" and T <= 64 and n_out % 64 == 0
-                and (k_in % 128 == 0))
+                and k_in % 128 == 0 and n_out * k_in <= 42_000_000)
 
     def _splitk_for(self, weight: torch.Tensor) -> int:
         n_out = weight.shape[0]
