@@ -143,7 +143,7 @@ class AgentProcessor:
         # 2. admit
         kind, call_id = _admission_key(delivery)
         admission = carriage.admit(kind, call_id)
-        ledger.note(admission=admission.value, kind=kind)
+        ledger.note(admission=admission.value, delivery_kind=kind)
         if admission == Admission.DISCARD:
             await ledger.flush(self.machinery.bus, self.machinery.namespace)
             return
@@ -172,10 +172,16 @@ class AgentProcessor:
         try:
             await self._proceed(carriage, delivery, ledger, hop)
         except Exception as exc:  # the hop-level fault floor
+            # echo the ORIGINAL caller's marker so the fault correlates at
+            # the client / calling agent
+            marker = getattr(delivery, "marker", None)
+            pm = carriage.record.metadata.get("peer_marker")
+            if marker is None and pm:
+                marker = EchoMarker.from_wire(pm)
             await self.funnel.emit(exc, thread_id=thread_id,
                                    reply=_stored_reply(carriage) or
                                    getattr(delivery, "reply", None),
-                                   marker=getattr(delivery, "marker", None))
+                                   marker=marker)
             ledger.note(outcome="fault", error=str(exc))
         await ledger.flush(self.machinery.bus, self.machinery.namespace)
 
