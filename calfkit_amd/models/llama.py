@@ -124,8 +124,12 @@ class LlamaShard:
         hipBLASLt streams at ~6.5 TB/s (roofline) and keeps them; the
         all-skinny experiment (r6) measured a net regression 70.4 -> 66.6
         turns/s and was reverted."""
+        import os
+
+        big = os.environ.get("CALF_DECODE_BIG_SKINNY", "0") == "1"
+        limit = 10 ** 18 if big else 42_000_000
         return (self.device.type == "cuda" and T <= 64 and n_out % 64 == 0
-                and k_in % 128 == 0 and n_out * k_in <= 42_000_000)
+                and k_in % 128 == 0 and n_out * k_in <= limit)
 
     def _splitk_for(self, weight: torch.Tensor) -> int:
         n_out = weight.shape[0]

@@ -258,7 +258,11 @@ __global__ void rope_qkv_write_slab_kernel(
   const float* cs = cos_sin + (int64_t)pos * dh;
   __hip_bfloat16* row = qkv + token * (int64_t)W;
 
-  for (int job = wid; job < hq + 2 * hkv; job += nwave) {
+  // grid.y spreads the head-jobs: block (x=token, y) covers jobs
+  // [y*nwave .. ) striding gridDim.y*nwave — T x 3 blocks for llama-8b
+  // instead of T (the single-block version was launch-latency bound)
+  for (int job = blockIdx.y * nwave + wid; job < hq + 2 * hkv;
+       job += gridDim.y * nwave) {
     const int col0 = job * dh;
     // reduce this head's slab columns; lane l owns elements (l, l+64)
     float v0 = 0.f, v1 = 0.f;
@@ -308,13 +312,20 @@ __global__ void fused_add_rmsnorm_slab_kernel(
   bf16x8* rv = reinterpret_cast<bf16x8*>(r);
   for (int i = threadIdx.x; i < nvec; i += BLOCK) {
     bf16x8 b = rv[i];
+    float acc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = bf2f(b.v[j]);
+    for (int sk = 0; sk < splitk; ++sk) {
+      const float4* wp = reinterpret_cast<const float4*>(
+          work + ((int64_t)sk * Mp + row) * d + i * 8);
+      const float4 w0 = wp[0], w1 = wp[1];
+      acc[0] += w0.x; acc[1] += w0.y; acc[2] += w0.z; acc[3] += w0.w;
+      acc[4] += w1.x; acc[5] += w1.y; acc[6] += w1.z; acc[7] += w1.w;
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float f = bf2f(b.v[j]);
-      for (int sk = 0; sk < splitk; ++sk)
-        f += work[((int64_t)sk * Mp + row) * d + i * 8 + j];
-      b.v[j] = f2bf(f);
-      float fr = bf2f(b.v[j]);  // norm over the bf16-rounded sum
+      b.v[j] = f2bf(acc[j]);
+      const float fr = bf2f(b.v[j]);  // norm over the bf16-rounded sum
       ss += fr * fr;
     }
     rv[i] = b;
@@ -437,8 +448,9 @@ void rope_qkv_write_slab(torch::Tensor qkv, torch::Tensor work,
   const int page = kcache.size(2);
   const int dh = kcache.size(3);
   TORCH_CHECK(dh == 2 * WAVE);
-  hipLaunchKernelGGL(rope_qkv_write_slab_kernel, dim3(T), dim3(256), 0,
-                     cur_stream(),
+  const int jobs = (int)(hq + 2 * hkv);
+  hipLaunchKernelGGL(rope_qkv_write_slab_kernel, dim3(T, (jobs + 3) / 4),
+                     dim3(256), 0, cur_stream(),
                      (__hip_bfloat16*)qkv.data_ptr(),
                      (const float*)work.data_ptr(),
                      (__hip_bfloat16*)kcache.data_ptr(),
