@@ -26,9 +26,9 @@ import time
 
 import torch
 
-PROMPT_TOKENS = 128
-DECODE_TOKENS = 64
-TURNS_PER_STEP = 32
+PROMPT_TOKENS = int(os.environ.get("CALF_BENCH_PROMPT", "128"))
+DECODE_TOKENS = int(os.environ.get("CALF_BENCH_DECODE", "64"))
+TURNS_PER_STEP = int(os.environ.get("CALF_BENCH_TURNS", "32"))
 
 
 def log(msg: str) -> None:
