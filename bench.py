@@ -80,26 +80,49 @@ async def run_rank(args, dist) -> dict:
         if record:
             latencies.append((time.perf_counter() - t0) * 1e3)
 
-    async def one_step(client, step: int, record: bool) -> None:
-        await asyncio.gather(*(one_turn(client, step * TURNS_PER_STEP + i, record)
-                               for i in range(TURNS_PER_STEP)))
-
     async with Worker(agent, bus=bus, store=store, directory=directory):
         client = await Client.connect(bus, directory)
-        # engine spin-up (weights + graph capture) happens on first turn
-        for wstep in range(args.warmup):
-            await one_step(client, -1 - wstep, record=False)
+        # rolling pool: TURNS_PER_STEP turns stay in flight continuously
+        # (steady-state serving — no wave boundaries idling the engine);
+        # one "step" = the next TURNS_PER_STEP completions
+        done = asyncio.Queue()
+        counter = 0
+        recording = False
+
+        def launch():
+            nonlocal counter
+            counter += 1
+
+            async def turn(i=counter):
+                await one_turn(client, i, recording)
+                done.put_nowait(None)
+
+            return asyncio.ensure_future(turn())
+
+        tasks = {launch() for _ in range(TURNS_PER_STEP)}
+
+        async def drain(n: int) -> None:
+            for _ in range(n):
+                await done.get()
+                tasks.add(launch())
+
+        # warmup covers engine spin-up (weights + graph capture)
+        await drain(args.warmup * TURNS_PER_STEP)
 
         barrier()
         if on_gpu:
             torch.cuda.synchronize()
+        recording = True
         t0 = time.perf_counter()
-        for step in range(args.steps):
-            await one_step(client, step, record=True)
+        await drain(args.steps * TURNS_PER_STEP)
         if on_gpu:
             torch.cuda.synchronize()
         barrier()
         elapsed = time.perf_counter() - t0
+        recording = False
+        for t in tasks:
+            t.cancel()
+        await asyncio.gather(*tasks, return_exceptions=True)
 
     return {"elapsed": elapsed, "latencies": latencies,
             "engine_stats": provider.engine.stats, "model": cfg.name}
