@@ -131,11 +131,23 @@ async def drive_step(engine: Engine, surface: StepSurface, policy: TurnPolicy,
                                 call_id=call.call_id, tool_name=call.tool_name,
                                 content=f"message sent to {effect.target}"))
                 elif binding is not None:
-                    # D4 — mesh tool fan-out
-                    pending_tools.append(PendingToolCall(
-                        call_id=call.call_id, tool_name=call.tool_name,
-                        arguments=dict(call.arguments),
-                        target_inbox=binding.target_inbox))
+                    # D4 — mesh tool fan-out; args are validated against the
+                    # ADVERTISED schema before anything hits the wire, so a
+                    # malformed call costs one correction step, not a mesh
+                    # round trip (caller-side validation,
+                    # /root/reference/CHANGELOG.md:43)
+                    err = _validate_args(binding.schema.parameters,
+                                         call.arguments)
+                    if err is not None:
+                        rejects.append(ToolReturnPart(
+                            call_id=call.call_id, tool_name=call.tool_name,
+                            content=f"invalid arguments: {err}",
+                            is_error=True))
+                    else:
+                        pending_tools.append(PendingToolCall(
+                            call_id=call.call_id, tool_name=call.tool_name,
+                            arguments=dict(call.arguments),
+                            target_inbox=binding.target_inbox))
                 else:
                     unknown.append(ToolReturnPart(
                         call_id=call.call_id, tool_name=call.tool_name,
@@ -176,6 +188,27 @@ async def drive_step(engine: Engine, surface: StepSurface, policy: TurnPolicy,
             return resolution
         resolution.new_messages.append(shape.correction)
         work = work + (shape.correction,)
+
+
+def _validate_args(schema: dict[str, Any], args: dict[str, Any]) -> str | None:
+    """Structural check of call args against a JSON-schema parameters dict."""
+    from ._policy import _check_schema
+
+    if not isinstance(args, dict):
+        return "arguments must be an object"
+    props = schema.get("properties", {})
+    unknown = set(args) - set(props) if props else set()
+    if props and unknown:
+        return f"unknown arguments: {sorted(unknown)}"
+    for req in schema.get("required", ()):  # required fields
+        if req not in args:
+            return f"missing required argument {req!r}"
+    for k, v in args.items():
+        if k in props:
+            err = _check_schema(v, props[k], f"$.{k}")
+            if err:
+                return err
+    return None
 
 
 def _realize_capability(surface: StepSurface, call: ToolCallPart,

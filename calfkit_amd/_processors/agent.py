@@ -198,6 +198,9 @@ class AgentProcessor:
                 carriage.working_extend(list(delivery.seed))
             if delivery.metadata.get("marker"):
                 carriage.record.metadata["peer_marker"] = delivery.metadata["marker"]
+            if delivery.metadata.get("model_settings"):
+                carriage.record.metadata["model_settings"] = \
+                    delivery.metadata["model_settings"]
             carriage.working_extend([delivery.message])
             await self._step_and_realize(carriage, ledger, hop,
                                          handoff=delivery.handoff)
@@ -293,8 +296,19 @@ class AgentProcessor:
         loop = asyncio.get_running_loop()
         sink = _BusSink(self, thread_id, _stored_reply(carriage), loop)
         sink.emit(StreamChunk(event="step_started", data={"hop": hop}))
+        settings_override = amendment.settings
+        if settings_override is None:
+            raw_ms = carriage.record.metadata.get("model_settings")
+            if raw_ms:
+                from ..schemas import ModelSettings
+
+                known = {k: v for k, v in raw_ms.items()
+                         if k in ModelSettings.__dataclass_fields__}
+                if "stop" in known:
+                    known["stop"] = tuple(known["stop"])
+                settings_override = ModelSettings(**known)
         resolution = await drive_step(self.engine, surface, self.policy, history,
-                                      sink, ctx, settings_override=amendment.settings)
+                                      sink, ctx, settings_override=settings_override)
         ledger.note(steps=len([m for m in resolution.new_messages
                                if isinstance(m, AssistantMessage)]),
                     pending_tools=len(resolution.pending_tool_calls),

@@ -160,8 +160,13 @@ class Client:
 
     async def execute(self, agent: str, content: Any, *,
                       thread_id: str | None = None, timeout: float = 300.0,
-                      output_type: Any = None) -> TurnResult:
-        """Request/response: publish, await the reply on the private inbox."""
+                      output_type: Any = None,
+                      model_settings: Any = None) -> TurnResult:
+        """Request/response: publish, await the reply on the private inbox.
+
+        ``model_settings`` overrides the agent's settings for this one turn
+        (runtime model_settings, /root/reference/CHANGELOG.md:406).
+        """
         thread_id = thread_id or uuid.uuid4().hex
         call_id = uuid.uuid4().hex
         fut: asyncio.Future = asyncio.get_running_loop().create_future()
@@ -169,7 +174,8 @@ class Client:
         message = self._as_message(content)
         reply = ReplySlot(inbox=self.reply_inbox, key=thread_id, node=self.client_id)
         marker = EchoMarker(call_id=call_id, caller=self.client_id)
-        await self._publish_call(agent, thread_id, message, reply, marker)
+        await self._publish_call(agent, thread_id, message, reply, marker,
+                                 model_settings=model_settings)
         try:
             result: TurnResult = await asyncio.wait_for(fut, timeout)
         except asyncio.TimeoutError:
@@ -214,11 +220,18 @@ class Client:
 
     # ------------------------------------------------------------- internals
     async def _publish_call(self, agent: str, thread_id: str, message: Message,
-                            reply: ReplySlot | None, marker: EchoMarker | None) -> None:
+                            reply: ReplySlot | None, marker: EchoMarker | None,
+                            model_settings: Any = None) -> None:
         inbox = topology.inbox_topic(agent, self.namespace)
+        metadata = None
+        if model_settings is not None:
+            from dataclasses import asdict
+
+            metadata = {"model_settings": asdict(model_settings)}
         value, headers = encode_envelope(
             "agent_call", to=agent, thread_id=thread_id,
-            payload=agent_call_payload(message), reply=reply, marker=marker,
+            payload=agent_call_payload(message, metadata=metadata),
+            reply=reply, marker=marker,
             emitter=self.client_id)
         if len(value) > self.max_message_bytes:
             raise ValueError(f"message {len(value)}B exceeds max_message_bytes")

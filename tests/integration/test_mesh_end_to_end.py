@@ -185,3 +185,39 @@ async def test_multi_turn_same_thread_keeps_history():
         r2 = await client.execute("memory", "two", thread_id="t1", timeout=5)
         assert r1.text == "seen 1 user messages"
         assert r2.text == "seen 2 user messages"
+
+
+async def test_runtime_model_settings_override():
+    from calfkit_amd.schemas import ModelSettings
+
+    bus, store, directory = mesh()
+    provider = FakeProvider(["ok"])
+    agent = Agent(name="tunable", model_client=provider,
+                  model_settings=ModelSettings(max_tokens=100))
+    async with Worker(agent, bus=bus, store=store, directory=directory):
+        client = await Client.connect(bus, directory)
+        await client.execute("tunable", "hi", timeout=5,
+                             model_settings=ModelSettings(max_tokens=7,
+                                                          temperature=0.5))
+    req = provider.requests[0]
+    assert req.settings.max_tokens == 7
+    assert req.settings.temperature == 0.5
+
+
+async def test_invalid_tool_args_rejected_before_the_wire():
+    bus, store, directory = mesh()
+    from calfkit_amd.messages import AssistantMessage, ToolCallPart
+
+    provider = FakeProvider([
+        AssistantMessage((ToolCallPart("c1", "get_weather",
+                                       {"town": "Lisbon"}),)),  # wrong key
+        "fixed it",
+    ])
+    agent = Agent(name="validator", tools=[get_weather], model_client=provider)
+    async with Worker(agent, get_weather, bus=bus, store=store,
+                      directory=directory):
+        client = await Client.connect(bus, directory)
+        r = await client.execute("validator", "weather", timeout=5)
+        assert r.text == "fixed it"
+    tool_msgs = [m for m in provider.requests[-1].messages if m.role == "tool"]
+    assert "invalid arguments" in str(tool_msgs[0].parts[0].content)
