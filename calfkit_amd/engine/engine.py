@@ -82,6 +82,11 @@ class InferenceEngine:
     def __init__(self, model: LlamaShard, *, max_batch: int = 64,
                  kv_pages: int | None = None, use_graphs: bool | None = None,
                  max_model_len: int | None = None):
+        #: TP>1 serving runs in lockstep: rank 0 schedules and broadcasts
+        #: each action; ranks > 0 replay them via follow() — every rank then
+        #: executes identical forwards (and RCCL collectives) in identical
+        #: order, and identical seeds make the sampled tokens identical.
+        self.tp_lockstep = model.tp_size > 1
         self.model = model
         self.cfg: ModelConfig = model.cfg
         self.device = model.device
@@ -156,6 +161,11 @@ class InferenceEngine:
             self._queue.put(None)
             self._thread.join(timeout=30)
             self._started = False
+            if self.tp_lockstep:
+                try:
+                    self._tp_bcast(("stop",))
+                except Exception:
+                    _log.exception("tp stop broadcast failed")
 
     @property
     def stats(self) -> dict:
@@ -177,6 +187,44 @@ class InferenceEngine:
 
     def generate(self, request: GenRequest, timeout: float = 600.0) -> GenResult:
         return self.submit(request).result(timeout)
+
+    # ----------------------------------------------------------- TP lockstep
+    def _tp_bcast(self, op) -> None:
+        if self.tp_lockstep:
+            import torch.distributed as td
+
+            objs = [op]
+            td.broadcast_object_list(objs, src=0)
+
+    def _tp_recv(self):
+        import torch.distributed as td
+
+        objs = [None]
+        td.broadcast_object_list(objs, src=0)
+        return objs[0]
+
+    def follow(self) -> None:
+        """Follower loop for TP ranks > 0 (call instead of start())."""
+        assert self.tp_lockstep
+        if self.use_graphs:
+            try:
+                self._capture_graphs()
+            except Exception:
+                _log.exception("follower graph capture failed; eager")
+                self.use_graphs = False
+        while True:
+            op = self._tp_recv()
+            if op[0] == "stop":
+                return
+            if op[0] == "prefill":
+                seqs = []
+                for sid, ids, mx, temp, topk, seed in op[1]:
+                    req = GenRequest(prompt_ids=list(ids), max_new_tokens=mx,
+                                     temperature=temp, top_k=topk, seed=seed)
+                    seqs.append(_Seq(seq_id=sid, request=req, future=Future()))
+                self._prefill_batch(seqs)
+            elif op[0] == "decode":
+                self._decode_step()
 
     # ------------------------------------------------------------------ loop
     def _loop(self) -> None:
@@ -206,10 +254,21 @@ class InferenceEngine:
                         budget -= n
                         batch.append(self._waiting.pop(0))
                     if batch:
+                        self._tp_bcast(("prefill", [
+                            (b.seq_id, b.request.prompt_ids,
+                             b.request.max_new_tokens, b.request.temperature,
+                             b.request.top_k, b.request.seed) for b in batch]))
                         self._prefill_batch(batch)
                     elif self._waiting:
-                        self._prefill(self._waiting.pop(0))
+                        solo = self._waiting.pop(0)
+                        self._tp_bcast(("prefill", [
+                            (solo.seq_id, solo.request.prompt_ids,
+                             solo.request.max_new_tokens,
+                             solo.request.temperature, solo.request.top_k,
+                             solo.request.seed)]))
+                        self._prefill_batch([solo])
                 if self._running:
+                    self._tp_bcast(("decode",))
                     self._decode_step()
             except Exception as e:
                 _log.exception("engine step failed")

@@ -64,7 +64,8 @@ def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
             else:
                 model = Shard(config, tp_rank=0, tp_size=1, device=device)
             eng = InferenceEngine(model, **engine_kw)
-            eng.start()
+            if autostart:
+                eng.start()
             _ENGINES[key] = eng
     return eng
 
