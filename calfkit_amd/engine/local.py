@@ -29,7 +29,7 @@ _ENGINES_LOCK = threading.Lock()
 
 
 def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
-                  **engine_kw: Any) -> InferenceEngine:
+                  autostart: bool = True, **engine_kw: Any) -> InferenceEngine:
     """One engine per (model, device) per process — all agents share it.
 
     ``tp`` > 1 shards the model across the torch.distributed world (every
@@ -75,11 +75,12 @@ class LocalModelClient:
 
     def __init__(self, model: Any = "llama-3-8b", *, device: str | None = None,
                  max_batch: int = 64, max_model_len: int | None = None,
-                 use_graphs: bool | None = None, kv_pages: int | None = None):
+                 use_graphs: bool | None = None, kv_pages: int | None = None,
+                 tp: int = 1):
         self._config = model
         self._device = device
         self._engine_kw = dict(max_batch=max_batch, max_model_len=max_model_len,
-                               use_graphs=use_graphs, kv_pages=kv_pages)
+                               use_graphs=use_graphs, kv_pages=kv_pages, tp=tp)
         self._engine: InferenceEngine | None = None
         self._tokenizer: SyntheticTokenizer | None = None
 
