@@ -90,6 +90,11 @@ class InferenceEngine:
         self.model = model
         self.cfg: ModelConfig = model.cfg
         self.device = model.device
+        # round up to a graph batch size so padding never outgrows buffers
+        for g in self.GRAPH_BATCHES:
+            if max_batch <= g:
+                max_batch = g
+                break
         self.max_batch = max_batch
         self.max_model_len = max_model_len or self.cfg.max_seq
         self.on_gpu = self.device.type == "cuda"
@@ -413,7 +418,7 @@ class InferenceEngine:
     def _pad_batch(self, b: int) -> int:
         for g in self.GRAPH_BATCHES:
             if b <= g:
-                return g
+                return min(g, self.max_batch)
         return self.max_batch
 
     def _run_len(self) -> int:

@@ -28,6 +28,25 @@ _ENGINES: dict[tuple, InferenceEngine] = {}
 _ENGINES_LOCK = threading.Lock()
 
 
+def shutdown_engines() -> None:
+    """Stop every shared engine (clean interpreter exit: a daemon scheduler
+    thread killed mid-HIP-call aborts with 'terminate called without an
+    active exception')."""
+    with _ENGINES_LOCK:
+        engines = list(_ENGINES.values())
+        _ENGINES.clear()
+    for eng in engines:
+        try:
+            eng.stop()
+        except Exception:
+            pass
+
+
+import atexit  # noqa: E402
+
+atexit.register(shutdown_engines)
+
+
 def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
                   autostart: bool = True, **engine_kw: Any) -> InferenceEngine:
     """One engine per (model, device) per process — all agents share it.
