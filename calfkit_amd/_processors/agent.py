@@ -76,11 +76,14 @@ class _BusSink(ChunkSink):
             running = asyncio.get_running_loop()
         except RuntimeError:
             running = None
-        if running is self._loop:
-            self._track(self._loop.create_task(self._send(chunk)))
-        else:
-            self._loop.call_soon_threadsafe(
-                lambda: self._track(self._loop.create_task(self._send(chunk))))
+        try:
+            if running is self._loop:
+                self._track(self._loop.create_task(self._send(chunk)))
+            else:
+                self._loop.call_soon_threadsafe(
+                    lambda: self._track(self._loop.create_task(self._send(chunk))))
+        except RuntimeError:
+            pass  # loop already closed (process teardown) — drop the chunk
 
     def _track(self, t: asyncio.Task) -> None:
         self._tasks.add(t)
