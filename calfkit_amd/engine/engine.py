@@ -77,7 +77,7 @@ class _Seq:
 
 class InferenceEngine:
     #: padded decode batch sizes that get their own captured graph
-    GRAPH_BATCHES = (1, 2, 4, 8, 16, 32, 64, 96, 128)
+    GRAPH_BATCHES = (1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256)
 
     def __init__(self, model: LlamaShard, *, max_batch: int = 64,
                  kv_pages: int | None = None, use_graphs: bool | None = None,
