@@ -338,6 +338,8 @@ class InferenceEngine:
         """Pack every waiting prompt into ONE varlen prefill forward."""
         if len(seqs) == 1 and len(seqs[0].request.prompt_ids) > PREFILL_CHUNK:
             return self._prefill(seqs[0])
+        if self.on_gpu:  # rocprofv3-visible phase marker (roctx via nvtx shim)
+            torch.cuda.nvtx.range_push(f"calfkit.prefill[{len(seqs)}]")
         t0 = time.perf_counter()
         model = self.model
         dev = self.device
@@ -377,6 +379,8 @@ class InferenceEngine:
         if self.on_gpu:
             torch.cuda.synchronize(self.device)
         ms = (time.perf_counter() - t0) * 1e3
+        if self.on_gpu:
+            torch.cuda.nvtx.range_pop()
         self._stats["prefills"] += len(seqs)
         self._stats["prefill_ms"] += ms
         for s, n, tok in zip(seqs, lens, firsts):
@@ -511,6 +515,8 @@ class InferenceEngine:
         B = len(self._running)
         pad = self._pad_batch(B)
         M = self._run_len()
+        if self.on_gpu:
+            torch.cuda.nvtx.range_push(f"calfkit.decode[{B}x{M}]")
         self._write_run_state(pad, M)
         if self.use_graphs and (pad, M) in self._graphs:
             self._graphs[(pad, M)].replay()
@@ -518,6 +524,8 @@ class InferenceEngine:
         else:
             self._decode_run_gpu(pad, M)
         sampled = self._sampled[:M, :B].cpu()  # the ONE sync per run
+        if self.on_gpu:
+            torch.cuda.nvtx.range_pop()
         dt = (time.perf_counter() - t0) * 1e3
         self._stats["decode_steps"] += M
         self._stats["decode_tokens"] += M * B
