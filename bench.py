@@ -28,7 +28,7 @@ import torch
 
 PROMPT_TOKENS = int(os.environ.get("CALF_BENCH_PROMPT", "128"))
 DECODE_TOKENS = int(os.environ.get("CALF_BENCH_DECODE", "64"))
-TURNS_PER_STEP = int(os.environ.get("CALF_BENCH_TURNS", "128"))
+TURNS_PER_STEP = int(os.environ.get("CALF_BENCH_TURNS", "256"))
 
 
 def log(msg: str) -> None:

@@ -499,7 +499,10 @@ class InferenceEngine:
         """One hipGraph per (padded batch, run length), largest first."""
         torch.cuda.synchronize(self.device)
         for b in reversed([g for g in self.GRAPH_BATCHES if g <= self.max_batch]):
-            for m in RUN_STEPS:
+            # full-length runs for every pad; short tail runs (M < 8) only
+            # for small pads — large-batch tails are rare and run eager
+            run_lens = RUN_STEPS if b <= 32 else RUN_STEPS[:1]
+            for m in run_lens:
                 self._decode_run_gpu(b, m)  # warm up allocations
                 torch.cuda.synchronize(self.device)
                 g = torch.cuda.CUDAGraph()
