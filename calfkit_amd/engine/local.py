@@ -155,11 +155,17 @@ class LocalModelClient:
         fut = engine.submit(gen)
         result = await asyncio.wrap_future(fut)
         text = tok.decode(result.token_ids)
+        finish = result.finish_reason
+        for stop in s.stop:  # stop-sequence truncation (client side)
+            idx = text.find(stop)
+            if idx >= 0:
+                text = text[:idx]
+                finish = "stop"
         facts = ProviderFacts(
             model_name=engine.cfg.name,
             usage=Usage(prompt_tokens=result.prompt_tokens,
                         completion_tokens=len(result.token_ids)),
-            finish_reason=result.finish_reason,
+            finish_reason=finish,
             prefill_ms=result.prefill_ms,
             decode_ms=result.decode_ms,
         )
