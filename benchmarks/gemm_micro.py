@@ -18,6 +18,11 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import calfkit_amd.ops as ops  # noqa: E402
 
 SHAPES = [  # (name, M, K, N)
+    ("qkv_8b_m256", 256, 4096, 6144),
+    ("o_8b_m256", 256, 4096, 4096),
+    ("gateup_8b_m256", 256, 4096, 28672),
+    ("down_8b_m256", 256, 14336, 4096),
+    ("lmhead_8b_m256", 256, 4096, 128256),
     ("qkv_8b", 32, 4096, 6144),
     ("o_8b", 32, 4096, 4096),
     ("gateup_8b", 32, 4096, 28672),
@@ -57,7 +62,7 @@ def main() -> None:
         out_lib = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
         out_sk = torch.empty_like(out_lib)
         sk = ops.choose_splitk(N, K)
-        work = torch.empty(sk * 64 * N, dtype=torch.float32, device="cuda")
+        work = torch.empty(sk * 256 * N, dtype=torch.float32, device="cuda")
 
         def lib():
             torch.nn.functional.linear(x, w, out=out_lib)

@@ -128,22 +128,23 @@ class LlamaShard:
 
         big = os.environ.get("CALF_DECODE_BIG_SKINNY", "0") == "1"
         limit = 10 ** 18 if big else 42_000_000
-        return (self.device.type == "cuda" and T <= 64 and n_out % 64 == 0
-                and k_in % 128 == 0 and n_out * k_in <= limit)
+        return (self.device.type == "cuda" and T <= 256 and n_out % 64 == 0
+                and k_in % 128 == 0
+                and (T > 64 or n_out * k_in <= limit))
 
-    def _splitk_for(self, weight: torch.Tensor) -> int:
-        n_out = weight.shape[0]
-        sk = self._splitk_cache.get(n_out)
+    def _splitk_for(self, weight: torch.Tensor, m: int = 32) -> int:
+        key = (weight.shape[0], m > 64)
+        sk = self._splitk_cache.get(key)
         if sk is None:
-            sk = ops.choose_splitk(n_out, weight.shape[1])
-            self._splitk_cache[n_out] = sk
+            sk = ops.choose_splitk(weight.shape[0], weight.shape[1], m)
+            self._splitk_cache[key] = sk
         return sk
 
     def _proj(self, x: torch.Tensor, weight: torch.Tensor,
               decode: bool) -> torch.Tensor:
         n_out = weight.shape[0]
         if decode and self._use_skinny(x.shape[0], n_out, weight.shape[1]):
-            sk = self._splitk_for(weight)
+            sk = self._splitk_for(weight, x.shape[0])
             out = torch.empty(x.shape[0], n_out, dtype=x.dtype, device=x.device)
             ops.skinny_gemm(out, x, weight, self._gemm_work, sk)
             return out
@@ -172,7 +173,7 @@ class LlamaShard:
         if decode and self._use_skinny(T, qkv_w.shape[0], qkv_w.shape[1]):
             # slab-fused path: split-K slabs -> {reduce + RoPE + cache write}
             # in the consumer kernel (no reduce_slabs pass, no re-read)
-            sk = self._splitk_for(qkv_w)
+            sk = self._splitk_for(qkv_w, T)
             Mp = ops.skinny_gemm_slabs(hidden, qkv_w, self._gemm_work, sk)
             qkv = torch.empty(T, qkv_w.shape[0], dtype=hidden.dtype,
                               device=hidden.device)
@@ -213,7 +214,7 @@ class LlamaShard:
                 and self._use_skinny(T, lw.wo.shape[0], lw.wo.shape[1])):
             # leave the o-projection as slabs; the following residual norm
             # fuses the split-K reduce
-            sk = self._splitk_for(lw.wo)
+            sk = self._splitk_for(lw.wo, T)
             Mp = ops.skinny_gemm_slabs(o2, lw.wo, self._gemm_work, sk)
             return ("slabs", Mp, sk)
         out = self._proj(o2, lw.wo, decode=decode)
@@ -225,7 +226,7 @@ class LlamaShard:
         act = torch.empty(T, self.ffn_s, dtype=hidden.dtype,
                           device=hidden.device)
         if decode and self._use_skinny(T, 2 * self.ffn_s, self.cfg.hidden):
-            sk = ops.choose_splitk(2 * self.ffn_s, self.cfg.hidden)
+            sk = ops.choose_splitk(2 * self.ffn_s, self.cfg.hidden, T)
             ops.skinny_gemm_silu(act, hidden, lw.wgateup, self._gemm_work, sk)
         else:
             gateup = torch.nn.functional.linear(hidden, lw.wgateup)
@@ -233,7 +234,7 @@ class LlamaShard:
         if (decode and self.tp_size == 1
                 and self._use_skinny(T, lw.wdown.shape[0], lw.wdown.shape[1])):
             # down stays as slabs; the NEXT residual norm fuses the reduce
-            sk = self._splitk_for(lw.wdown)
+            sk = self._splitk_for(lw.wdown, T)
             Mp = ops.skinny_gemm_slabs(act, lw.wdown, self._gemm_work2, sk)
             return ("slabs2", Mp, sk)
         down = self._proj(act, lw.wdown, decode=decode)

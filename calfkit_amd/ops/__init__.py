@@ -283,13 +283,21 @@ def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
 
 # -------------------------------------------------------------- skinny GEMM
 
-def choose_splitk(N: int, K: int, target_blocks: int = 384) -> int:
-    """Split-K so (N/64)*splitk lands near 1-2x the 256 CUs, K dividing."""
+def choose_splitk(N: int, K: int, m: int = 32) -> int:
+    """Split-K so (N/64)*splitk lands near 1-2x the 256 CUs, K dividing.
+
+    The M>64 path doesn't wave-split K, so its divisibility needs only
+    K % (64*sk); its block target is ~256 (the guide's 1/2-1x CU rule)."""
+    target = 384 if m <= 64 else 256
     best = 1
     for sk in (1, 2, 4, 8):
-        if K % (64 * sk * 4) != 0 and K % (32 * sk * 4) != 0:
-            continue
-        if (N // 64) * sk <= target_blocks:
+        if m <= 64:
+            if K % (64 * sk * 4) != 0 and K % (32 * sk * 4) != 0:
+                continue
+        else:
+            if K % (64 * sk) != 0:
+                continue
+        if (N // 64) * sk <= target:
             best = sk
     return best
 
@@ -315,10 +323,17 @@ def skinny_gemm_silu(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     out.copy_((torch.nn.functional.silu(g) * u).to(out.dtype))
 
 
+def pad_m(M: int) -> int:
+    for p in (16, 32, 64, 128, 256):
+        if M <= p:
+            return p
+    raise ValueError(f"decode batch {M} > 256")
+
+
 def skinny_gemm_slabs(x: torch.Tensor, w: torch.Tensor, work: torch.Tensor,
                       splitk: int) -> int:
     """Run only the slab stage; returns Mp. Consumer fuses the reduce."""
-    Mp = 16 if x.shape[0] <= 16 else (32 if x.shape[0] <= 32 else 64)
+    Mp = pad_m(x.shape[0])
     if _native(x):
         _K.skinny_gemm_slabs(x, w, work, splitk)
         return Mp

@@ -353,3 +353,21 @@ def test_slab_fused_residual_norm_matches_unfused():
     assert torch.allclose(r1.float(), r2.float(), atol=3e-2, rtol=3e-2)
     assert torch.allclose(h1.float(), h2.float(), atol=3e-2, rtol=3e-2), \
         (h1.float() - h2.float()).abs().max()
+
+
+@pytest.mark.parametrize("M,K,N,sk", [
+    (256, 4096, 6144, 2),
+    (256, 14336, 4096, 4),
+    (200, 4096, 4096, 4),   # ragged M, pads to 256
+    (100, 4096, 1280, 2),   # pads to 128
+])
+def test_skinny_gemm_m256_vs_fp32(M, K, N, sk):
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / K ** 0.5)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    work = torch.empty(sk * 256 * N, dtype=torch.float32, device="cuda")
+    ops.skinny_gemm(out, x, w, work, sk)
+    ref = x.float() @ w.float().T
+    scale = ref.abs().max()
+    assert ((out.float() - ref).abs().max() / scale) < 2e-2, \
+        (out.float() - ref).abs().max()
