@@ -103,7 +103,7 @@ class LlamaShard:
         if self.device.type == "cuda":
             self._gemm_work = torch.empty(work_elems, dtype=torch.float32,
                                           device=self.device)
-            self._gemm_work2 = torch.empty(8 * 64 * d, dtype=torch.float32,
+            self._gemm_work2 = torch.empty(8 * 256 * d, dtype=torch.float32,
                                            device=self.device)
         else:
             self._gemm_work = torch.empty(0)
@@ -128,9 +128,11 @@ class LlamaShard:
 
         big = os.environ.get("CALF_DECODE_BIG_SKINNY", "0") == "1"
         limit = 10 ** 18 if big else 42_000_000
-        return (self.device.type == "cuda" and T <= 256 and n_out % 64 == 0
-                and k_in % 128 == 0
-                and (T > 64 or n_out * k_in <= limit))
+        # the M>64 path exists (skinny_gemm_m_kernel) but measured 0.2-0.4x
+        # of hipBLASLt (r14: the 8 waves re-read the shared W tile; it needs
+        # the guide's LDS W-staging before it can route) — library for now
+        return (self.device.type == "cuda" and T <= 64 and n_out % 64 == 0
+                and k_in % 128 == 0 and n_out * k_in <= limit)
 
     def _splitk_for(self, weight: torch.Tensor, m: int = 32) -> int:
         key = (weight.shape[0], m > 64)
