@@ -258,7 +258,11 @@ class LlamaShard:
             past_lens=torch.tensor([past_len], dtype=torch.int32, device=dev),
             kv_write_start=past_len,
             kv_total=past_len + T,
-            kv_in_qkv=(past_len == 0))
+            # chunked prefill MUST land every chunk in the scratch buffers:
+            # a past==0 first chunk that skipped the write would leave stale
+            # K/V for the later chunks (caught by
+            # test_long_prompt_chunked_prefill_matches_short_path)
+            kv_in_qkv=False)
 
     def forward(self, tokens: torch.Tensor, positions: torch.Tensor,
                 kcaches: list[torch.Tensor], vcaches: list[torch.Tensor],

@@ -120,3 +120,31 @@ def test_model_forward_prefill_decode_consistency():
         (logits_a.float() - logits_b.float()).abs().max()
     # and the argmax (greedy next token) agrees
     assert int(logits_a.argmax()) == int(logits_b.argmax())
+
+
+def test_long_prompt_chunked_prefill_matches_short_path():
+    """A prompt longer than PREFILL_CHUNK goes through the chunked solo
+    path; its continuation must match an equivalent packed prefill."""
+    import calfkit_amd.engine.engine as eng_mod
+
+    old_chunk, old_budget = eng_mod.PREFILL_CHUNK, eng_mod.PREFILL_BUDGET
+    eng_mod.PREFILL_CHUNK, eng_mod.PREFILL_BUDGET = 16, 48
+    try:
+        eng = make_engine(max_model_len=256)
+        try:
+            prompt = [1] + [(7 * i) % 200 + 10 for i in range(60)]  # 61 > 48
+            r_long = eng.generate(GenRequest(prompt_ids=prompt,
+                                             max_new_tokens=4), timeout=120)
+        finally:
+            eng.stop()
+        # same prompt through the packed path (budget large enough)
+        eng_mod.PREFILL_CHUNK, eng_mod.PREFILL_BUDGET = 512, 8192
+        eng2 = make_engine(max_model_len=256)
+        try:
+            r_pack = eng2.generate(GenRequest(prompt_ids=prompt,
+                                              max_new_tokens=4), timeout=120)
+        finally:
+            eng2.stop()
+        assert r_long.token_ids == r_pack.token_ids
+    finally:
+        eng_mod.PREFILL_CHUNK, eng_mod.PREFILL_BUDGET = old_chunk, old_budget
