@@ -335,7 +335,12 @@ class InferenceEngine:
             self._running.append(seq)
 
     def _prefill_batch(self, seqs: list["_Seq"]) -> None:
-        """Pack every waiting prompt into ONE varlen prefill forward."""
+        """Pack every waiting prompt into ONE varlen prefill forward.
+
+        Prefix-cached prompts prefill only their uncached SUFFIX: the cached
+        pages are gathered into the contiguous scratch as the per-sequence
+        past, and the new rows scatter into their packed positions.
+        """
         if len(seqs) == 1 and len(seqs[0].request.prompt_ids) > PREFILL_CHUNK:
             return self._prefill(seqs[0])
         if self.on_gpu:  # rocprofv3-visible phase marker (roctx via nvtx shim)
@@ -343,31 +348,48 @@ class InferenceEngine:
         t0 = time.perf_counter()
         model = self.model
         dev = self.device
-        lens = [len(s.request.prompt_ids) for s in seqs]
-        total = sum(lens)
+        cached: list[int] = []
+        suffixes: list[list[int]] = []
+        for s in seqs:
+            c = self.kv.match_prefix(s.seq_id, s.request.prompt_ids)
+            cached.append(c)
+            suffixes.append(s.request.prompt_ids[c:])
+        lens = [len(sx) for sx in suffixes]
         ids: list[int] = []
         positions: list[int] = []
         slots: list[int] = []
-        starts = [0]
-        for s, n in zip(seqs, lens):
-            self.kv.open(s.seq_id)
-            ids.extend(s.request.prompt_ids)
-            positions.extend(range(n))
-            slots.extend(self.kv.slots_for(s.seq_id, n))
-            starts.append(starts[-1] + n)
+        q_starts_l = [0]
+        kv_starts_l = [0]
+        write_index: list[int] = []
+        for s, sx, c in zip(seqs, suffixes, cached):
+            ids.extend(sx)
+            positions.extend(range(c, c + len(sx)))
+            slots.extend(self.kv.slots_for(s.seq_id, len(sx)))
+            write_index.extend(range(kv_starts_l[-1] + c,
+                                     kv_starts_l[-1] + c + len(sx)))
+            q_starts_l.append(q_starts_l[-1] + len(sx))
+            kv_starts_l.append(kv_starts_l[-1] + c + len(sx))
+        kv_total = kv_starts_l[-1]
+        any_cached = any(cached)
         from .. import ops
         from ..models.llama import PrefillPlan
 
-        q_starts = torch.as_tensor(starts, dtype=torch.int32).to(dev)
+        if any_cached:
+            self._gather_past(seqs, cached, kv_starts_l)
+        q_starts = torch.as_tensor(q_starts_l, dtype=torch.int32).to(dev)
+        kv_starts = torch.as_tensor(kv_starts_l, dtype=torch.int32).to(dev)
         plan = PrefillPlan(
             scratch=self._prefill_kv,
             block_map=ops.build_prefill_blocks(lens).to(dev),
             q_starts=q_starts,
-            kv_starts=q_starts,
-            past_lens=torch.zeros(len(seqs), dtype=torch.int32, device=dev),
+            kv_starts=kv_starts,
+            past_lens=torch.as_tensor(cached, dtype=torch.int32).to(dev),
             kv_write_start=0,
-            kv_total=total,
-            kv_in_qkv=True)
+            kv_total=kv_total,
+            kv_in_qkv=not any_cached,
+            kv_write_index=(torch.as_tensor(write_index,
+                                            dtype=torch.int64).to(dev)
+                            if any_cached else None))
         logits = model.forward(
             torch.as_tensor(ids, dtype=torch.int64).to(dev),
             torch.as_tensor(positions, dtype=torch.int32).to(dev),
@@ -375,6 +397,8 @@ class InferenceEngine:
             torch.as_tensor(slots, dtype=torch.int32).to(dev),
             prefill_plan=plan,
             logits_rows=(q_starts[1:].long() - 1))
+        for s in seqs:
+            self.kv.register_prefix(s.seq_id, s.request.prompt_ids)
         firsts = self._sample_batch_host(logits, seqs)
         if self.on_gpu:
             torch.cuda.synchronize(self.device)
@@ -383,12 +407,32 @@ class InferenceEngine:
             torch.cuda.nvtx.range_pop()
         self._stats["prefills"] += len(seqs)
         self._stats["prefill_ms"] += ms
+        total_new = sum(lens)
         for s, n, tok in zip(seqs, lens, firsts):
-            s.result.prompt_tokens = n
-            s.result.prefill_ms += ms * n / max(1, total)  # attribute by share
+            s.result.prompt_tokens = len(s.request.prompt_ids)
+            s.result.prefill_ms += ms * n / max(1, total_new)  # by share
             self._emit_token(s, tok)
             if not self._finish_if_done(s):
                 self._running.append(s)
+
+    def _gather_past(self, seqs, cached, kv_starts_l) -> None:
+        """Copy each sequence's cached pages into the prefill scratch."""
+        from .kvcache import PAGE_SIZE as PG
+
+        for layer in range(self.cfg.layers):
+            kbuf, vbuf = self._prefill_kv[layer]
+            kc, vc = self.kv.kcache[layer], self.kv.vcache[layer]
+            for s, c, start in zip(seqs, cached, kv_starts_l[:-1]):
+                if c == 0:
+                    continue
+                pages = self.kv._slots[s.seq_id].pages[: c // PG]
+                idx = torch.as_tensor(pages, dtype=torch.long,
+                                      device=kc.device)
+                # [P, hkv, PG, dh] -> [P*PG, hkv, dh]
+                kbuf[start:start + c] = kc[idx].permute(0, 2, 1, 3).reshape(
+                    -1, self.model.hkv, self.model.dh)
+                vbuf[start:start + c] = vc[idx].permute(0, 2, 1, 3).reshape(
+                    -1, self.model.hkv, self.model.dh)
 
     def _sample_batch_host(self, logits: torch.Tensor,
                            seqs: list["_Seq"]) -> list[int]:

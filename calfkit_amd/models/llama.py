@@ -39,6 +39,9 @@ class PrefillPlan:
     past_lens: torch.Tensor   # [B] i32
     kv_write_start: int       # where this forward's new kv rows land
     kv_total: int             # total kv rows the attention reads
+    #: scattered landing spots for the new kv rows (packed multi-seq prefill
+    #: with per-seq cached pasts); None = contiguous at kv_write_start
+    kv_write_index: "torch.Tensor | None" = None
     #: True when every past_len is 0 and kv rows == q rows: attention reads
     #: K/V straight from the packed qkv GEMM output (no scratch copy at all)
     kv_in_qkv: bool = False
@@ -199,12 +202,18 @@ class LlamaShard:
                 vv = vv.view(T, self.hkv, self.dh)
             else:
                 kfull, vfull = prefill_kv
-                w0 = plan.kv_write_start
-                kfull[w0:w0 + T] = qkv[:, self.hq * self.dh:
-                                       (self.hq + self.hkv) * self.dh].view(
-                                           T, self.hkv, self.dh)
-                vfull[w0:w0 + T] = qkv[:, (self.hq + self.hkv) * self.dh:].view(
+                knew = qkv[:, self.hq * self.dh:
+                           (self.hq + self.hkv) * self.dh].view(
+                               T, self.hkv, self.dh)
+                vnew = qkv[:, (self.hq + self.hkv) * self.dh:].view(
                     T, self.hkv, self.dh)
+                if plan.kv_write_index is not None:
+                    kfull.index_copy_(0, plan.kv_write_index, knew)
+                    vfull.index_copy_(0, plan.kv_write_index, vnew)
+                else:
+                    w0 = plan.kv_write_start
+                    kfull[w0:w0 + T] = knew
+                    vfull[w0:w0 + T] = vnew
                 kk, vv = kfull[:plan.kv_total], vfull[:plan.kv_total]
             ops.attn_prefill_varlen(
                 o, q, kk, vv, plan.block_map, plan.q_starts, plan.kv_starts,
