@@ -17,7 +17,7 @@ nothing else escapes an agent hop.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Union
+from typing import Union
 
 from .messages import Message, TextPart, ToolCallPart, UserMessage
 from .ports import Advert

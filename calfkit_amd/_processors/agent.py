@@ -11,7 +11,6 @@ aggregation per /root/reference/CHANGELOG.md:192,196.
 from __future__ import annotations
 
 import asyncio
-import uuid
 from typing import Any
 
 from .._capabilities import Dispatch, Transfer
@@ -28,21 +27,17 @@ from .._wire.deliveries import (
     EchoMarker,
     Fault,
     ReplySlot,
-    ToolInvoke,
 )
 from .._wire.machinery import NodeMachinery
 from .._wire.publisher import FaultFunnel
 from .._wire import topology
 from ..context import TurnContext
 from ..definitions.agent import Agent
-from ..errors import DecodeError, ErrorReport
+from ..errors import DecodeError
 from ..messages import (
     AssistantMessage,
-    Message,
-    TextPart,
     ToolMessage,
     ToolReturnPart,
-    UserMessage,
     parts_to_text,
 )
 from .._projection import project_for_model, project_for_transfer

@@ -23,7 +23,6 @@ import asyncio
 import json
 import logging
 import time
-from collections import defaultdict
 from typing import Any, AsyncIterator
 
 from ...ports import Advert, DeliveryHandler, DirectoryView, GrainKey, RawDelivery

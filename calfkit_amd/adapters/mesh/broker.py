@@ -21,14 +21,7 @@ from collections import defaultdict
 from dataclasses import dataclass, field
 from typing import Any
 
-from .protocol import (
-    b64d,
-    b64e,
-    encode_frame,
-    headers_from_wire,
-    headers_to_wire,
-    read_frame,
-)
+from .protocol import encode_frame, read_frame
 
 _log = logging.getLogger("calfkit.mesh.broker")
 

@@ -11,12 +11,10 @@ from __future__ import annotations
 
 import asyncio
 import importlib.util
-import json
 import os
 import signal
 import subprocess
 import sys
-import time
 from pathlib import Path
 
 import click

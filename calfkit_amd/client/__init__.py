@@ -15,7 +15,7 @@ from __future__ import annotations
 import asyncio
 import json
 import uuid
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, AsyncIterator
 
 from .._wire import topology
@@ -26,7 +26,7 @@ from .._wire.deliveries import (
     agent_call_payload,
     encode_envelope,
 )
-from ..definitions.tools import schema_for_type, _coerce
+from ..definitions.tools import _coerce
 from ..errors import NodeFaultError, ErrorReport, TimeoutError_
 from ..messages import (
     AssistantMessage,

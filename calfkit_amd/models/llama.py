@@ -22,7 +22,7 @@ from dataclasses import dataclass
 import torch
 
 from .. import ops
-from ..parallel import all_reduce_sum, shard_dim
+from ..parallel import all_reduce_sum
 from .config import ModelConfig
 
 __all__ = ["LlamaShard", "LayerWeights", "PrefillPlan"]

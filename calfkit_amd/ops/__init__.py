@@ -11,7 +11,6 @@ Dispatch contract:
 
 from __future__ import annotations
 
-import math
 from typing import Any
 
 import torch

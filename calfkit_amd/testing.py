@@ -8,14 +8,12 @@ Capability parity: reference ``testing.py``
 
 from __future__ import annotations
 
-import json
 import uuid
 from typing import Any, Callable, Sequence
 
 from .mcp import MCPResult, MCPToolInfo
 from .messages import (
     AssistantMessage,
-    Part,
     ProviderFacts,
     TextPart,
     ToolCallPart,
