@@ -113,11 +113,15 @@ class InferenceEngine:
                           device=self.device)
         self.max_pages = (self.max_model_len + PAGE_SIZE - 1) // PAGE_SIZE
 
-        # prefill scratch KV (contiguous [max_ctx, hkv, dh] per layer)
+        # prefill scratch KV (contiguous [rows, hkv, dh] per layer).  The
+        # packed varlen prefill with prefix-cache gathering lands the WHOLE
+        # batch (pasts + new rows) here, so it must hold the full token
+        # budget, not just one sequence
+        scratch_rows = max(self.max_model_len, PREFILL_BUDGET)
         self._prefill_kv = [
-            (torch.zeros(self.max_model_len, model.hkv, model.dh,
+            (torch.zeros(scratch_rows, model.hkv, model.dh,
                          dtype=model.dtype, device=self.device),
-             torch.zeros(self.max_model_len, model.hkv, model.dh,
+             torch.zeros(scratch_rows, model.hkv, model.dh,
                          dtype=model.dtype, device=self.device))
             for _ in range(self.cfg.layers)]
 

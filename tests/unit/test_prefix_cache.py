@@ -102,3 +102,24 @@ async def test_multi_turn_agent_thread_reuses_prefix():
         await client.execute("cached", "turn two", thread_id="T", timeout=120)
         # the second turn's prompt (same system prefix) hit the cache
         assert kv.prefix_hit_tokens >= 128
+
+
+def test_identical_prompts_across_packed_batches():
+    """Identical prompts (the synthetic tokenizer can legitimately produce
+    them) share pages across packed batches; outputs stay identical.
+    Regression for the r19 scratch-overflow crash."""
+    eng = make_engine(max_model_len=256)
+    try:
+        p = [1] + [20 + i % 40 for i in range(140)]
+        futs = [eng.submit(GenRequest(prompt_ids=p, max_new_tokens=3))
+                for _ in range(3)]
+        outs = [f.result(60).token_ids for f in futs]
+        assert outs[0] == outs[1] == outs[2]
+        futs = [eng.submit(GenRequest(prompt_ids=p, max_new_tokens=3)),
+                eng.submit(GenRequest(prompt_ids=[1, 2, 3, 4],
+                                      max_new_tokens=3))]
+        outs2 = [f.result(60).token_ids for f in futs]
+        assert outs2[0] == outs[0]
+        assert eng.kv.prefix_hit_tokens >= 128
+    finally:
+        eng.stop()
