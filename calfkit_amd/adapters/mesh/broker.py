@@ -41,6 +41,8 @@ class _Conn:
     id: int
     alive: bool = True
     subs: list[_Subscription] = field(default_factory=list)
+    #: nodes this connection advertised (liveness: they die with it)
+    adverts: set[str] = field(default_factory=set)
 
     def send(self, obj: dict[str, Any]) -> None:
         if not self.alive:
@@ -117,6 +119,13 @@ class MeshBroker:
             self._assign.pop((sub.topic, sub.group), None)
         if conn in self._watchers:
             self._watchers.remove(conn)
+        # liveness: a dead worker's adverts leave the directory with it
+        dropped = False
+        for node in conn.adverts:
+            if self._adverts.pop(node, None) is not None:
+                dropped = True
+        if dropped:
+            self._push_directory()
 
     async def _handle(self, conn: _Conn, f: dict[str, Any]) -> None:
         op = f.get("op")
@@ -135,6 +144,7 @@ class MeshBroker:
             advert = dict(f["advert"])
             advert["ts"] = time.time()
             self._adverts[advert["node"]] = advert
+            conn.adverts.add(advert["node"])
             self._push_directory()
         elif op == "withdraw":
             self._adverts.pop(f.get("node", ""), None)

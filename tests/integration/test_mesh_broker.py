@@ -94,3 +94,23 @@ async def test_consumer_group_replicas_share_by_key():
         assert not overlap, (seen, overlap)
     finally:
         await broker.stop()
+
+
+async def test_dead_worker_adverts_leave_the_directory():
+    """Liveness: a worker connection dying withdraws its adverts."""
+    broker = await MeshBroker().start()
+    try:
+        bus_w, dir_w = await connect_mesh(broker.address)
+        agent = Agent(name="mortal", model_client=FakeProvider(["hi"]))
+        worker = Worker(agent, bus=bus_w, store=InProcessThreadStore(),
+                        directory=dir_w)
+        await worker.start()
+        bus_c, dir_c = await connect_mesh(broker.address)
+        client = await Client.connect(bus_c, dir_c)
+        assert (await client.mesh()).get("mortal") is not None
+        # kill the worker's CONNECTION without a clean withdraw
+        await bus_w.close()
+        await asyncio.sleep(0.2)
+        assert (await client.mesh()).get("mortal") is None
+    finally:
+        await broker.stop()
