@@ -214,11 +214,29 @@ def dev_chat(mesh_url: str, agent_name: str) -> None:
                 break
             if not line.strip():
                 continue
-            result = await client.execute(target, line, thread_id=thread_id,
-                                          timeout=600)
-            thread_id = result.thread_id  # handoffs keep the thread id, so
-            # the next turn follows the conversation wherever it moved
-            click.echo(result.text)
+            tid = thread_id or __import__("uuid").uuid4().hex
+
+            async def stream() -> None:
+                # live step events while the turn runs (text deltas, tool
+                # calls, handoffs) — the reference's wait-progress
+                async for chunk in client.firehose(tid):
+                    if chunk.event == "text_delta":
+                        click.echo(chunk.text, nl=False)
+                    elif chunk.event == "tool_call":
+                        click.echo(f"\n[{chunk.node} -> "
+                                   f"{chunk.data.get('tool_name')}"
+                                   f"({chunk.data.get('arguments')})]")
+                    elif chunk.event == "handoff":
+                        click.echo(f"\n[handoff -> {chunk.data.get('to')}]")
+
+            watcher = asyncio.ensure_future(stream())
+            try:
+                result = await client.execute(target, line, thread_id=tid,
+                                              timeout=600)
+            finally:
+                watcher.cancel()
+            thread_id = result.thread_id or tid  # handoffs keep the thread
+            click.echo("")  # newline after the streamed text
 
     asyncio.run(run())
 
