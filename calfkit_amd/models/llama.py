@@ -79,12 +79,16 @@ class LlamaShard:
         self.dh = cfg.head_dim
         self.scale = 1.0 / math.sqrt(self.dh)
 
-        gen = torch.Generator(device="cpu").manual_seed(seed + tp_rank)
+        # random-init directly on the device: CPU-side randn of 8-70B params
+        # costs minutes; on-GPU generation is seconds (seeded per TP rank)
+        gen_device = "cpu" if self.device.type == "cpu" else self.device
+        gen = torch.Generator(device=gen_device).manual_seed(seed + tp_rank)
         d = cfg.hidden
 
         def w(rows: int, cols: int, fan_in: int) -> torch.Tensor:
             std = 1.0 / math.sqrt(fan_in)
-            t = torch.randn(rows, cols, generator=gen, dtype=torch.float32) * std
+            t = torch.randn(rows, cols, generator=gen, dtype=torch.float32,
+                            device=gen_device) * std
             return t.to(device=self.device, dtype=dtype)
 
         self.embed = w(cfg.vocab_size, d, d)

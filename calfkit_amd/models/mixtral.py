@@ -41,12 +41,14 @@ class MixtralShard(LlamaShard):
         assert cfg.is_moe, "MixtralShard needs an MoE config"
         super().__init__(cfg, **kw)
         # replace the dense MLP weights with stacked expert weights + router
-        gen = torch.Generator(device="cpu").manual_seed(4321 + self.tp_rank)
+        gen_device = "cpu" if self.device.type == "cpu" else self.device
+        gen = torch.Generator(device=gen_device).manual_seed(4321 + self.tp_rank)
         d, e = cfg.hidden, cfg.num_experts
 
         def w(*shape: int, fan_in: int) -> torch.Tensor:
             std = 1.0 / math.sqrt(fan_in)
-            t = torch.randn(*shape, generator=gen, dtype=torch.float32) * std
+            t = torch.randn(*shape, generator=gen, dtype=torch.float32,
+                            device=gen_device) * std
             return t.to(device=self.device, dtype=self.dtype)
 
         layers: list[MoeLayerWeights] = []
