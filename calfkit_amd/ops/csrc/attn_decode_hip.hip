@@ -23,7 +23,7 @@
 #define HD 128
 #define NWAVES 4
 
-template <int G>
+template <int G, bool NT>
 __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
     __hip_bfloat16* __restrict__ out,
     const __hip_bfloat16* __restrict__ q,
@@ -88,7 +88,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
           kcache + base + (live ? lane : 0) * HD);
       bf16x8 kreg[HD / 8];
 #pragma unroll
-      for (int i = 0; i < HD / 8; ++i) kreg[i] = krow[i];
+      for (int i = 0; i < HD / 8; ++i)
+        kreg[i] = NT ? __builtin_nontemporal_load(krow + i) : krow[i];
 #pragma unroll
       for (int i = 0; i < HD / 8; ++i) {
 #pragma unroll
@@ -138,7 +139,9 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
       float2 vA[8], vB[8];
 #define PV_LOAD(BUF, C)                                                     \
   _Pragma("unroll") for (int j = 0; j < 8; ++j)                             \
-      BUF[j] = vbase[(t0 + min((C) * 8 + j, nt - 1)) * (HD / 4)];
+      BUF[j] = NT ? __builtin_nontemporal_load(                             \
+                        vbase + (t0 + min((C) * 8 + j, nt - 1)) * (HD / 4)) \
+                  : vbase[(t0 + min((C) * 8 + j, nt - 1)) * (HD / 4)];
 #define PV_FMA(BUF, C)                                                      \
   _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
     const int t = (C) * 8 + j;                                              \
@@ -218,6 +221,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
 
 void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                  torch::Tensor vcache, torch::Tensor page_table,
@@ -237,8 +241,15 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   const int G = hq / hkv;
   dim3 grid(B, hkv), block(NWAVES * WAVE);
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  const char* nt_env = getenv("CALF_ATTN_NT");
+  const bool use_nt = nt_env && nt_env[0] == '1';
 #define LAUNCH(GV)                                                        \
-  hipLaunchKernelGGL(attn_decode4_kernel<GV>, grid, block, 0, s,          \
+  do {                                                                    \
+    if (use_nt) LAUNCH_NT(GV, true);                                      \
+    else LAUNCH_NT(GV, false);                                            \
+  } while (0)
+#define LAUNCH_NT(GV, NTV)                                                \
+  hipLaunchKernelGGL((attn_decode4_kernel<GV, NTV>), grid, block, 0, s,   \
                      (__hip_bfloat16*)out.data_ptr(),                     \
                      (const __hip_bfloat16*)q.data_ptr(),                 \
                      (const __hip_bfloat16*)kcache.data_ptr(),            \
@@ -253,6 +264,7 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
     case 8: LAUNCH(8); break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
+#undef LAUNCH_NT
 #undef LAUNCH
   HIP_CHECK_LAST();
 }
