@@ -22,7 +22,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
-TURNS_PER_STEP = 16
+TURNS_PER_STEP = int(os.environ.get("C3_TURNS", "48"))
 STEPS = int(os.environ.get("C3_STEPS", "3"))
 WARMUP = int(os.environ.get("C3_WARMUP", "1"))
 
