@@ -188,6 +188,8 @@ class InferenceEngine:
         with self._lock:
             self._next_id += 1
             sid = self._next_id
+        request.max_new_tokens = min(request.max_new_tokens,
+                                     self.max_model_len - 1)
         if len(request.prompt_ids) + request.max_new_tokens > self.max_model_len:
             request.prompt_ids = request.prompt_ids[
                 -(self.max_model_len - request.max_new_tokens):]
