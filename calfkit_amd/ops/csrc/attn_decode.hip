@@ -22,7 +22,7 @@
 #define HD 128
 #define NWAVES 4
 
-template <int G, bool NT>
+template <int G>
 __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
     __hip_bfloat16* __restrict__ out,
     const __hip_bfloat16* __restrict__ q,
@@ -87,8 +87,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
           kcache + base + (live ? lane : 0) * HD);
       bf16x8 kreg[HD / 8];
 #pragma unroll
-      for (int i = 0; i < HD / 8; ++i)
-        kreg[i] = NT ? __builtin_nontemporal_load(krow + i) : krow[i];
+      for (int i = 0; i < HD / 8; ++i) kreg[i] = krow[i];
 #pragma unroll
       for (int i = 0; i < HD / 8; ++i) {
 #pragma unroll
@@ -138,9 +137,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
       float2 vA[8], vB[8];
 #define PV_LOAD(BUF, C)                                                     \
   _Pragma("unroll") for (int j = 0; j < 8; ++j)                             \
-      BUF[j] = NT ? __builtin_nontemporal_load(                             \
-                        vbase + (t0 + min((C) * 8 + j, nt - 1)) * (HD / 4)) \
-                  : vbase[(t0 + min((C) * 8 + j, nt - 1)) * (HD / 4)];
+      BUF[j] = vbase[(t0 + min((C) * 8 + j, nt - 1)) * (HD / 4)];
 #define PV_FMA(BUF, C)                                                      \
   _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
     const int t = (C) * 8 + j;                                              \
@@ -220,7 +217,6 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
-#include <cstdlib>
 
 void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                  torch::Tensor vcache, torch::Tensor page_table,
@@ -240,15 +236,8 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   const int G = hq / hkv;
   dim3 grid(B, hkv), block(NWAVES * WAVE);
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
-  const char* nt_env = getenv("CALF_ATTN_NT");
-  const bool use_nt = nt_env && nt_env[0] == '1';
 #define LAUNCH(GV)                                                        \
-  do {                                                                    \
-    if (use_nt) LAUNCH_NT(GV, true);                                      \
-    else LAUNCH_NT(GV, false);                                            \
-  } while (0)
-#define LAUNCH_NT(GV, NTV)                                                \
-  hipLaunchKernelGGL((attn_decode4_kernel<GV, NTV>), grid, block, 0, s,   \
+  hipLaunchKernelGGL(attn_decode4_kernel<GV>, grid, block, 0, s,          \
                      (__hip_bfloat16*)out.data_ptr(),                     \
                      (const __hip_bfloat16*)q.data_ptr(),                 \
                      (const __hip_bfloat16*)kcache.data_ptr(),            \
@@ -263,7 +252,6 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
     case 8: LAUNCH(8); break;
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
-#undef LAUNCH_NT
 #undef LAUNCH
   HIP_CHECK_LAST();
 }
