@@ -99,8 +99,14 @@ class InferenceEngine:
         self.max_model_len = max_model_len or self.cfg.max_seq
         self.on_gpu = self.device.type == "cuda"
         if use_graphs is None:
-            # MoE's bucketing has a host sync per layer — not capturable yet
-            use_graphs = self.on_gpu and not self.cfg.is_moe
+            # MoE decode is capturable through the grouped kernels
+            # (ops/csrc/moe_gemm.hip): buckets are built device-side, so
+            # there is no host sync.  The host-grouped fallback (no
+            # extension) is not capturable — the capture try/except below
+            # would catch it, but don't even try.
+            from .. import ops as _ops
+            use_graphs = self.on_gpu and (
+                not self.cfg.is_moe or _ops.have_extension())
         self.use_graphs = use_graphs
 
         if kv_pages is None:

@@ -3,12 +3,18 @@
 Attention is identical to Llama (GQA, dh=128, paged KV) and is inherited;
 the MLP becomes a top-2 router + 8 SwiGLU experts.
 
-Expert compute (round-1 shape): tokens are bucketed per expert on-device
-(argsort over the flat (token, expert) assignment), each expert runs one
-dense gate-up/down GEMM pair over its bucket (hipBLASLt), and results
-scatter-add back with the router weights.  This is the grouped-GEMM
-decomposition with the group loop on the host; a single fused grouped-GEMM
-kernel is the planned next step (SURVEY §2.4 row "Grouped GEMM (MoE)").
+Expert compute:
+- DECODE (T <= 256, GPU): the grouped CDNA4 kernel pair
+  (ops/csrc/moe_gemm.hip) — buckets built device-side (argsort + cumsum,
+  no host sync), token gather fused into the gate-up GEMM's A-fragment
+  loads, router scale fused into the down GEMM's epilogue.  The whole MoE
+  layer is hipGraph-capturable, so Mixtral decode runs under the engine's
+  multi-step graphs exactly like the dense models (SURVEY §2.4 row
+  "Grouped GEMM (MoE)").
+- PREFILL / CPU: tokens bucketed with a host-side count, each expert runs
+  one dense gate-up/down GEMM pair over its bucket (hipBLASLt) — the
+  compute-bound regime where library tiles win and the one-time host sync
+  amortizes over the whole chunk.
 
 TP: every expert's ffn shards over the TP group exactly like the dense MLP
 (row/column parallel + one all-reduce); the router is replicated.  Expert
@@ -62,6 +68,14 @@ class MixtralShard(LlamaShard):
             ))
         self.layers = layers
         self.topk = cfg.experts_per_token
+        # fp32 gate-up slab for the grouped decode kernel: R = 256 (max
+        # decode batch) * topk rows of 2F columns (~59 MB for 8x7B)
+        if self.device.type == "cuda":
+            self._moe_work = torch.empty(
+                256 * self.topk * 2 * self.ffn_s, dtype=torch.float32,
+                device=self.device)
+        else:
+            self._moe_work = torch.empty(0)
 
     def _mlp(self, lw: MoeLayerWeights, hidden: torch.Tensor, *,
              decode: bool = False) -> torch.Tensor:
@@ -71,6 +85,23 @@ class MixtralShard(LlamaShard):
         logits = torch.nn.functional.linear(hidden, lw.router).float()  # [T,E]
         weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
         weights = weights / weights.sum(-1, keepdim=True)   # renormalize top-k
+
+        if decode and self.device.type == "cuda" and ops.have_extension() \
+                and T <= 256:
+            # grouped CDNA4 path: no host sync, hipGraph-capturable
+            tok, off, scale = ops.moe_buckets(experts, weights, e)
+            R = T * k
+            act = torch.empty(R, self.ffn_s, dtype=hidden.dtype,
+                              device=hidden.device)
+            ops.moe_gateup_silu(act, hidden, lw.wgateup, tok, off,
+                                self._moe_work)
+            y = torch.empty(R, d, dtype=hidden.dtype, device=hidden.device)
+            ops.moe_down(y, act, lw.wdown, off, scale)
+            out32 = torch.zeros(T, d, dtype=torch.float32,
+                                device=hidden.device)
+            out32.index_add_(0, tok.long(), y.float())  # k rows per token
+            out = out32.to(hidden.dtype)
+            return all_reduce_sum(out) if self.tp_size > 1 else out
 
         flat_expert = experts.reshape(-1)                   # [T*k]
         flat_token = (torch.arange(T, device=hidden.device)

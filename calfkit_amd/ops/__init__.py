@@ -33,6 +33,9 @@ __all__ = [
     "rope_qkv_write_slab",
     "fused_add_rmsnorm_slab",
     "skinny_gemm_silu",
+    "moe_gateup_silu",
+    "moe_down",
+    "moe_buckets",
     "choose_splitk",
     "build_rope_table",
 ]
@@ -320,6 +323,74 @@ def skinny_gemm_silu(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     y = x.float() @ w.float().T
     g, u = y[:, :F], y[:, F:]
     out.copy_((torch.nn.functional.silu(g) * u).to(out.dtype))
+
+
+# ------------------------------------------------------------- grouped MoE
+
+def moe_buckets(experts: torch.Tensor, weights: torch.Tensor,
+                num_experts: int) -> tuple[torch.Tensor, torch.Tensor,
+                                           torch.Tensor]:
+    """Device-side expert bucketing for the grouped MoE kernels.
+
+    experts/weights: [T, topk] router picks.  Returns (tok, off, scale):
+    tok[r] i32 source token for bucket row r (rows sorted by expert),
+    off[E+1] i32 bucket offsets, scale[r] f32 router weight.  Built from
+    sort/cumsum/gather only — no host sync, hipGraph-capturable (unlike
+    bincount().tolist() in the host-grouped fallback).
+    """
+    T, k = experts.shape
+    flat_expert = experts.reshape(-1)
+    order = torch.argsort(flat_expert, stable=True)
+    flat_token = torch.arange(T, device=experts.device,
+                              dtype=torch.int32).repeat_interleave(k)
+    counts = torch.zeros(num_experts, device=experts.device,
+                         dtype=torch.int32)
+    counts.scatter_add_(0, flat_expert.long(),
+                        torch.ones_like(flat_expert, dtype=torch.int32))
+    off = torch.zeros(num_experts + 1, device=experts.device,
+                      dtype=torch.int32)
+    torch.cumsum(counts, 0, out=off[1:])
+    tok = flat_token[order].contiguous()
+    scale = weights.reshape(-1)[order].float().contiguous()
+    return tok, off, scale
+
+
+def moe_gateup_silu(act: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                    tok: torch.Tensor, off: torch.Tensor,
+                    work: torch.Tensor) -> None:
+    """Grouped gate-up: act[r,:F] = silu(g)*u of x[tok[r]] @ w[e(r)]^T.
+
+    w: [E, 2F, K] stacked [gate rows | up rows]; rows r in [off[e], off[e+1])
+    belong to expert e; work: fp32 scratch >= R*2F.
+    """
+    if _native(x):
+        _K.moe_gateup_silu(act, x, w, tok, off, work)
+        return
+    F = act.shape[-1]
+    offs = off.tolist()
+    for e in range(w.shape[0]):
+        lo, hi = offs[e], offs[e + 1]
+        if lo == hi:
+            continue
+        xs = x[tok[lo:hi].long()].float()
+        y = xs @ w[e].float().T
+        g, u = y[:, :F], y[:, F:]
+        act[lo:hi] = (torch.nn.functional.silu(g) * u).to(act.dtype)
+
+
+def moe_down(y: torch.Tensor, act: torch.Tensor, w: torch.Tensor,
+             off: torch.Tensor, scale: torch.Tensor) -> None:
+    """Grouped down: y[r] = scale[r] * (act[r] @ w[e(r)]^T); w: [E, d, F]."""
+    if _native(act):
+        _K.moe_down(y, act, w, off, scale)
+        return
+    offs = off.tolist()
+    for e in range(w.shape[0]):
+        lo, hi = offs[e], offs[e + 1]
+        if lo == hi:
+            continue
+        out = act[lo:hi].float() @ w[e].float().T
+        y[lo:hi] = (out * scale[lo:hi, None].float()).to(y.dtype)
 
 
 def pad_m(M: int) -> int:

@@ -39,6 +39,10 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor work, int64_t splitk);
 void skinny_gemm_silu(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                       torch::Tensor work, int64_t splitk);
+void moe_gateup_silu(torch::Tensor act, torch::Tensor x, torch::Tensor w,
+                     torch::Tensor tok, torch::Tensor off, torch::Tensor work);
+void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w,
+              torch::Tensor off, torch::Tensor scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
@@ -64,4 +68,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "decode projection GEMM: out = x @ w^T, w stored [N,K]");
   m.def("skinny_gemm_silu", &skinny_gemm_silu,
         "fused SiLU-MLP gate: out = silu(x@Wg^T) * (x@Wu^T)");
+  m.def("moe_gateup_silu", &moe_gateup_silu,
+        "grouped MoE gate-up: act[r] = silu/mul of x[tok[r]] @ W[e(r)]^T");
+  m.def("moe_down", &moe_down,
+        "grouped MoE down: y[r] = scale[r] * act[r] @ W[e(r)]^T");
 }

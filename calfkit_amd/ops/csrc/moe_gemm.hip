@@ -1,0 +1,223 @@
+// Grouped MoE decode GEMM for MI355X (gfx950).
+//
+// Capability parity: the reference's MoE expert-parallel GEMM path
+// (SURVEY §2.4 "Grouped GEMM (MoE)") — here as a CDNA4 weight-streaming
+// kernel instead of a CUTLASS grouped GEMM.
+//
+// Regime: Mixtral decode.  R = T*topk token-expert assignments are
+// bucketed per expert (device-side argsort; offsets via cumsum).  Every
+// active expert's weights are read exactly once per step, so — like the
+// dense skinny GEMM this is modeled on (skinny_gemm.hip) — the kernel's
+// job is to stream W at the HBM ceiling with the MFMA work riding along.
+// The expert loop moves onto the GRID (blockIdx.y = expert), the token
+// gather rides the A-fragment load (indirection through tok[]), and the
+// router scale rides the epilogue — so the whole MoE layer runs without
+// a host sync and is hipGraph-capturable (the host-looped fallback in
+// models/mixtral.py syncs on counts.tolist() every layer).
+//
+// Decomposition:
+//   grid = (N/64, E); 4 waves; each wave owns 1/4 of K (in-block split-K,
+//   splitk always 1 — block count N/64*E far exceeds 256 CUs already).
+//   Buckets larger than MT=64 rows loop in MT chunks (weights re-stream
+//   per chunk; at the decode shapes this path serves, avg bucket size is
+//   R/E <= 64, so the common case is one pass).
+//
+// Fragments (mfma_f32_16x16x32_bf16, lane maps verified by GPU tests):
+//   A[i][k]: lane i=l&15, k=(l>>4)*8+m  -> x row tok[i], 16 B contiguous
+//   B[k][j]: lane j=l&15, k=(l>>4)*8+m  -> W^T row n0+j, 16 B contiguous
+//   C[r][c]: lane c=l&15, r=(l>>4)*4+reg
+
+#include "common.h"
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
+
+#define MFMA16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
+
+#define BN 64       // output columns per block
+#define MGWAVES 4   // waves per block (in-block K split)
+#define MOE_MT 64   // rows per M-chunk
+
+// x rows are gathered through tok[] (nullptr = identity: rows are already
+// bucket-ordered, the down projection's case).  direct=0 writes the fp32
+// slab work[r, :N] (gate-up; the SiLU epilogue mixes columns); direct=1
+// writes bf16 out[r, :N] scaled by scale[r] (down projection; the router
+// weight fuses here so the scatter-add back to tokens is a plain
+// index_add over pre-scaled rows).
+template <int KU>
+__global__ __launch_bounds__(MGWAVES * WAVE) void moe_gemm_kernel(
+    float* __restrict__ work,             // [R, N] fp32 slab (direct=0)
+    __hip_bfloat16* __restrict__ out,     // [R, N] bf16 (direct=1)
+    const __hip_bfloat16* __restrict__ x, // [T, K] source rows
+    const __hip_bfloat16* __restrict__ w, // [E, N, K] stacked expert W^T
+    const int* __restrict__ tok,          // [R] row gather index (or null)
+    const int* __restrict__ off,          // [E+1] bucket offsets
+    const float* __restrict__ scale,      // [R] row scale (or null)
+    const int N, const int K, const int direct) {
+  constexpr int MTILES = MOE_MT / 16;
+  const int n0 = blockIdx.x * BN;
+  const int e = blockIdx.y;
+  const int m_begin = off[e], m_end = off[e + 1];
+  if (m_begin >= m_end) return;
+
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int l16 = lane & 15;
+  const int l4 = lane >> 4;
+
+  const __hip_bfloat16* we = w + (int64_t)e * N * K;
+  const int wave_k = K / MGWAVES;
+  const int k_begin = wid * wave_k;
+  const int k_end = k_begin + wave_k;
+  const int wrow = n0 + l16;  // W^T row for this lane's B fragments
+
+  __shared__ float red[MGWAVES][MOE_MT][BN];
+
+  for (int m0 = m_begin; m0 < m_end; m0 += MOE_MT) {
+    f32x4 acc[MTILES][BN / 16];
+#pragma unroll
+    for (int mi = 0; mi < MTILES; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < BN / 16; ++ni)
+        acc[mi][ni] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    // A row per m-tile for this lane; clamp into the bucket, gather via tok
+    int arow[MTILES];
+#pragma unroll
+    for (int mi = 0; mi < MTILES; ++mi) {
+      const int r = min(m0 + mi * 16 + l16, m_end - 1);
+      arow[mi] = tok ? tok[r] : r;
+    }
+
+    for (int k = k_begin; k < k_end; k += KU) {
+      bf16x8v a[KU / 32][MTILES], b[KU / 32][BN / 16];
+#pragma unroll
+      for (int kk = 0; kk < KU; kk += 32) {
+#pragma unroll
+        for (int mi = 0; mi < MTILES; ++mi)
+          a[kk / 32][mi] = *reinterpret_cast<const bf16x8v*>(
+              x + (int64_t)arow[mi] * K + k + kk + l4 * 8);
+#pragma unroll
+        for (int ni = 0; ni < BN / 16; ++ni)
+          b[kk / 32][ni] = *reinterpret_cast<const bf16x8v*>(
+              we + (int64_t)(wrow + ni * 16) * K + k + kk + l4 * 8);
+      }
+#pragma unroll
+      for (int kk = 0; kk < KU; kk += 32)
+#pragma unroll
+        for (int ni = 0; ni < BN / 16; ++ni)
+#pragma unroll
+          for (int mi = 0; mi < MTILES; ++mi)
+            acc[mi][ni] = MFMA16(a[kk / 32][mi], b[kk / 32][ni], acc[mi][ni]);
+    }
+
+    // cross-wave reduce through LDS, then one guarded row-range write.
+    // NOTE the guard r < m_end is load-bearing in slab mode: rows past this
+    // bucket belong to the NEXT expert's blocks.
+    __syncthreads();  // red[] reuse across m-chunks
+#pragma unroll
+    for (int mi = 0; mi < MTILES; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < BN / 16; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          red[wid][mi * 16 + l4 * 4 + r][ni * 16 + l16] = acc[mi][ni][r];
+    __syncthreads();
+
+    for (int i = threadIdx.x; i < MOE_MT * BN; i += MGWAVES * WAVE) {
+      const int rr = i / BN, c = i % BN;
+      const int r = m0 + rr;
+      if (r >= m_end) continue;
+      float v = red[0][rr][c];
+#pragma unroll
+      for (int wv = 1; wv < MGWAVES; ++wv) v += red[wv][rr][c];
+      if (direct) {
+        out[(int64_t)r * N + n0 + c] = f2bf(scale ? v * scale[r] : v);
+      } else {
+        work[(int64_t)r * N + n0 + c] = v;
+      }
+    }
+  }
+}
+
+// gate/up slab -> silu(gate)*up bf16.  work rows are [gate | up] pairs of
+// F columns each (same layout as skinny_gemm's reduce_slabs_silu, splitk=1;
+// duplicated here because -fno-gpu-rdc keeps __global__ symbols per-TU).
+__global__ void moe_silu_kernel(__hip_bfloat16* __restrict__ out,  // [R,F]
+                                const float* __restrict__ work,    // [R,2F]
+                                const int R, const int F) {
+  const int64_t total = (int64_t)R * F;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int r = i / F, c = i % F;
+    const float* row = work + (int64_t)r * 2 * F;
+    const float g = row[c], u = row[F + c];
+    const float s = g / (1.f + __expf(-g));
+    out[i] = f2bf(s * u);
+  }
+}
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+static void launch_moe(torch::Tensor work, torch::Tensor out, torch::Tensor x,
+                       torch::Tensor w, const int* tok, torch::Tensor off,
+                       const float* scale, int direct, hipStream_t s) {
+  const int E = w.size(0), N = w.size(1), K = w.size(2);
+  TORCH_CHECK(x.size(1) == K, "moe_gemm: K mismatch");
+  TORCH_CHECK(N % BN == 0, "moe_gemm: N must be a multiple of 64");
+  TORCH_CHECK(off.numel() == E + 1, "moe_gemm: off must be [E+1]");
+  TORCH_CHECK(off.scalar_type() == torch::kInt32);
+  const int wave_k = K / MGWAVES;
+  dim3 grid(N / BN, E), block(MGWAVES * WAVE);
+#define L(KU)                                                           \
+  hipLaunchKernelGGL((moe_gemm_kernel<KU>), grid, block, 0, s,          \
+                     (float*)work.data_ptr(),                           \
+                     (__hip_bfloat16*)out.data_ptr(),                   \
+                     (const __hip_bfloat16*)x.data_ptr(),               \
+                     (const __hip_bfloat16*)w.data_ptr(), tok,          \
+                     (const int*)off.data_ptr(), scale, N, K, direct)
+  if (wave_k % 64 == 0) L(64);
+  else if (wave_k % 32 == 0) L(32);
+  else TORCH_CHECK(false, "moe_gemm: K=", K, " not divisible by 128");
+#undef L
+  HIP_CHECK_LAST();
+}
+
+void moe_gateup_silu(torch::Tensor act, torch::Tensor x, torch::Tensor w,
+                     torch::Tensor tok, torch::Tensor off,
+                     torch::Tensor work) {
+  // act[R,F] = silu(g)*u where [g|u] rows = x[tok[r]] @ w[expert(r)]^T,
+  // w is [E, 2F, K] stacked gate|up, work is fp32 scratch >= R*2F.
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && act.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(tok.scalar_type() == torch::kInt32);
+  const int R = tok.numel();
+  const int N = w.size(1), F = N / 2;
+  TORCH_CHECK(act.size(0) == R && act.size(1) == F);
+  TORCH_CHECK(work.numel() >= (int64_t)R * N, "moe workspace too small");
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  launch_moe(work, act, x, w, (const int*)tok.data_ptr(), off, nullptr,
+             /*direct=*/0, s);
+  const int64_t total = (int64_t)R * F;
+  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(moe_silu_kernel, dim3(blocks), dim3(256), 0, s,
+                     (__hip_bfloat16*)act.data_ptr(),
+                     (const float*)work.data_ptr(), R, F);
+  HIP_CHECK_LAST();
+}
+
+void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w,
+              torch::Tensor off, torch::Tensor scale) {
+  // y[R,d] = scale[r] * (act[r] @ w[expert(r)]^T); act rows already bucket
+  // ordered (identity gather), w is [E, d, F], scale fp32 [R].
+  TORCH_CHECK(act.is_contiguous() && w.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(act.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(scale.scalar_type() == torch::kFloat32);
+  const int R = act.size(0);
+  TORCH_CHECK(y.size(0) == R && y.size(1) == w.size(1));
+  TORCH_CHECK(scale.numel() == R);
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  launch_moe(y /*unused in direct mode: any fp32-typed arg*/, y, act, w,
+             nullptr, off, (const float*)scale.data_ptr(), /*direct=*/1, s);
+}

@@ -47,3 +47,52 @@ def test_moe_engine_generates():
         assert len(r.token_ids) == 4
     finally:
         eng.stop()
+
+
+def test_moe_buckets_device_free():
+    """moe_buckets: bucket rows sorted by expert, offsets = cumcounts."""
+    from calfkit_amd import ops
+
+    torch.manual_seed(3)
+    T, E, k = 11, 4, 2
+    logits = torch.randn(T, E)
+    weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
+    tok, off, scale = ops.moe_buckets(experts, weights, E)
+    assert tok.dtype == torch.int32 and off.dtype == torch.int32
+    assert int(off[0]) == 0 and int(off[-1]) == T * k
+    # each bucket row must be a (token, expert) pair the router picked,
+    # with the matching weight
+    offs = off.tolist()
+    for e in range(E):
+        for r in range(offs[e], offs[e + 1]):
+            t = int(tok[r])
+            j = (experts[t] == e).nonzero()
+            assert len(j) == 1, (t, e)
+            assert abs(float(scale[r]) - float(weights[t, j[0, 0]])) < 1e-6
+
+
+def test_moe_grouped_fallback_matches_host_loop():
+    """ops.moe_gateup_silu/moe_down CPU references == the host-loop MLP."""
+    from calfkit_amd import ops
+
+    torch.manual_seed(4)
+    model = MixtralShard(TINY_MOE, device="cpu")
+    lw = model.layers[0]
+    T, k, E = 7, 2, TINY_MOE.num_experts
+    d, F = TINY_MOE.hidden, model.ffn_s
+    x = torch.randn(T, d, dtype=torch.bfloat16)
+    host = model._mlp(lw, x)  # host-grouped path
+
+    logits = torch.nn.functional.linear(x, lw.router).float()
+    weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
+    weights = weights / weights.sum(-1, keepdim=True)
+    tok, off, scale = ops.moe_buckets(experts, weights, E)
+    R = T * k
+    act = torch.empty(R, F, dtype=torch.bfloat16)
+    ops.moe_gateup_silu(act, x, lw.wgateup, tok, off, torch.empty(0))
+    y = torch.empty(R, d, dtype=torch.bfloat16)
+    ops.moe_down(y, act, lw.wdown, off, scale)
+    out = torch.zeros(T, d, dtype=torch.float32)
+    out.index_add_(0, tok.long(), y.float())
+    assert torch.allclose(out, host.float(), atol=5e-2, rtol=5e-2), \
+        (out - host.float()).abs().max()
