@@ -42,8 +42,10 @@ async def run() -> None:
         from tests.unit.test_mixtral import TINY_MOE as cfg  # type: ignore
 
     def provider(directed: bool):
+        # use_graphs defaults on: MoE decode is capturable through the
+        # grouped kernels (device-side bucketing, no host sync)
         inner = LocalModelClient(cfg, max_batch=TURNS_PER_STEP * 4,
-                                 max_model_len=1024, use_graphs=False)
+                                 max_model_len=1024)
         if not directed:
             return inner
         return ToolCallDirector(inner, calls=[
