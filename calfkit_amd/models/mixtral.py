@@ -16,9 +16,13 @@ Expert compute:
   compute-bound regime where library tiles win and the one-time host sync
   amortizes over the whole chunk.
 
-TP: every expert's ffn shards over the TP group exactly like the dense MLP
-(row/column parallel + one all-reduce); the router is replicated.  Expert
-parallelism with RCCL all-to-all is a post-baseline extension for EP>1.
+Parallel layouts (router always replicated):
+- TP: every expert's ffn shards over the TP group exactly like the dense
+  MLP (row/column parallel + one all-reduce).
+- EP: experts partition over the EP group; tokens travel to their experts
+  over RCCL all-to-all on xGMI and combine back (``_mlp_ep``).  Per-expert
+  weight seeds make any layout produce identical experts, so EP_n == EP_1
+  numerically (tests/unit/test_expert_parallel.py, gloo world 2).
 """
 
 from __future__ import annotations
@@ -29,7 +33,7 @@ from dataclasses import dataclass
 import torch
 
 from .. import ops
-from ..parallel import all_reduce_sum
+from ..parallel import all_reduce_sum, all_to_all_rows
 from .config import ModelConfig
 from .llama import LayerWeights, LlamaShard
 
@@ -43,28 +47,59 @@ class MoeLayerWeights(LayerWeights):
 
 
 class MixtralShard(LlamaShard):
-    def __init__(self, cfg: ModelConfig, **kw):
+    def __init__(self, cfg: ModelConfig, *, ep_rank: int = 0, ep_size: int = 1,
+                 **kw):
         assert cfg.is_moe, "MixtralShard needs an MoE config"
         super().__init__(cfg, **kw)
-        # replace the dense MLP weights with stacked expert weights + router
+        # Expert parallelism: experts partition over the EP group (attention
+        # and router stay replicated; tokens travel to their experts over
+        # RCCL all-to-all on xGMI).  Round-1 scope: EP and TP are alternative
+        # layouts for the MoE MLP — EP>1 requires TP=1.
+        if ep_size > 1 and self.tp_size > 1:
+            raise ValueError("EP>1 with TP>1 is not supported yet")
+        if cfg.num_experts % ep_size:
+            raise ValueError(f"{cfg.num_experts} experts must divide ep={ep_size}")
+        self.ep_rank = ep_rank
+        self.ep_size = ep_size
+        self.e_loc = cfg.num_experts // ep_size
+
+        # replace the dense MLP weights with stacked expert weights + router.
+        # Per-expert generators (seeded by layer*E+expert) make the weights
+        # layout-independent: any EP sharding reproduces the same experts,
+        # so EP2 == EP1 holds by construction (tests/unit/test_expert_parallel).
         gen_device = "cpu" if self.device.type == "cpu" else self.device
-        gen = torch.Generator(device=gen_device).manual_seed(4321 + self.tp_rank)
         d, e = cfg.hidden, cfg.num_experts
 
-        def w(*shape: int, fan_in: int) -> torch.Tensor:
+        def w(*shape: int, fan_in: int, seed: int) -> torch.Tensor:
+            gen = torch.Generator(device=gen_device).manual_seed(seed)
             std = 1.0 / math.sqrt(fan_in)
             t = torch.randn(*shape, generator=gen, dtype=torch.float32,
                             device=gen_device) * std
             return t.to(device=self.device, dtype=self.dtype)
 
+        f0 = self.tp_rank * self.ffn_s  # TP slice of each expert's ffn
+
+        def expert_gateup(li: int, ei: int) -> torch.Tensor:
+            full = w(2 * cfg.ffn, d, fan_in=d, seed=4321 + li * e + ei)
+            gate, up = full[:cfg.ffn], full[cfg.ffn:]
+            return torch.cat([gate[f0:f0 + self.ffn_s],
+                              up[f0:f0 + self.ffn_s]]).contiguous()
+
+        def expert_down(li: int, ei: int) -> torch.Tensor:
+            full = w(d, cfg.ffn, fan_in=cfg.ffn, seed=9876 + li * e + ei)
+            return full[:, f0:f0 + self.ffn_s].contiguous()
+
+        e0 = ep_rank * self.e_loc
         layers: list[MoeLayerWeights] = []
-        for lw in self.layers:
+        for li, lw in enumerate(self.layers):
             layers.append(MoeLayerWeights(
                 wqkv=lw.wqkv, wo=lw.wo,
-                wgateup=w(e, 2 * self.ffn_s, d, fan_in=d),
-                wdown=w(e, d, self.ffn_s, fan_in=cfg.ffn),
+                wgateup=torch.stack([expert_gateup(li, ei)
+                                     for ei in range(e0, e0 + self.e_loc)]),
+                wdown=torch.stack([expert_down(li, ei)
+                                   for ei in range(e0, e0 + self.e_loc)]),
                 norm_attn=lw.norm_attn, norm_mlp=lw.norm_mlp,
-                router=w(e, d, fan_in=d),
+                router=w(e, d, fan_in=d, seed=5555 + li),
             ))
         self.layers = layers
         self.topk = cfg.experts_per_token
@@ -85,6 +120,9 @@ class MixtralShard(LlamaShard):
         logits = torch.nn.functional.linear(hidden, lw.router).float()  # [T,E]
         weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
         weights = weights / weights.sum(-1, keepdim=True)   # renormalize top-k
+
+        if self.ep_size > 1:
+            return self._mlp_ep(lw, hidden, weights, experts)
 
         if decode and self.device.type == "cuda" and ops.have_extension() \
                 and T <= 256:
@@ -128,3 +166,60 @@ class MixtralShard(LlamaShard):
                            .to(out.dtype))
             start += n
         return all_reduce_sum(out) if self.tp_size > 1 else out
+
+    def _mlp_ep(self, lw: MoeLayerWeights, hidden: torch.Tensor,
+                weights: torch.Tensor, experts: torch.Tensor) -> torch.Tensor:
+        """Expert-parallel MLP: dispatch tokens to their experts' owner ranks
+        over all-to-all, compute the local expert shard, combine back.
+
+        The exchange is the reference's EP dispatch/combine pattern
+         on RCCL point-to-point xGMI.  Variable splits need a
+        counts exchange (host sync), so EP runs eager — decode stays on the
+        TP/grouped layout for graph capture; EP is the capacity layout for
+        models whose experts exceed one rank's HBM.
+        """
+        T, d = hidden.shape
+        e, k, e_loc = self.cfg.num_experts, self.topk, self.e_loc
+        tok, off, scale = ops.moe_buckets(experts, weights, e)
+        R = T * k
+        counts = off.diff()
+        # rows for rank r = buckets of its expert range, already contiguous
+        send_splits = [int(counts[r * e_loc:(r + 1) * e_loc].sum())
+                       for r in range(self.ep_size)]
+        eids = torch.repeat_interleave(
+            torch.arange(e, device=hidden.device, dtype=torch.int32), counts)
+        rows = hidden[tok.long()]
+        recv_rows, recv_splits = all_to_all_rows(rows, send_splits)
+        recv_eids, _ = all_to_all_rows(eids, send_splits, recv_splits)
+
+        # regroup across sources by local expert (each source segment is
+        # expert-sorted; the concatenation is not)
+        local_eid = recv_eids - self.ep_rank * e_loc
+        order2 = torch.argsort(local_eid, stable=True).int()
+        counts2 = torch.zeros(e_loc, device=hidden.device, dtype=torch.int32)
+        counts2.scatter_add_(0, local_eid.long(),
+                             torch.ones_like(local_eid, dtype=torch.int32))
+        off2 = torch.zeros(e_loc + 1, device=hidden.device, dtype=torch.int32)
+        torch.cumsum(counts2, 0, out=off2[1:])
+        R2 = recv_rows.shape[0]
+
+        if R2 > 0:
+            act = torch.empty(R2, self.ffn_s, dtype=hidden.dtype,
+                              device=hidden.device)
+            work = self._moe_work if self._moe_work.numel() >= R2 * 2 * self.ffn_s \
+                else torch.empty(R2 * 2 * self.ffn_s, dtype=torch.float32,
+                                 device=hidden.device)
+            ops.moe_gateup_silu(act, recv_rows, lw.wgateup, order2, off2, work)
+            y2 = torch.empty(R2, d, dtype=hidden.dtype, device=hidden.device)
+            ones = torch.ones(R2, dtype=torch.float32, device=hidden.device)
+            ops.moe_down(y2, act, lw.wdown, off2, ones)
+            # un-permute to the received order so sources can reassemble
+            y_src = torch.empty_like(y2)
+            y_src[order2.long()] = y2
+        else:
+            y_src = torch.empty(0, d, dtype=hidden.dtype, device=hidden.device)
+        y_back, _ = all_to_all_rows(y_src, recv_splits, send_splits)
+
+        out32 = torch.zeros(T, d, dtype=torch.float32, device=hidden.device)
+        out32.index_add_(0, tok.long(), y_back.float() * scale[:, None])
+        return out32.to(hidden.dtype)

@@ -11,6 +11,7 @@ prefill all-reduces chunked.
 from .pg import (
     DistInfo,
     all_reduce_sum,
+    all_to_all_rows,
     barrier,
     get_dist,
     init_distributed,
@@ -21,6 +22,7 @@ from .pg import (
 __all__ = [
     "DistInfo",
     "all_reduce_sum",
+    "all_to_all_rows",
     "barrier",
     "get_dist",
     "init_distributed",

@@ -20,7 +20,7 @@ import torch
 import torch.distributed as dist
 
 __all__ = ["DistInfo", "init_distributed", "get_dist", "is_initialized",
-           "all_reduce_sum", "barrier", "shard_dim"]
+           "all_reduce_sum", "all_to_all_rows", "barrier", "shard_dim"]
 
 
 @dataclass(frozen=True)
@@ -83,6 +83,32 @@ def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
     if d.world_size > 1:
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
     return t
+
+
+def all_to_all_rows(t: torch.Tensor, send_splits: list[int],
+                    recv_splits: list[int] | None = None,
+                    ) -> tuple[torch.Tensor, list[int]]:
+    """Variable-row all-to-all: rank r receives t's send_splits[r] rows from
+    every peer, concatenated in rank order.
+
+    The expert-parallel token exchange (MoE dispatch/combine).  When
+    recv_splits is unknown it is exchanged first with a counts all-to-all —
+    a host sync, which is why EP decode runs eager (the
+    fixed-capacity capturable variant is the planned next step).  On GPU
+    this rides RCCL's point-to-point xGMI sends; gloo serves the CPU lane.
+    """
+    d = get_dist()
+    if d.world_size == 1:
+        return t, [t.shape[0]]
+    if recv_splits is None:
+        counts = torch.tensor(send_splits, dtype=torch.int64, device=t.device)
+        rcounts = torch.empty_like(counts)
+        dist.all_to_all_single(rcounts, counts)
+        recv_splits = [int(c) for c in rcounts]
+    out = torch.empty(sum(recv_splits), *t.shape[1:], dtype=t.dtype,
+                      device=t.device)
+    dist.all_to_all_single(out, t.contiguous(), recv_splits, send_splits)
+    return out, recv_splits
 
 
 def barrier() -> None:

@@ -1,0 +1,80 @@
+"""Multi-process expert parallelism on CPU (gloo, world_size 2).
+
+EP=2 (experts split across ranks, tokens exchanged over all-to-all) must
+reproduce the EP=1 output.  Per-expert weight seeds make every layout
+generate identical experts, so no state copying is needed — the layouts
+agree by construction (models/mixtral.py).
+"""
+
+import os
+from dataclasses import replace
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from calfkit_amd.models.config import ModelConfig
+
+pytestmark = pytest.mark.dist
+
+CFG = ModelConfig(
+    name="tiny-moe-ep", vocab_size=128, hidden=128, layers=2, heads=2,
+    kv_heads=1, head_dim=128, ffn=256, rope_theta=10000.0, max_seq=256,
+    num_experts=4, experts_per_token=2)
+
+
+def _mlp_inputs(cfg):
+    torch.manual_seed(11)
+    return torch.randn(6, cfg.hidden, dtype=torch.bfloat16)
+
+
+def _worker(rank, world, port, result_q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import calfkit_amd.parallel.pg as pg
+
+    pg._DIST = None
+    pg.init_distributed(backend="gloo", device_override="cpu")
+    from calfkit_amd.models.mixtral import MixtralShard
+
+    model = MixtralShard(CFG, ep_rank=rank, ep_size=world, device="cpu")
+    x = _mlp_inputs(CFG)
+    out = model._mlp(model.layers[0], x)
+    if rank == 0:
+        result_q.put(out.float())
+    import torch.distributed as td
+
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_ep2_matches_ep1():
+    from calfkit_amd.models.mixtral import MixtralShard
+
+    single = MixtralShard(CFG, device="cpu")
+    x = _mlp_inputs(CFG)
+    ref = single._mlp(single.layers[0], x).float()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, 2, 29519, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    ep_out = q.get(timeout=120)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert torch.allclose(ep_out, ref, atol=5e-2, rtol=5e-2), \
+        (ep_out - ref).abs().max()
+
+
+def test_ep_rejects_uneven_and_tp_mix():
+    from calfkit_amd.models.mixtral import MixtralShard
+
+    with pytest.raises(ValueError):
+        MixtralShard(replace(CFG, num_experts=3), ep_size=2, device="cpu")
+    with pytest.raises(ValueError):
+        MixtralShard(replace(CFG, heads=4, kv_heads=2, ffn=512),
+                     ep_size=2, tp_size=2, device="cpu")
