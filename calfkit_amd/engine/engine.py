@@ -82,11 +82,12 @@ class InferenceEngine:
     def __init__(self, model: LlamaShard, *, max_batch: int = 64,
                  kv_pages: int | None = None, use_graphs: bool | None = None,
                  max_model_len: int | None = None):
-        #: TP>1 serving runs in lockstep: rank 0 schedules and broadcasts
+        #: TP/EP>1 serving runs in lockstep: rank 0 schedules and broadcasts
         #: each action; ranks > 0 replay them via follow() — every rank then
         #: executes identical forwards (and RCCL collectives) in identical
         #: order, and identical seeds make the sampled tokens identical.
-        self.tp_lockstep = model.tp_size > 1
+        self.tp_lockstep = (model.tp_size > 1
+                            or getattr(model, "ep_size", 1) > 1)
         self.model = model
         self.cfg: ModelConfig = model.cfg
         self.device = model.device
@@ -106,7 +107,9 @@ class InferenceEngine:
             # would catch it, but don't even try.
             from .. import ops as _ops
             use_graphs = self.on_gpu and (
-                not self.cfg.is_moe or _ops.have_extension())
+                not self.cfg.is_moe
+                or (_ops.have_extension()
+                    and getattr(model, "ep_size", 1) == 1))
         self.use_graphs = use_graphs
 
         if kv_pages is None:

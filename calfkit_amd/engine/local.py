@@ -48,13 +48,16 @@ atexit.register(shutdown_engines)
 
 
 def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
-                  autostart: bool = True, **engine_kw: Any) -> InferenceEngine:
+                  ep: int = 1, autostart: bool = True,
+                  **engine_kw: Any) -> InferenceEngine:
     """One engine per (model, device) per process — all agents share it.
 
     ``tp`` > 1 shards the model across the torch.distributed world (every
     rank must construct the same engine and drive the same forward order);
-    the default is one independent replica per process (the
-    one-Worker-per-GPU scaling story).
+    ``ep`` > 1 does the same with expert parallelism (MoE models only; the
+    lockstep discipline is identical — every rank steps the same batch and
+    the MoE all-to-alls line up).  The default is one independent replica
+    per process (the one-Worker-per-GPU scaling story).
     """
     from ..models.config import ModelConfig, get_config
     from ..models.llama import LlamaShard
@@ -64,7 +67,9 @@ def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
     assert isinstance(config, ModelConfig)
     if device is None:
         device = "cuda" if torch.cuda.is_available() else "cpu"
-    key = (config.name, device, tp)
+    if ep > 1 and not config.is_moe:
+        raise ValueError(f"{config.name} is dense; ep={ep} needs an MoE model")
+    key = (config.name, device, tp, ep)
     with _ENGINES_LOCK:
         eng = _ENGINES.get(key)
         if eng is None:
@@ -72,14 +77,22 @@ def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
                 from ..models.mixtral import MixtralShard as Shard
             else:
                 Shard = LlamaShard
-            if tp > 1:
+            if tp > 1 or ep > 1:
                 from ..parallel import get_dist
 
                 dist = get_dist()
-                if dist.world_size != tp:
+                world = max(tp, ep)
+                if dist.world_size != world:
                     raise ValueError(
-                        f"tp={tp} but torch.distributed world is {dist.world_size}")
-                model = Shard(config, tp_rank=dist.rank, tp_size=tp, device=device)
+                        f"tp={tp}/ep={ep} but torch.distributed world is "
+                        f"{dist.world_size}")
+                if config.is_moe:
+                    model = Shard(config, tp_rank=dist.rank if tp > 1 else 0,
+                                  tp_size=tp, ep_rank=dist.rank if ep > 1 else 0,
+                                  ep_size=ep, device=device)
+                else:
+                    model = Shard(config, tp_rank=dist.rank, tp_size=tp,
+                                  device=device)
             else:
                 model = Shard(config, tp_rank=0, tp_size=1, device=device)
             eng = InferenceEngine(model, **engine_kw)
@@ -95,11 +108,12 @@ class LocalModelClient:
     def __init__(self, model: Any = "llama-3-8b", *, device: str | None = None,
                  max_batch: int = 64, max_model_len: int | None = None,
                  use_graphs: bool | None = None, kv_pages: int | None = None,
-                 tp: int = 1):
+                 tp: int = 1, ep: int = 1):
         self._config = model
         self._device = device
         self._engine_kw = dict(max_batch=max_batch, max_model_len=max_model_len,
-                               use_graphs=use_graphs, kv_pages=kv_pages, tp=tp)
+                               use_graphs=use_graphs, kv_pages=kv_pages, tp=tp,
+                               ep=ep)
         self._engine: InferenceEngine | None = None
         self._tokenizer: SyntheticTokenizer | None = None
 

@@ -78,3 +78,47 @@ def test_ep_rejects_uneven_and_tp_mix():
     with pytest.raises(ValueError):
         MixtralShard(replace(CFG, heads=4, kv_heads=2, ffn=512),
                      ep_size=2, tp_size=2, device="cpu")
+
+
+def _serve_worker(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import calfkit_amd.parallel.pg as pg
+
+    pg._DIST = None
+    pg.init_distributed(backend="gloo", device_override="cpu")
+    from calfkit_amd.engine.engine import GenRequest
+    from calfkit_amd.engine.local import shared_engine
+
+    engine = shared_engine(CFG, "cpu", ep=2, max_batch=4, use_graphs=False,
+                           max_model_len=128, autostart=False)
+    assert engine.tp_lockstep
+    if rank == 0:
+        engine.start()
+        fut = engine.submit(GenRequest(prompt_ids=[1, 7, 9], max_new_tokens=4))
+        out = fut.result(120).token_ids
+        engine.stop()
+        q.put(out)
+    else:
+        engine.follow()
+    import torch.distributed as td
+
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_ep_lockstep_serving_completes():
+    """EP=2 engine serves in lockstep (rank 0 schedules, rank 1 follows,
+    MoE all-to-alls line up) and completes a generation."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_serve_worker, args=(r, 2, 29523, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert len(out) == 4
