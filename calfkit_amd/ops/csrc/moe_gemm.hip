@@ -44,7 +44,12 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 // writes bf16 out[r, :N] scaled by scale[r] (down projection; the router
 // weight fuses here so the scatter-add back to tokens is a plain
 // index_add over pre-scaled rows).
-template <int KU>
+// MT is the rows-per-chunk template: every chunk past the first re-streams
+// the expert's weights, so MT=128 (accumulators for 8 m-tiles live across
+// the whole k-loop) halves the re-read when buckets run big (decode batch
+// 256 -> avg bucket 64, tails ~90).  The LDS reduce stays a 64-row panel
+// (two passes at MT=128) so occupancy holds at 2 blocks/CU.
+template <int KU, int MT>
 __global__ __launch_bounds__(MGWAVES * WAVE) void moe_gemm_kernel(
     float* __restrict__ work,             // [R, N] fp32 slab (direct=0)
     __hip_bfloat16* __restrict__ out,     // [R, N] bf16 (direct=1)
@@ -54,7 +59,7 @@ __global__ __launch_bounds__(MGWAVES * WAVE) void moe_gemm_kernel(
     const int* __restrict__ off,          // [E+1] bucket offsets
     const float* __restrict__ scale,      // [R] row scale (or null)
     const int N, const int K, const int direct) {
-  constexpr int MTILES = MOE_MT / 16;
+  constexpr int MTILES = MT / 16;
   const int n0 = blockIdx.x * BN;
   const int e = blockIdx.y;
   const int m_begin = off[e], m_end = off[e + 1];
@@ -71,9 +76,9 @@ __global__ __launch_bounds__(MGWAVES * WAVE) void moe_gemm_kernel(
   const int k_end = k_begin + wave_k;
   const int wrow = n0 + l16;  // W^T row for this lane's B fragments
 
-  __shared__ float red[MGWAVES][MOE_MT][BN];
+  __shared__ float red[MGWAVES][MOE_MT][BN];  // 64-row panel, 64 KB
 
-  for (int m0 = m_begin; m0 < m_end; m0 += MOE_MT) {
+  for (int m0 = m_begin; m0 < m_end; m0 += MT) {
     f32x4 acc[MTILES][BN / 16];
 #pragma unroll
     for (int mi = 0; mi < MTILES; ++mi)
@@ -111,30 +116,35 @@ __global__ __launch_bounds__(MGWAVES * WAVE) void moe_gemm_kernel(
             acc[mi][ni] = MFMA16(a[kk / 32][mi], b[kk / 32][ni], acc[mi][ni]);
     }
 
-    // cross-wave reduce through LDS, then one guarded row-range write.
-    // NOTE the guard r < m_end is load-bearing in slab mode: rows past this
-    // bucket belong to the NEXT expert's blocks.
-    __syncthreads();  // red[] reuse across m-chunks
+    // cross-wave reduce through LDS in 64-row panels (MT/64 passes), then
+    // one guarded row-range write per panel.  NOTE the guard r < m_end is
+    // load-bearing in slab mode: rows past this bucket belong to the NEXT
+    // expert's blocks.
 #pragma unroll
-    for (int mi = 0; mi < MTILES; ++mi)
+    for (int p = 0; p < MT / MOE_MT; ++p) {
+      __syncthreads();  // red[] reuse across panels/m-chunks
 #pragma unroll
-      for (int ni = 0; ni < BN / 16; ++ni)
+      for (int mi = 0; mi < MOE_MT / 16; ++mi)
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          red[wid][mi * 16 + l4 * 4 + r][ni * 16 + l16] = acc[mi][ni][r];
-    __syncthreads();
+        for (int ni = 0; ni < BN / 16; ++ni)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            red[wid][mi * 16 + l4 * 4 + r][ni * 16 + l16] =
+                acc[p * (MOE_MT / 16) + mi][ni][r];
+      __syncthreads();
 
-    for (int i = threadIdx.x; i < MOE_MT * BN; i += MGWAVES * WAVE) {
-      const int rr = i / BN, c = i % BN;
-      const int r = m0 + rr;
-      if (r >= m_end) continue;
-      float v = red[0][rr][c];
+      for (int i = threadIdx.x; i < MOE_MT * BN; i += MGWAVES * WAVE) {
+        const int rr = i / BN, c = i % BN;
+        const int r = m0 + p * MOE_MT + rr;
+        if (r >= m_end) continue;
+        float v = red[0][rr][c];
 #pragma unroll
-      for (int wv = 1; wv < MGWAVES; ++wv) v += red[wv][rr][c];
-      if (direct) {
-        out[(int64_t)r * N + n0 + c] = f2bf(scale ? v * scale[r] : v);
-      } else {
-        work[(int64_t)r * N + n0 + c] = v;
+        for (int wv = 1; wv < MGWAVES; ++wv) v += red[wv][rr][c];
+        if (direct) {
+          out[(int64_t)r * N + n0 + c] = f2bf(scale ? v * scale[r] : v);
+        } else {
+          work[(int64_t)r * N + n0 + c] = v;
+        }
       }
     }
   }
@@ -159,10 +169,16 @@ __global__ void moe_silu_kernel(__hip_bfloat16* __restrict__ out,  // [R,F]
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
+
+static int moe_env_int(const char* name, int dflt) {
+  const char* v = getenv(name);
+  return v ? atoi(v) : dflt;
+}
 
 static void launch_moe(torch::Tensor work, torch::Tensor out, torch::Tensor x,
                        torch::Tensor w, const int* tok, torch::Tensor off,
-                       const float* scale, int direct, hipStream_t s) {
+                       const float* scale, int direct, int R, hipStream_t s) {
   const int E = w.size(0), N = w.size(1), K = w.size(2);
   TORCH_CHECK(x.size(1) == K, "moe_gemm: K mismatch");
   TORCH_CHECK(N % BN == 0, "moe_gemm: N must be a multiple of 64");
@@ -170,16 +186,25 @@ static void launch_moe(torch::Tensor work, torch::Tensor out, torch::Tensor x,
   TORCH_CHECK(off.scalar_type() == torch::kInt32);
   const int wave_k = K / MGWAVES;
   dim3 grid(N / BN, E), block(MGWAVES * WAVE);
-#define L(KU)                                                           \
-  hipLaunchKernelGGL((moe_gemm_kernel<KU>), grid, block, 0, s,          \
+  // big-bucket regime: avg rows/expert >= ~40 makes >64-row buckets (and
+  // their weight re-streams) common, so accumulate 128 rows per pass
+  int mt = moe_env_int("CALF_MOE_MT", R >= moe_env_int("CALF_MOE_MT_AT", 320)
+                                          ? 128 : 64);
+  int ku = moe_env_int("CALF_MOE_KU", 64);
+  while (ku > 32 && wave_k % ku != 0) ku /= 2;
+#define L(KU, MT)                                                       \
+  hipLaunchKernelGGL((moe_gemm_kernel<KU, MT>), grid, block, 0, s,      \
                      (float*)work.data_ptr(),                           \
                      (__hip_bfloat16*)out.data_ptr(),                   \
                      (const __hip_bfloat16*)x.data_ptr(),               \
                      (const __hip_bfloat16*)w.data_ptr(), tok,          \
                      (const int*)off.data_ptr(), scale, N, K, direct)
-  if (wave_k % 64 == 0) L(64);
-  else if (wave_k % 32 == 0) L(32);
-  else TORCH_CHECK(false, "moe_gemm: K=", K, " not divisible by 128");
+  TORCH_CHECK(wave_k % 32 == 0, "moe_gemm: K=", K, " not divisible by 128");
+  if (mt == 128) {
+    if (ku >= 64) L(64, 128); else L(32, 128);
+  } else {
+    if (ku >= 64) L(64, 64); else L(32, 64);
+  }
 #undef L
   HIP_CHECK_LAST();
 }
@@ -198,7 +223,7 @@ void moe_gateup_silu(torch::Tensor act, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(work.numel() >= (int64_t)R * N, "moe workspace too small");
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
   launch_moe(work, act, x, w, (const int*)tok.data_ptr(), off, nullptr,
-             /*direct=*/0, s);
+             /*direct=*/0, R, s);
   const int64_t total = (int64_t)R * F;
   const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(moe_silu_kernel, dim3(blocks), dim3(256), 0, s,
@@ -219,5 +244,5 @@ void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w,
   TORCH_CHECK(scale.numel() == R);
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
   launch_moe(y /*unused in direct mode: any fp32-typed arg*/, y, act, w,
-             nullptr, off, (const float*)scale.data_ptr(), /*direct=*/1, s);
+             nullptr, off, (const float*)scale.data_ptr(), /*direct=*/1, R, s);
 }
