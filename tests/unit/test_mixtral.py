@@ -1,5 +1,6 @@
 """Unit: Mixtral MoE layer + engine integration (tiny MoE config, CPU)."""
 
+import pytest
 import torch
 
 from calfkit_amd.engine.engine import GenRequest, InferenceEngine
@@ -96,3 +97,19 @@ def test_moe_grouped_fallback_matches_host_loop():
     out.index_add_(0, tok.long(), y.float())
     assert torch.allclose(out, host.float(), atol=5e-2, rtol=5e-2), \
         (out - host.float()).abs().max()
+
+
+@pytest.mark.gpu
+def test_moe_engine_decodes_under_graphs_gpu():
+    """On GPU the MoE engine must run decode under hipGraphs (the grouped
+    kernels make it capturable) — a silent eager fallback is a regression."""
+    model = MixtralShard(TINY_MOE, device="cuda")
+    eng = InferenceEngine(model, max_batch=2, max_model_len=256)
+    try:
+        assert eng.use_graphs
+        r = eng.generate(GenRequest(prompt_ids=[1, 5, 9], max_new_tokens=8),
+                         timeout=120)
+        assert len(r.token_ids) == 8
+        assert eng._stats["graph_replays"] > 0
+    finally:
+        eng.stop()
