@@ -190,7 +190,10 @@ static void launch_moe(torch::Tensor work, torch::Tensor out, torch::Tensor x,
   // their weight re-streams) common, so accumulate 128 rows per pass
   int mt = moe_env_int("CALF_MOE_MT", R >= moe_env_int("CALF_MOE_MT_AT", 320)
                                           ? 128 : 64);
-  int ku = moe_env_int("CALF_MOE_KU", 64);
+  // MT=128/KU=64 sits at 172 VGPRs -> 1 wave/SIMD and LOSES across the
+  // board; KU=32 keeps 2 waves/SIMD and wins the big-bucket regime
+  // (profiles/r01_moe_micro.md sweep: T=256 1.24 -> 1.02 ms)
+  int ku = moe_env_int("CALF_MOE_KU", mt == 128 ? 32 : 64);
   while (ku > 32 && wave_k % ku != 0) ku /= 2;
 #define L(KU, MT)                                                       \
   hipLaunchKernelGGL((moe_gemm_kernel<KU, MT>), grid, block, 0, s,      \
