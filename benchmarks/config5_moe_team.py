@@ -41,11 +41,24 @@ async def run() -> None:
     else:
         from tests.unit.test_mixtral import TINY_MOE as cfg  # type: ignore
 
+    # Multi-rank launch (torchrun) shards the experts: EP = world size,
+    # rank 0 drives the mesh + engine schedule, followers replay in
+    # lockstep (the per-layer all-to-alls line up).  Single rank = the
+    # BASELINE single-GPU measurement.
+    ep = dist.world_size if dist.world_size > 1 else 1
+    if dist.rank != 0:
+        from calfkit_amd.engine.local import shared_engine
+
+        engine = shared_engine(cfg, ep=ep, max_batch=TURNS_PER_STEP * 4,
+                               max_model_len=1024, autostart=False)
+        engine.follow()
+        return
+
     def provider(directed: bool):
-        # use_graphs defaults on: MoE decode is capturable through the
-        # grouped kernels (device-side bucketing, no host sync)
+        # use_graphs defaults on (EP=1): MoE decode is capturable through
+        # the grouped kernels (device-side bucketing, no host sync)
         inner = LocalModelClient(cfg, max_batch=TURNS_PER_STEP * 4,
-                                 max_model_len=1024)
+                                 max_model_len=1024, ep=ep)
         if not directed:
             return inner
         return ToolCallDirector(inner, calls=[
