@@ -179,9 +179,8 @@ class MixtralShard(LlamaShard):
         models whose experts exceed one rank's HBM.
         """
         T, d = hidden.shape
-        e, k, e_loc = self.cfg.num_experts, self.topk, self.e_loc
+        e, e_loc = self.cfg.num_experts, self.e_loc
         tok, off, scale = ops.moe_buckets(experts, weights, e)
-        R = T * k
         counts = off.diff()
         # rows for rank r = buckets of its expert range, already contiguous
         send_splits = [int(counts[r * e_loc:(r + 1) * e_loc].sum())
