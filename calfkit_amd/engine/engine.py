@@ -431,23 +431,32 @@ class InferenceEngine:
                 self._running.append(s)
 
     def _gather_past(self, seqs, cached, kv_starts_l) -> None:
-        """Copy each sequence's cached pages into the prefill scratch."""
+        """Copy cached pages into the prefill scratch — ONE indexed gather +
+        scatter per layer tensor.  (The per-sequence loop this replaced
+        issued layers x seqs x 2 small torch ops; at 32 layers x 256
+        sequences that was the dominant host gap in the batch-256 bench —
+        see profiles/r01_bench256_gaps.md.)"""
         from .kvcache import PAGE_SIZE as PG
 
+        pages_l: list[int] = []
+        rows_l: list[int] = []
+        for s, c, start in zip(seqs, cached, kv_starts_l[:-1]):
+            if c == 0:
+                continue
+            pages_l.extend(self.kv._slots[s.seq_id].pages[: c // PG])
+            rows_l.extend(range(start, start + c))
+        if not pages_l:
+            return
+        dev = self.device
+        pages = torch.as_tensor(pages_l, dtype=torch.long, device=dev)
+        rows = torch.as_tensor(rows_l, dtype=torch.long, device=dev)
+        hkv, dh = self.model.hkv, self.model.dh
         for layer in range(self.cfg.layers):
             kbuf, vbuf = self._prefill_kv[layer]
             kc, vc = self.kv.kcache[layer], self.kv.vcache[layer]
-            for s, c, start in zip(seqs, cached, kv_starts_l[:-1]):
-                if c == 0:
-                    continue
-                pages = self.kv._slots[s.seq_id].pages[: c // PG]
-                idx = torch.as_tensor(pages, dtype=torch.long,
-                                      device=kc.device)
-                # [P, hkv, PG, dh] -> [P*PG, hkv, dh]
-                kbuf[start:start + c] = kc[idx].permute(0, 2, 1, 3).reshape(
-                    -1, self.model.hkv, self.model.dh)
-                vbuf[start:start + c] = vc[idx].permute(0, 2, 1, 3).reshape(
-                    -1, self.model.hkv, self.model.dh)
+            # [P, hkv, PG, dh] -> gathered [n*PG, hkv, dh] at scratch rows
+            kbuf[rows] = kc[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)
+            vbuf[rows] = vc[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)
 
     def _sample_batch_host(self, logits: torch.Tensor,
                            seqs: list["_Seq"]) -> list[int]:
