@@ -506,40 +506,47 @@ class InferenceEngine:
         return 1
 
     def _write_run_state(self, pad: int, M: int) -> None:
+        """Assemble the run's device state host-side with numpy (the
+        python-nested-list version cost several ms per run at batch 256 —
+        one of the per-run host gaps in profiles/r01_bench256_gaps.md)."""
+        import numpy as np
+
         B = len(self._running)
-        tokens = [s.last_token for s in self._running] + [0] * (pad - B)
-        slots = [[0] * pad for _ in range(M)]
-        positions = [[0] * pad for _ in range(M)]
-        lens = [[1] * pad for _ in range(M)]
-        pt_rows = []
+        tokens = np.zeros(pad, dtype=np.int64)
+        slots = np.empty((M, pad), dtype=np.int32)
+        lens = np.ones((M, pad), dtype=np.int32)
+        pt = np.zeros((pad, self.max_pages), dtype=np.int32)
+        base = np.empty(pad, dtype=np.int32)
+        base[B:] = 0
+        run_slots = np.empty((B, M), dtype=np.int32)
+        temps = np.zeros(pad, dtype=np.float32)
+        topks = np.zeros(pad, dtype=np.int32)
         for b, s in enumerate(self._running):
-            base = self.kv.length(s.seq_id)
-            run_slots = self.kv.slots_for(s.seq_id, M)
-            for i in range(M):
-                slots[i][b] = run_slots[i]
-                positions[i][b] = base + i
-                lens[i][b] = base + i + 1
+            tokens[b] = s.last_token
+            base[b] = self.kv.length(s.seq_id)
+            run_slots[b] = self.kv.slots_for(s.seq_id, M)
             pages = self.kv._slots[s.seq_id].pages
-            pt_rows.append(pages + [0] * (self.max_pages - len(pages)))
-        for b in range(B, pad):
-            for i in range(M):
-                slots[i][b] = self._dummy_page * PAGE_SIZE + i
-            pt_rows.append([self._dummy_page] + [0] * (self.max_pages - 1))
-        self._tokens[:pad].copy_(torch.as_tensor(tokens, dtype=torch.int64),
-                                 non_blocking=True)
-        self._positions[:M, :pad].copy_(
-            torch.as_tensor(positions, dtype=torch.int32), non_blocking=True)
-        self._slots[:M, :pad].copy_(torch.as_tensor(slots, dtype=torch.int32),
-                                    non_blocking=True)
-        self._seq_lens[:M, :pad].copy_(
-            torch.as_tensor(lens, dtype=torch.int32), non_blocking=True)
-        self._page_table[:pad].copy_(
-            torch.as_tensor(pt_rows, dtype=torch.int32), non_blocking=True)
-        temps = [s.request.temperature for s in self._running] + [0.0] * (pad - B)
-        topks = [s.request.top_k for s in self._running] + [0] * (pad - B)
-        self._temperature[:pad].copy_(torch.as_tensor(temps), non_blocking=True)
-        self._topk[:pad].copy_(torch.as_tensor(topks, dtype=torch.int32),
-                               non_blocking=True)
+            pt[b, :len(pages)] = pages
+            temps[b] = s.request.temperature
+            topks[b] = s.request.top_k
+        steps = np.arange(M, dtype=np.int32)
+        positions = base[None, :] + steps[:, None]
+        lens = positions + 1
+        slots[:, :B] = run_slots.T
+        # padding rows decode against the dummy page
+        slots[:, B:] = (self._dummy_page * PAGE_SIZE + steps)[:, None]
+        pt[B:, 0] = self._dummy_page
+        lens[:, B:] = 1
+        self._tokens[:pad].copy_(torch.from_numpy(tokens), non_blocking=True)
+        self._positions[:M, :pad].copy_(torch.from_numpy(positions),
+                                        non_blocking=True)
+        self._slots[:M, :pad].copy_(torch.from_numpy(slots), non_blocking=True)
+        self._seq_lens[:M, :pad].copy_(torch.from_numpy(lens),
+                                       non_blocking=True)
+        self._page_table[:pad].copy_(torch.from_numpy(pt), non_blocking=True)
+        self._temperature[:pad].copy_(torch.from_numpy(temps),
+                                      non_blocking=True)
+        self._topk[:pad].copy_(torch.from_numpy(topks), non_blocking=True)
         base_seed = self._running[0].request.seed or 12345
         seeds = [[base_seed, self._step_counter + i + 1] for i in range(M)]
         self._seeds[:M].copy_(torch.as_tensor(seeds, dtype=torch.int64),
