@@ -194,6 +194,14 @@ def main() -> None:
             },
         }
         print(json.dumps(result), flush=True)
+        # diagnostics (stderr, not part of the bench contract): where the
+        # engine spent the run — GPU phases vs starved-scheduler idle
+        es = out.get("engine_stats")
+        if es:
+            stats = es() if callable(es) else es
+            print("engine_stats: " + json.dumps(
+                {k: round(v, 1) if isinstance(v, float) else v
+                 for k, v in stats.items()}), file=sys.stderr, flush=True)
 
 
 if __name__ == "__main__":

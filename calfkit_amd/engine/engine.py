@@ -160,7 +160,8 @@ class InferenceEngine:
         self._next_id = 0
         self._step_counter = 0
         self._stats = {"decode_steps": 0, "decode_tokens": 0, "prefills": 0,
-                       "decode_ms": 0.0, "prefill_ms": 0.0, "graph_replays": 0}
+                       "decode_ms": 0.0, "prefill_ms": 0.0, "graph_replays": 0,
+                       "idle_ms": 0.0}  # scheduler starved (no work queued)
         self._stop = threading.Event()
         self._thread = threading.Thread(target=self._loop, name="calfkit-engine",
                                         daemon=True)
@@ -255,7 +256,12 @@ class InferenceEngine:
             _log.exception("hipGraph capture failed; falling back to eager")
             self.use_graphs = False
         while not self._stop.is_set():
-            self._drain_queue(block=not (self._running or self._waiting))
+            starved = not (self._running or self._waiting)
+            if starved:
+                t_idle = time.perf_counter()
+            self._drain_queue(block=starved)
+            if starved:
+                self._stats["idle_ms"] += (time.perf_counter() - t_idle) * 1e3
             if self._stop.is_set():
                 break
             try:
