@@ -54,6 +54,12 @@ class GenRequest:
     seed: int = 0
     ignore_eos: bool = True
     on_token: Callable[[int], None] | None = None
+    #: batched delivery — called once per decode RUN with all tokens the
+    #: run produced (and once with the prefill's first token).  When set,
+    #: on_token is NOT called: one callback per host visit instead of one
+    #: per token keeps the asyncio side from drowning in per-token tasks
+    #: at large batch (profiles/r01_bench256_gaps.md).
+    on_tokens: Callable[[list[int]], None] | None = None
 
 
 @dataclass
@@ -352,6 +358,7 @@ class InferenceEngine:
         self._stats["prefills"] += 1
         self._stats["prefill_ms"] += seq.result.prefill_ms
         self._emit_token(seq, first)
+        self._emit_batch(seq, [first])
         if not self._finish_if_done(seq):
             self._running.append(seq)
 
@@ -433,6 +440,7 @@ class InferenceEngine:
             s.result.prompt_tokens = len(s.request.prompt_ids)
             s.result.prefill_ms += ms * n / max(1, total_new)  # by share
             self._emit_token(s, tok)
+            self._emit_batch(s, [tok])
             if not self._finish_if_done(s):
                 self._running.append(s)
 
@@ -618,8 +626,11 @@ class InferenceEngine:
         for b, seq in enumerate(self._running):
             seq.result.decode_ms += dt
             done = False
+            emitted: list[int] = []
             for i in range(M):
-                self._emit_token(seq, int(sampled[i, b]))
+                t = int(sampled[i, b])
+                self._emit_token(seq, t)
+                emitted.append(t)
                 done = self._check_done(seq)
                 if done:
                     overshoot = M - 1 - i
@@ -627,6 +638,7 @@ class InferenceEngine:
                         self.kv.truncate(seq.seq_id,
                                          self.kv.length(seq.seq_id) - overshoot)
                     break
+            self._emit_batch(seq, emitted)
             if done:
                 self._finish(seq)
             else:
@@ -638,12 +650,23 @@ class InferenceEngine:
         seq.last_token = token
         seq.new_tokens += 1
         seq.result.token_ids.append(token)
+        if seq.request.on_tokens is not None:
+            return  # batched delivery at the call site
         cb = seq.request.on_token
         if cb is not None:
             try:
                 cb(token)
             except Exception:
                 _log.exception("on_token callback failed")
+
+    @staticmethod
+    def _emit_batch(seq: _Seq, tokens: list[int]) -> None:
+        cb = seq.request.on_tokens
+        if cb is not None and tokens:
+            try:
+                cb(tokens)
+            except Exception:
+                _log.exception("on_tokens callback failed")
 
     def _check_done(self, seq: _Seq) -> bool:
         r = seq.request

@@ -151,11 +151,13 @@ class LocalModelClient:
         s = request.settings
         seq = 0
 
-        def on_token(t: int) -> None:
+        def on_tokens(ts: list[int]) -> None:
+            # one chunk per decode RUN (the engine's host-visit granularity)
+            # — per-token chunks flooded the asyncio loop at large batch
             nonlocal seq
             seq += 1
             sink.emit(StreamChunk(event="text_delta", seq=seq,
-                                  text=tok.decode([t]) + " "))
+                                  text=tok.decode(ts) + " "))
 
         gen = GenRequest(
             prompt_ids=ids,
@@ -164,7 +166,7 @@ class LocalModelClient:
             top_k=s.top_k,
             seed=s.seed or 0,
             ignore_eos=False,
-            on_token=on_token,
+            on_tokens=on_tokens,
         )
         fut = engine.submit(gen)
         result = await asyncio.wrap_future(fut)
