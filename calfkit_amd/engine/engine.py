@@ -81,6 +81,17 @@ class _Seq:
     new_tokens: int = 0
 
 
+@dataclass
+class _PendingPrefill:
+    """A prefill forward launched (possibly on the side stream) whose host
+    finalization — first-token readback, bookkeeping, admission — is
+    deferred until after the overlapped decode run."""
+    seqs: list
+    lens: list
+    sampled: "torch.Tensor"  # [B] int32 on device
+    t0: float
+
+
 class InferenceEngine:
     #: padded decode batch sizes that get their own captured graph
     GRAPH_BATCHES = (1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256)
@@ -159,6 +170,18 @@ class InferenceEngine:
         #: one page sacrificed as the write target of padding rows
         self._dummy_page = self.kv.num_pages - 1
         self.kv._free.remove(self._dummy_page)
+
+        # ---- prefill/decode overlap: waiting prompts prefill on a side
+        # stream WHILE the decode graph replays (prefill is MFMA-bound,
+        # decode HBM-bound — complementary).  Disjoint state: new seqs'
+        # pages + prefill scratch vs running seqs' pages + decode buffers;
+        # sampling gets its own candidate scratch so a concurrent decode
+        # graph at the same batch size cannot race the shared one.
+        self._s2 = torch.cuda.Stream(device=dev) if self.on_gpu else None
+        self._prefill_cand = (torch.empty(B * 64 * 2, dtype=torch.float32,
+                                          device=dev)
+                              if self.on_gpu else None)
+        self._pending: "_PendingPrefill | None" = None
 
         self._queue: "queue.Queue[_Seq | None]" = queue.Queue()
         self._running: list[_Seq] = []
@@ -244,14 +267,26 @@ class InferenceEngine:
             if op[0] == "stop":
                 return
             if op[0] == "prefill":
+                if self._pending is not None:  # safety: rank 0 never does this
+                    p, self._pending = self._pending, None
+                    self._prefill_finalize(p)
                 seqs = []
                 for sid, ids, mx, temp, topk, seed in op[1]:
                     req = GenRequest(prompt_ids=list(ids), max_new_tokens=mx,
                                      temperature=temp, top_k=topk, seed=seed)
                     seqs.append(_Seq(seq_id=sid, request=req, future=Future()))
-                self._prefill_batch(seqs)
+                # mirror rank 0's deterministic overlap rule (lockstep state
+                # is identical, so every rank takes the same branch)
+                if self._overlap_ok(seqs):
+                    with torch.cuda.stream(self._s2):
+                        self._pending = self._prefill_launch(seqs)
+                else:
+                    self._prefill_batch(seqs)
             elif op[0] == "decode":
                 self._decode_step()
+                if self._pending is not None:
+                    p, self._pending = self._pending, None
+                    self._prefill_finalize(p)
 
     # ------------------------------------------------------------------ loop
     def _loop(self) -> None:
@@ -290,6 +325,12 @@ class InferenceEngine:
                             (b.seq_id, b.request.prompt_ids,
                              b.request.max_new_tokens, b.request.temperature,
                              b.request.top_k, b.request.seed) for b in batch]))
+                        if self._overlap_ok(batch):
+                            # side-stream launch; the decode run below hides
+                            # it — finalize (readback + admit) comes after
+                            with torch.cuda.stream(self._s2):
+                                self._pending = self._prefill_launch(batch)
+                            break
                         self._prefill_batch(batch)
                     elif self._waiting:
                         solo = self._waiting.pop(0)
@@ -302,9 +343,14 @@ class InferenceEngine:
                 if self._running:
                     self._tp_bcast(("decode",))
                     self._decode_step()
+                if self._pending is not None:
+                    p, self._pending = self._pending, None
+                    self._prefill_finalize(p)
             except Exception as e:
                 _log.exception("engine step failed")
-                for seq in self._running + self._waiting:
+                pending = self._pending.seqs if self._pending else []
+                self._pending = None
+                for seq in self._running + self._waiting + pending:
                     if not seq.future.done():
                         seq.future.set_exception(e)
                     self.kv.release(seq.seq_id)
@@ -362,15 +408,35 @@ class InferenceEngine:
         if not self._finish_if_done(seq):
             self._running.append(seq)
 
+    def _overlap_ok(self, seqs: list["_Seq"]) -> bool:
+        """Prefill rides the side stream iff there is a decode run to hide
+        it behind and the batch is not the chunked-solo long-prompt path.
+
+        TP/EP lockstep engines never overlap: prefill and decode both issue
+        RCCL collectives, and two streams' collectives interleaving
+        differently across ranks is a classic deadlock."""
+        import os
+
+        return (self._s2 is not None and bool(self._running)
+                and not self.tp_lockstep
+                and os.environ.get("CALF_PREFILL_OVERLAP", "1") != "0"
+                and not (len(seqs) == 1
+                         and len(seqs[0].request.prompt_ids) > PREFILL_CHUNK))
+
     def _prefill_batch(self, seqs: list["_Seq"]) -> None:
-        """Pack every waiting prompt into ONE varlen prefill forward.
+        """Synchronous prefill: launch + finalize in one call."""
+        if len(seqs) == 1 and len(seqs[0].request.prompt_ids) > PREFILL_CHUNK:
+            return self._prefill(seqs[0])
+        self._prefill_finalize(self._prefill_launch(seqs))
+
+    def _prefill_launch(self, seqs: list["_Seq"]) -> _PendingPrefill:
+        """Pack every waiting prompt into ONE varlen prefill forward and
+        enqueue it (forward + first-token sampling) WITHOUT a host sync.
 
         Prefix-cached prompts prefill only their uncached SUFFIX: the cached
         pages are gathered into the contiguous scratch as the per-sequence
         past, and the new rows scatter into their packed positions.
         """
-        if len(seqs) == 1 and len(seqs[0].request.prompt_ids) > PREFILL_CHUNK:
-            return self._prefill(seqs[0])
         if self.on_gpu:  # rocprofv3-visible phase marker (roctx via nvtx shim)
             torch.cuda.nvtx.range_push(f"calfkit.prefill[{len(seqs)}]")
         t0 = time.perf_counter()
@@ -427,16 +493,27 @@ class InferenceEngine:
             logits_rows=(q_starts[1:].long() - 1))
         for s in seqs:
             self.kv.register_prefix(s.seq_id, s.request.prompt_ids)
-        firsts = self._sample_batch_host(logits, seqs)
-        if self.on_gpu:
-            torch.cuda.synchronize(self.device)
-        ms = (time.perf_counter() - t0) * 1e3
+        sampled = self._sample_launch(logits, seqs)
         if self.on_gpu:
             torch.cuda.nvtx.range_pop()
-        self._stats["prefills"] += len(seqs)
+        return _PendingPrefill(seqs=seqs, lens=lens, sampled=sampled, t0=t0)
+
+    def _prefill_finalize(self, p: _PendingPrefill) -> None:
+        """Host side of a launched prefill: read back first tokens, account,
+        emit, admit to the running batch.  Under overlap this runs after
+        the decode replay, so ``prefill_ms`` measures launch→finalize wall
+        (it overstates GPU prefill time when overlapped — that is the
+        point)."""
+        if self._s2 is not None:
+            self._s2.synchronize()  # no-op when launched on the default stream
+        firsts = [int(t) for t in p.sampled.cpu()]
+        if self.on_gpu:
+            torch.cuda.synchronize(self.device)
+        ms = (time.perf_counter() - p.t0) * 1e3
+        self._stats["prefills"] += len(p.seqs)
         self._stats["prefill_ms"] += ms
-        total_new = sum(lens)
-        for s, n, tok in zip(seqs, lens, firsts):
+        total_new = sum(p.lens)
+        for s, n, tok in zip(p.seqs, p.lens, firsts):
             s.result.prompt_tokens = len(s.request.prompt_ids)
             s.result.prefill_ms += ms * n / max(1, total_new)  # by share
             self._emit_token(s, tok)
@@ -472,8 +549,11 @@ class InferenceEngine:
             kbuf[rows] = kc[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)
             vbuf[rows] = vc[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)
 
-    def _sample_batch_host(self, logits: torch.Tensor,
-                           seqs: list["_Seq"]) -> list[int]:
+    def _sample_launch(self, logits: torch.Tensor,
+                       seqs: list["_Seq"]) -> torch.Tensor:
+        """Enqueue first-token sampling; returns the device tensor (no
+        sync).  Uses the engine's own candidate scratch so it can run on
+        the side stream concurrently with a decode graph."""
         from .. import ops
 
         B = len(seqs)
@@ -486,8 +566,9 @@ class InferenceEngine:
             [seqs[0].request.seed or seqs[0].seq_id, self._step_counter],
             dtype=torch.int64).to(logits.device)
         ops.sample(out, logits, temp, seed,
-                   topk if any(s.request.top_k for s in seqs) else None)
-        return [int(t) for t in out.cpu()]
+                   topk if any(s.request.top_k for s in seqs) else None,
+                   cand=self._prefill_cand)
+        return out
 
     def _sample_host(self, logits: torch.Tensor, seq: _Seq) -> int:
         r = seq.request

@@ -252,16 +252,23 @@ _SAMPLE_SCRATCH: dict[tuple, torch.Tensor] = {}
 
 
 def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
-           seed_step: torch.Tensor, topk: torch.Tensor | None = None) -> None:
+           seed_step: torch.Tensor, topk: torch.Tensor | None = None,
+           cand: torch.Tensor | None = None) -> None:
     """Gumbel-max sampling; temperature 0 = greedy. out: int32 [B].
-    Accepts fp32 or bf16 logits (bf16 avoids materializing [B,V] fp32)."""
+    Accepts fp32 or bf16 logits (bf16 avoids materializing [B,V] fp32).
+    ``cand`` overrides the shared two-stage scratch — callers running on a
+    side stream (overlapped prefill) MUST pass their own so a concurrent
+    decode graph replaying the same batch size cannot race on it."""
     if _native(logits):
-        key = (logits.device.index, int(out.shape[0]))
-        cand = _SAMPLE_SCRATCH.get(key)
-        if cand is None or cand.numel() < out.shape[0] * 64 * 2:
-            cand = torch.empty(out.shape[0] * 64 * 2, dtype=torch.float32,
-                               device=logits.device)
-            _SAMPLE_SCRATCH[key] = cand
+        if cand is None:
+            key = (logits.device.index, int(out.shape[0]))
+            cand = _SAMPLE_SCRATCH.get(key)
+            if cand is None or cand.numel() < out.shape[0] * 64 * 2:
+                cand = torch.empty(out.shape[0] * 64 * 2, dtype=torch.float32,
+                                   device=logits.device)
+                _SAMPLE_SCRATCH[key] = cand
+        else:
+            assert cand.numel() >= out.shape[0] * 64 * 2
         _K.sample(out, logits, temperature,
                   seed_step, topk if topk is not None else torch.Tensor(),
                   cand)
