@@ -148,3 +148,37 @@ def test_long_prompt_chunked_prefill_matches_short_path():
         assert r_long.token_ids == r_pack.token_ids
     finally:
         eng_mod.PREFILL_CHUNK, eng_mod.PREFILL_BUDGET = old_chunk, old_budget
+
+
+@pytest.mark.gpu
+def test_prefill_overlap_matches_serial_gpu():
+    """Side-stream prefill overlap must not change greedy outputs: the
+    same staggered request stream produces identical tokens with the
+    overlap forced off (CALF_PREFILL_OVERLAP=0) and on."""
+    import os
+    import time as _time
+
+    from calfkit_amd.models.llama import LlamaShard
+
+    def run(overlap: bool) -> list[list[int]]:
+        os.environ["CALF_PREFILL_OVERLAP"] = "1" if overlap else "0"
+        try:
+            model = LlamaShard(TINY, device="cuda")
+            eng = InferenceEngine(model, max_batch=8, max_model_len=256)
+            try:
+                futs = [eng.submit(GenRequest(
+                    prompt_ids=[1, 3 + i, 7, 11 + i], max_new_tokens=24))
+                    for i in range(4)]
+                _time.sleep(0.05)  # second wave arrives mid-decode
+                futs += [eng.submit(GenRequest(
+                    prompt_ids=[2, 5 + i, 9], max_new_tokens=16))
+                    for i in range(4)]
+                return [f.result(120).token_ids for f in futs]
+            finally:
+                eng.stop()
+        finally:
+            os.environ.pop("CALF_PREFILL_OVERLAP", None)
+
+    serial = run(False)
+    overlapped = run(True)
+    assert serial == overlapped
