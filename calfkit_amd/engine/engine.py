@@ -417,9 +417,16 @@ class InferenceEngine:
         differently across ranks is a classic deadlock."""
         import os
 
+        # Default OFF: at the batch-256 operating point the A/B measured a
+        # wash (the aggregate sits at the HBM roofline — see
+        # profiles/r01_bench256_gaps.md), and overlapped admission shifts
+        # decode batch shapes, whose library-GEMM tilings round
+        # differently — greedy outputs then depend on arrival timing.
+        # Opt in with CALF_PREFILL_OVERLAP=1 for latency-sensitive mixed
+        # loads where hiding prefill behind decode matters more.
         return (self._s2 is not None and bool(self._running)
                 and not self.tp_lockstep
-                and os.environ.get("CALF_PREFILL_OVERLAP", "1") != "0"
+                and os.environ.get("CALF_PREFILL_OVERLAP", "0") == "1"
                 and not (len(seqs) == 1
                          and len(seqs[0].request.prompt_ids) > PREFILL_CHUNK))
 
