@@ -5,8 +5,11 @@
 // with KVPAGE == 64 so one (page, head) panel is a contiguous 16 KB run the
 // wave streams line-aligned.
 //
-// Decomposition: one workgroup per (sequence, kv_head); 4 waves stride the
-// sequence's pages.  Within a wave, LANE == TOKEN for the QK phase (each
+// Decomposition: one workgroup per (sequence, kv_head); NW waves stride the
+// sequence's pages.  NW is a template knob: agent decode lives at 2-4
+// pages per sequence, where 4-wave blocks idle 1-2 wave slots per CU —
+// 2-wave blocks pack twice as many blocks per CU and keep every slot hot
+// (env CALF_ATTN_NW overrides; default set by the r31 A/B).  Within a wave, LANE == TOKEN for the QK phase (each
 // lane owns one key row, so the G per-head scores are lane-local dots — no
 // cross-lane reduction), and LANE == DIM-PAIR for the PV phase (p broadcast
 // by shuffle, V row read coalesced 4 B/lane).  Online softmax per wave with
@@ -20,10 +23,9 @@
 
 #define KVPAGE 64
 #define HD 128
-#define NWAVES 4
 
-template <int G>
-__global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
+template <int G, int NW>
+__global__ __launch_bounds__(NW * WAVE) void attn_decode4_kernel(
     __hip_bfloat16* __restrict__ out,
     const __hip_bfloat16* __restrict__ q,
     const __hip_bfloat16* __restrict__ kcache,
@@ -40,10 +42,10 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
   const int npages = (slen + KVPAGE - 1) / KVPAGE;
 
   __shared__ float q_lds[G][HD];
-  __shared__ float o_lds[NWAVES][G][HD];
-  __shared__ float ml_lds[NWAVES][G][2];
+  __shared__ float o_lds[NW][G][HD];
+  __shared__ float ml_lds[NW][G][2];
 
-  for (int i = threadIdx.x; i < G * HD; i += NWAVES * WAVE) {
+  for (int i = threadIdx.x; i < G * HD; i += NW * WAVE) {
     const int g = i / HD, d = i % HD;
     q_lds[g][d] =
         bf2f(q[(int64_t)seq * q_stride + (kvh * G + g) * HD + d]) * scale;
@@ -65,7 +67,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
   const int thalf = lane >> 5;        // 0: tokens 0..31, 1: tokens 32..63
 
   const int64_t tbl = (int64_t)seq * max_pages;
-  for (int pi = wid; pi < npages; pi += NWAVES) {
+  for (int pi = wid; pi < npages; pi += NW) {
     // bar the compiler from hoisting the per-page q_lds re-reads out of
     // this loop: without it LICM keeps G*128 floats live across pages
     // (251-747 VGPR spills measured; see the r01/r03 kernel profiles)
@@ -195,10 +197,10 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
     for (int g = 0; g < G; ++g) {
       float M = -INFINITY;
 #pragma unroll
-      for (int w = 0; w < NWAVES; ++w) M = fmaxf(M, ml_lds[w][g][0]);
+      for (int w = 0; w < NW; ++w) M = fmaxf(M, ml_lds[w][g][0]);
       float L = 0.f, o[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int w = 0; w < NWAVES; ++w) {
+      for (int w = 0; w < NW; ++w) {
         const float mw = ml_lds[w][g][0];
         const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
         L += sc * ml_lds[w][g][1];
@@ -217,6 +219,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void attn_decode4_kernel(
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
 
 void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                  torch::Tensor vcache, torch::Tensor page_table,
@@ -234,10 +237,12 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   TORCH_CHECK(dh == HD, "decode kernel assumes head_dim == 128");
   TORCH_CHECK(page == KVPAGE, "decode kernel assumes page_size == 64");
   const int G = hq / hkv;
-  dim3 grid(B, hkv), block(NWAVES * WAVE);
+  const char* nw_env = getenv("CALF_ATTN_NW");
+  const int nw = nw_env ? atoi(nw_env) : 2;
+  dim3 grid(B, hkv), block(nw * WAVE);
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
-#define LAUNCH(GV)                                                        \
-  hipLaunchKernelGGL(attn_decode4_kernel<GV>, grid, block, 0, s,          \
+#define LAUNCH_NW(GV, NWV)                                                \
+  hipLaunchKernelGGL((attn_decode4_kernel<GV, NWV>), grid, block, 0, s,   \
                      (__hip_bfloat16*)out.data_ptr(),                     \
                      (const __hip_bfloat16*)q.data_ptr(),                 \
                      (const __hip_bfloat16*)kcache.data_ptr(),            \
@@ -245,6 +250,11 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
                      (const int32_t*)page_table.data_ptr(),               \
                      (const int32_t*)seq_lens.data_ptr(), hq, hkv,        \
                      max_pages, (float)scale, (int)q.stride(0))
+#define LAUNCH(GV)                                                        \
+  do {                                                                    \
+    if (nw == 4) LAUNCH_NW(GV, 4);                                        \
+    else LAUNCH_NW(GV, 2);                                                \
+  } while (0)
   switch (G) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -253,5 +263,6 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH
+#undef LAUNCH_NW
   HIP_CHECK_LAST();
 }
