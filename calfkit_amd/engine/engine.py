@@ -94,7 +94,7 @@ class _PendingPrefill:
 
 class InferenceEngine:
     #: padded decode batch sizes that get their own captured graph
-    GRAPH_BATCHES = (1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256, 384, 512)
+    GRAPH_BATCHES = (1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256, 384, 512, 768, 1024)
 
     def __init__(self, model: LlamaShard, *, max_batch: int = 64,
                  kv_pages: int | None = None, use_graphs: bool | None = None,
