@@ -286,7 +286,13 @@ def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
         if t <= 0:
             out[b] = int(lg[b].argmax())
         else:
-            g = torch.distributions.Gumbel(0.0, 1.0).sample((V,))
+            # counter-based like the HIP kernel: (seed, step, row) fully
+            # determine the draw — same request seed => same tokens
+            key = (int(seed_step[0]) * 1_000_003
+                   + int(seed_step[1]) * 7919 + b) & 0x7FFFFFFF
+            gen = torch.Generator().manual_seed(key)
+            u = torch.rand(V, generator=gen).clamp_min(1e-12)
+            g = -torch.log(-torch.log(u))
             out[b] = int((lg[b] / t + g).argmax())
 
 

@@ -181,3 +181,49 @@ def test_prefill_overlap_completes_gpu():
     finally:
         os.environ.pop("CALF_PREFILL_OVERLAP", None)
     assert [len(o) for o in outs] == [24] * 4 + [16] * 4
+
+
+def test_on_tokens_batched_delivery_matches_result():
+    """on_tokens delivers every generated token exactly once, in order,
+    batched per host visit (first chunk = the prefill's first token)."""
+    eng = make_engine()
+    try:
+        chunks: list[list[int]] = []
+        r = eng.generate(GenRequest(prompt_ids=[1, 5, 9], max_new_tokens=11,
+                                    on_tokens=chunks.append), timeout=120)
+        flat = [t for c in chunks for t in c]
+        assert flat == r.token_ids
+        assert len(r.token_ids) == 11
+        assert chunks[0] == r.token_ids[:1]  # prefill's first token alone
+        assert all(c for c in chunks)
+    finally:
+        eng.stop()
+
+
+def test_on_tokens_suppresses_per_token_callback():
+    eng = make_engine()
+    try:
+        per_token: list[int] = []
+        chunks: list[list[int]] = []
+        eng.generate(GenRequest(prompt_ids=[1, 2], max_new_tokens=4,
+                                on_token=per_token.append,
+                                on_tokens=chunks.append), timeout=120)
+        assert per_token == []  # batched delivery wins
+        assert sum(len(c) for c in chunks) == 4
+    finally:
+        eng.stop()
+
+
+def test_seeded_sampling_reproducible_across_engines():
+    """temperature>0 with a fixed seed: two engine instances produce the
+    same tokens (counter-based Gumbel sampling, no hidden RNG state)."""
+    def run() -> list[int]:
+        eng = make_engine()
+        try:
+            return eng.generate(GenRequest(prompt_ids=[1, 4, 7],
+                                           max_new_tokens=8, temperature=0.8,
+                                           seed=42), timeout=120).token_ids
+        finally:
+            eng.stop()
+
+    assert run() == run()
