@@ -129,3 +129,29 @@ async def test_emitter_headers_present():
     assert rec.header("x-calf-emitter") == b"emitter-node"
     assert rec.header("x-calf-hop") == b"7"
     assert rec.header("x-calf-kind") == b"agent_call"
+
+
+def test_envelope_golden_bytes():
+    """Wire-format pin: the exact bytes and headers of a canonical agent
+    call.  This is the byte-compatibility contract for Kafka (and the
+    bundled mesh, which carries the same records verbatim) — any drift in
+    field names, ordering, separators or header keys breaks consumers and
+    MUST show up here as a deliberate change."""
+    value, headers = encode_envelope(
+        "agent_call", to="researcher", thread_id="t-1",
+        payload={"message": {"role": "user",
+                             "parts": [{"type": "text", "text": "hi"}]}},
+        reply=ReplySlot(inbox="calf.reply.abc", key="t-1", node="client"),
+        marker=EchoMarker(call_id="c-5", frame_id="f-9", caller="boss"),
+        emitter="client-1", hop=2)
+    assert value == (
+        b'{"routing":{"to":"researcher","thread_id":"t-1",'
+        b'"reply":{"inbox":"calf.reply.abc","key":"t-1","node":"client"}},'
+        b'"state":{"marker":{"call_id":"c-5","frame_id":"f-9",'
+        b'"caller":"boss"},"hop":2},'
+        b'"payload":{"message":{"role":"user",'
+        b'"parts":[{"type":"text","text":"hi"}]}}}'
+    ), value
+    assert [(k if isinstance(k, str) else k.decode()) for k, _ in headers] \
+        == ["x-calf-kind", "x-calf-emitter", "x-calf-hop"]
+    assert [v for _, v in headers] == [b"agent_call", b"client-1", b"2"]
