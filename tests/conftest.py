@@ -20,6 +20,9 @@ def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: needs a real MI355X (run with -m gpu)")
     config.addinivalue_line("markers", "dist: multi-process distributed test")
     config.addinivalue_line("markers", "slow: long-running test")
+    config.addinivalue_line(
+        "markers",
+        "kafka: needs a live broker (CALF_TEST_KAFKA_BOOTSTRAP)")
 
 
 def _cuda_available() -> bool:
