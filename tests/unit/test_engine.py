@@ -227,3 +227,19 @@ def test_seeded_sampling_reproducible_across_engines():
             eng.stop()
 
     assert run() == run()
+
+
+def test_large_batch_rounding_to_new_graph_sizes():
+    """max_batch rounds up to the new 384/512/1024 graph sizes and the
+    pad table never exceeds max_batch."""
+    eng = make_engine(max_batch=300)
+    try:
+        assert eng.max_batch == 384
+        assert eng._pad_batch(3) == 4
+        assert eng._pad_batch(257) == 384
+        assert eng._pad_batch(384) == 384
+        r = eng.generate(GenRequest(prompt_ids=[1, 2, 3], max_new_tokens=3),
+                         timeout=120)
+        assert len(r.token_ids) == 3
+    finally:
+        eng.stop()
