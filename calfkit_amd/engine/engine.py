@@ -217,7 +217,13 @@ class InferenceEngine:
 
     @property
     def stats(self) -> dict:
-        return dict(self._stats)
+        s = dict(self._stats)
+        # KV/prefix-cache telemetry (observability parity with the step
+        # ledger: the serving side of the mesh view)
+        s["kv_pages_total"] = self.kv.num_pages
+        s["kv_pages_free"] = self.kv.free_pages
+        s["prefix_hit_tokens"] = self.kv.prefix_hit_tokens
+        return s
 
     # ---------------------------------------------------------------- submit
     def submit(self, request: GenRequest) -> Future:
