@@ -42,10 +42,10 @@ __all__ = ["GenRequest", "GenResult", "InferenceEngine"]
 PREFILL_CHUNK = 512
 PREFILL_BUDGET = 8192
 #: decode steps executed per host visit (graphs exist for each length).
-#: 16 front-loaded: at batch 512 each host visit costs ~50 ms (readback
-#: sync + emit + next-run state upload), and agent turns decode in
-#: multiples of 16, so halving the visits is free throughput
-RUN_STEPS = (16, 8, 4, 2, 1)
+#: 8 is the measured sweet spot: 16-step runs halve the per-visit host
+#: cost but delay admission by a full run, and the running batch dips
+#: enough to LOSE throughput end-to-end (233 -> 210 turns/s at 512)
+RUN_STEPS = (8, 4, 2, 1)
 
 
 @dataclass
@@ -685,9 +685,9 @@ class InferenceEngine:
         """One hipGraph per (padded batch, run length), largest first."""
         torch.cuda.synchronize(self.device)
         for b in reversed([g for g in self.GRAPH_BATCHES if g <= self.max_batch]):
-            # full-length runs for every pad; short tail runs only for
-            # small pads — large-batch tails are rare and run eager
-            run_lens = RUN_STEPS if b <= 32 else RUN_STEPS[:2]
+            # full-length runs for every pad; short tail runs (M < 8) only
+            # for small pads — large-batch tails are rare and run eager
+            run_lens = RUN_STEPS if b <= 32 else RUN_STEPS[:1]
             for m in run_lens:
                 self._decode_run_gpu(b, m)  # warm up allocations
                 torch.cuda.synchronize(self.device)
