@@ -221,3 +221,41 @@ async def test_invalid_tool_args_rejected_before_the_wire():
         assert r.text == "fixed it"
     tool_msgs = [m for m in provider.requests[-1].messages if m.role == "tool"]
     assert "invalid arguments" in str(tool_msgs[0].parts[0].content)
+
+
+async def test_handoff_chain_three_agents():
+    """A -> B -> C handoff chain: the thread follows to the terminal agent
+    and the caller gets C's answer (reference: handoff follows across
+    turns, CHANGELOG.md:47-area behavior)."""
+    bus, store, directory = mesh()
+    a = Agent(name="triage", model_client=FakeProvider([
+        call_tool("handoff", to="router", reason="route it"),
+    ]), peers=[Handoff(names=("router",))])
+    b = Agent(name="router", model_client=FakeProvider([
+        call_tool("handoff", to="expert", reason="needs the expert"),
+    ]), peers=[Handoff(names=("expert",))])
+    c = Agent(name="expert", model_client=FakeProvider(["expert verdict"]))
+    async with Worker(a, b, c, bus=bus, store=store, directory=directory):
+        client = await Client.connect(bus, directory)
+        result = await client.execute("triage", "deep question", timeout=5)
+        assert result.text == "expert verdict"
+
+
+async def test_handoff_followup_turn_reaches_new_owner():
+    """After a handoff, the NEXT turn on the same thread goes to the new
+    owner directly (the thread moved, not just the single reply)."""
+    bus, store, directory = mesh()
+    front = Agent(name="desk", model_client=FakeProvider([
+        call_tool("handoff", to="owner", reason="yours now"),
+    ]), peers=[Handoff(names=("owner",))])
+    owner = Agent(name="owner", model_client=FakeProvider([
+        "taking over", "still me on turn 2",
+    ]))
+    async with Worker(front, owner, bus=bus, store=store, directory=directory):
+        client = await Client.connect(bus, directory)
+        r1 = await client.execute("desk", "start", timeout=5)
+        assert r1.text == "taking over"
+        target = r1.agent if getattr(r1, "agent", None) else "owner"
+        r2 = await client.execute(target, "follow-up",
+                                  thread_id=r1.thread_id, timeout=5)
+        assert r2.text == "still me on turn 2"
