@@ -43,6 +43,12 @@ void moe_gateup_silu(torch::Tensor act, torch::Tensor x, torch::Tensor w,
                      torch::Tensor tok, torch::Tensor off, torch::Tensor work);
 void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w,
               torch::Tensor off, torch::Tensor scale);
+std::vector<std::tuple<int64_t, double>> hipblaslt_algo_bench(
+    torch::Tensor out, torch::Tensor x, torch::Tensor w, int64_t iters,
+    int64_t max_algos);
+void hipblaslt_gemm_indexed(torch::Tensor out, torch::Tensor x,
+                            torch::Tensor w, int64_t algo_index,
+                            torch::Tensor workspace);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
@@ -72,4 +78,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped MoE gate-up: act[r] = silu/mul of x[tok[r]] @ W[e(r)]^T");
   m.def("moe_down", &moe_down,
         "grouped MoE down: y[r] = scale[r] * act[r] @ W[e(r)]^T");
+  m.def("hipblaslt_algo_bench", &hipblaslt_algo_bench,
+        "time every supported hipBLASLt algo for out = x @ w^T");
+  m.def("hipblaslt_gemm_indexed", &hipblaslt_gemm_indexed,
+        "run out = x @ w^T with a searched hipBLASLt algo index");
 }

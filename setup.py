@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "sampling.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "moe_gemm.hip"),
+    os.path.join(CSRC, "gemm_search.hip"),
 ]
 
 setup(
@@ -37,6 +38,7 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
             },
+            libraries=["hipblaslt"],
         )
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
