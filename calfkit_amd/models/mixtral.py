@@ -42,8 +42,9 @@ __all__ = ["MixtralShard", "MoeLayerWeights"]
 
 @dataclass
 class MoeLayerWeights(LayerWeights):
-    # wgateup/wdown become stacked per-expert: [E, hidden, 2*ffn_s], [E, ffn_s, hidden]
-    router: torch.Tensor = None  # [hidden, E]
+    # wgateup/wdown become stacked per-expert (nn.Linear [out, in] layout):
+    # wgateup [E_loc, 2*ffn_s, hidden], wdown [E_loc, hidden, ffn_s]
+    router: torch.Tensor = None  # [E, hidden]
 
 
 class MixtralShard(LlamaShard):
