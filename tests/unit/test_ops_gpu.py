@@ -460,3 +460,22 @@ def test_moe_decode_graph_capture():
     eager = model._mlp(lw, x2.clone(), decode=True)
     assert torch.allclose(out.float(), eager.float(), atol=5e-2, rtol=5e-2), \
         (out.float() - eager.float()).abs().max()
+
+
+def test_hipblaslt_algo_search_and_indexed_replay():
+    """The offline GEMM algo-search binding: enumeration returns timed
+    algos and replaying the best index reproduces F.linear numerics."""
+    from calfkit_amd.ops import _C  # type: ignore
+
+    M, N, K = 64, 256, 512
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.05
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    timed = _C.hipblaslt_algo_bench(out, x, w, 3, 8)
+    assert timed, "no supported hipBLASLt algos for a plain bf16 GEMM"
+    best = min(timed, key=lambda t: t[1])[0]
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    _C.hipblaslt_gemm_indexed(out, x, w, best, ws)
+    torch.cuda.synchronize()
+    ref = torch.nn.functional.linear(x, w)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
