@@ -17,7 +17,9 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
-async def run(turns: int = 200, concurrency: int = 16) -> None:
+async def run(turns: int | None = None, concurrency: int | None = None) -> None:
+    turns = turns or int(os.environ.get("C1_TURNS", "200"))
+    concurrency = concurrency or int(os.environ.get("C1_CONCURRENCY", "16"))
     from calfkit_amd import Agent, Client, Worker, agent_tool
     from calfkit_amd.adapters.memory import InProcessThreadStore
     from calfkit_amd.adapters.mesh import MeshBroker, connect_mesh
