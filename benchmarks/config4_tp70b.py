@@ -76,11 +76,13 @@ async def run_rank0(dist, cfg) -> None:
 
         for w in range(WARMUP):
             await step(-1 - w, False)
-        torch.cuda.synchronize()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
         t0 = time.perf_counter()
         for s in range(STEPS):
             await step(s, True)
-        torch.cuda.synchronize()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
         wall = time.perf_counter() - t0
         provider.engine.stop()
 
@@ -103,7 +105,15 @@ def main() -> None:
     from calfkit_amd.parallel import init_distributed
 
     dist = init_distributed()
-    cfg = get_config("llama-3-70b" if torch.cuda.is_available() else "tiny")
+    if torch.cuda.is_available():
+        cfg = get_config("llama-3-70b")
+    else:
+        # sanity mode: tiny widened so heads/kv/ffn shard over the world
+        from dataclasses import replace
+
+        cfg = replace(get_config("tiny"), heads=2 * dist.world_size,
+                      kv_heads=dist.world_size, ffn=256 * dist.world_size,
+                      name="tiny")
     if dist.rank == 0:
         asyncio.run(run_rank0(dist, cfg))
     else:

@@ -47,3 +47,28 @@ def test_bench_default_contract_cpu():
                   "dtype", "data", "config"):
         assert field in r, field
     assert r["n_gpus"] == 1 and r["scaling"] == "weak"
+
+
+@pytest.mark.slow
+def test_config3_toolcalls_script_cpu():
+    r = _run("config3_toolcalls.py",
+             {"C3_TURNS": "6", "C3_STEPS": "1", "C3_WARMUP": "0"})
+    assert r["unit"] == "turns/s" and r["value"] > 0
+
+
+@pytest.mark.slow
+@pytest.mark.dist
+def test_config4_tp_script_cpu_world2():
+    """config4 (TP lockstep serving) under torchrun world=2 on gloo."""
+    env = dict(os.environ, C4_TURNS="4", C4_STEPS="1", C4_WARMUP="0")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29547", os.path.join(ROOT, "benchmarks",
+                                                "config4_tp70b.py")],
+        capture_output=True, text=True, timeout=420, env=env, cwd=ROOT)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert lines, out.stdout[-2000:]
+    r = json.loads(lines[-1])
+    assert r["value"] > 0 and r["n_gpus"] == 2
