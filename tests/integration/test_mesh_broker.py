@@ -114,3 +114,19 @@ async def test_dead_worker_adverts_leave_the_directory():
         assert (await client.mesh()).get("mortal") is None
     finally:
         await broker.stop()
+
+
+async def test_client_connect_bootstrap_string():
+    """The README path: Client.connect(bootstrap='host:port') dials the
+    broker without pre-built adapters."""
+    broker = await MeshBroker().start()
+    try:
+        bus_w, dir_w = await connect_mesh(broker.address)
+        agent = Agent(name="dialee", model_client=FakeProvider(["hello"]))
+        async with Worker(agent, bus=bus_w, store=InProcessThreadStore(),
+                          directory=dir_w):
+            client = await Client.connect(bootstrap=broker.address)
+            result = await client.execute("dialee", "hi", timeout=10)
+            assert result.text == "hello"
+    finally:
+        await broker.stop()
