@@ -27,7 +27,10 @@ def _run(script: str, env: dict, timeout: int = 300) -> dict:
 @pytest.mark.slow
 def test_config1_plumbing_script():
     r = _run("config1_plumbing.py", {"C1_TURNS": "24", "C1_CONCURRENCY": "8"})
-    assert r["unit"] == "turns/s" and r["value"] > 0
+    assert r["unit"] == "turns/s"
+    # catastrophic-regression floor only (measured ~1900 turns/s; a
+    # machine-speed-safe 20x margin below the slowest plausible CI box)
+    assert r["value"] > 20, r
 
 
 @pytest.mark.slow
