@@ -10,10 +10,19 @@ from __future__ import annotations
 
 import hashlib
 import re
+from functools import lru_cache
 
 __all__ = ["SyntheticTokenizer"]
 
 _WORD = re.compile(r"\S+")
+
+
+@lru_cache(maxsize=65536)
+def _word_id(word: str, span: int) -> int:
+    # cached: agent prompts repeat words heavily (a 512-turn rollover
+    # tokenizes ~65k words; the distinct set is a few hundred)
+    h = hashlib.blake2s(word.encode(), digest_size=4).digest()
+    return int.from_bytes(h, "little") % span
 
 
 class SyntheticTokenizer:
@@ -28,8 +37,7 @@ class SyntheticTokenizer:
         ids = [self.BOS] if bos else []
         span = self.vocab_size - self._RESERVED
         for m in _WORD.finditer(text):
-            h = hashlib.blake2s(m.group().encode(), digest_size=4).digest()
-            ids.append(self._RESERVED + int.from_bytes(h, "little") % span)
+            ids.append(self._RESERVED + _word_id(m.group(), span))
         return ids
 
     def decode(self, ids: list[int]) -> str:
