@@ -53,6 +53,12 @@ class StepLedger:
     async def flush(self, bus: MessageBus | None = None, namespace: str = "") -> dict[str, Any]:
         rec = self.record()
         _log.info("step %s", json.dumps(rec, default=str))
+        from ..utils.metrics import metrics  # no-op shim unless enabled
+
+        metrics.hops.labels(self.node, self.hop_kind,
+                            str(rec.get("outcome", "ok"))).inc()
+        metrics.hop_seconds.labels(self.node, self.hop_kind).observe(
+            rec["hop_ms"] / 1e3)
         if bus is not None:
             try:
                 await bus.publish(topology.ledger_topic(namespace), self.thread_id,

@@ -719,6 +719,10 @@ class InferenceEngine:
         self._stats["decode_steps"] += M
         self._stats["decode_tokens"] += M * B
         self._stats["decode_ms"] += dt
+        from ..utils.metrics import metrics  # no-op shim unless enabled
+
+        metrics.engine_tokens.labels(self.cfg.name, "decode").inc(M * B)
+        metrics.engine_batch.labels(self.cfg.name).set(B)
         still: list[_Seq] = []
         for b, seq in enumerate(self._running):
             seq.result.decode_ms += dt

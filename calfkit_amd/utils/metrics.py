@@ -40,11 +40,11 @@ class Metrics:
         except ImportError:
             return False
         if not self.enabled:
-            self.hops = Counter("calfkit_hops_total", "agent/tool hops",
+            self.hops = Counter("calfkit_hops", "agent/tool hops",
                                 ["node", "kind", "outcome"])
             self.hop_seconds = Histogram("calfkit_hop_seconds", "hop latency",
                                          ["node", "kind"])
-            self.engine_tokens = Counter("calfkit_engine_tokens_total",
+            self.engine_tokens = Counter("calfkit_engine_tokens",
                                          "generated tokens", ["model", "phase"])
             self.engine_batch = Gauge("calfkit_engine_batch", "decode batch",
                                       ["model"])

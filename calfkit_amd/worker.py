@@ -72,6 +72,13 @@ class Worker:
     async def start(self) -> None:
         if self.gpu is not None:
             self._pin_gpu(self.gpu)
+        port = os.environ.get("CALF_METRICS_PORT")
+        if port:  # opt-in Prometheus exporter (utils/metrics.py)
+            from .utils.metrics import metrics
+
+            if not metrics.enable(int(port)):
+                _log.warning("CALF_METRICS_PORT set but prometheus_client "
+                             "is not installed; metrics stay off")
         await self.lifecycle.startup()
         for node in self.nodes:
             processor = host(node, self.machinery)
