@@ -54,6 +54,11 @@ def _daemons() -> dict[str, int]:
 @click.group()
 def cli() -> None:
     """calfkit-amd developer tool."""
+    from ..utils.logging import setup_logging
+
+    # CALF_LOG_JSON=1 turns every calfkit log record into one JSON line
+    # (structured logging; the step ledger rides the same stream)
+    setup_logging(json_lines=os.environ.get("CALF_LOG_JSON", "") == "1")
 
 
 @cli.group()

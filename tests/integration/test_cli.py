@@ -66,3 +66,25 @@ async def test_worker_via_cli_module_against_broker(tmp_path):
             assert r.text == "echo: hi there"
     finally:
         await broker.stop()
+
+
+def test_json_logging_formatter():
+    """CALF_LOG_JSON structured logging: one JSON line per record with
+    stable keys (reference CHANGELOG.md:497 parity)."""
+    import io
+    import json
+    import logging
+
+    from calfkit_amd.utils.logging import setup_logging
+
+    buf = io.StringIO()
+    setup_logging(json_lines=True, stream=buf)
+    try:
+        log = logging.getLogger("calfkit.test")
+        log.info("hello %s", "world", extra={"node": "n1", "thread_id": "t9"})
+        rec = json.loads(buf.getvalue().strip())
+        assert rec["msg"] == "hello world"
+        assert rec["level"] == "info" and rec["logger"] == "calfkit.test"
+        assert rec["node"] == "n1" and rec["thread_id"] == "t9"
+    finally:
+        setup_logging()  # restore default handler for other tests
