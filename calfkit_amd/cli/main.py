@@ -155,13 +155,18 @@ def _load_nodes(spec: str) -> list:
 @click.option("--mesh", "mesh_url", default=DEFAULT_MESH, show_default=True)
 @click.option("--detach/--no-detach", default=False)
 @click.option("--gpu", type=int, default=None, help="pin this worker to a GPU")
-def dev_run(spec: str, mesh_url: str, detach: bool, gpu: int | None) -> None:
+@click.option("--mcp", "mcp_json", type=click.Path(exists=True), default=None,
+              help="also host every toolbox from this mcp.json")
+def dev_run(spec: str, mesh_url: str, detach: bool, gpu: int | None,
+            mcp_json: str | None) -> None:
     """Host the nodes of SPEC (file.py[:name]) as a worker on the mesh."""
     if detach:
         argv = [sys.executable, "-m", "calfkit_amd.cli", "dev", "run", spec,
                 "--mesh", mesh_url]
         if gpu is not None:
             argv += ["--gpu", str(gpu)]
+        if mcp_json:
+            argv += ["--mcp", mcp_json]
         _spawn_daemon(f"worker-{Path(spec.partition(':')[0]).stem}", argv)
         click.echo(f"worker for {spec} detached")
         return
@@ -172,6 +177,10 @@ def dev_run(spec: str, mesh_url: str, detach: bool, gpu: int | None) -> None:
         from ..worker import Worker
 
         nodes = _load_nodes(spec)
+        if mcp_json:
+            from ..utils.mcp_config import load_mcp_config
+
+            nodes.extend(load_mcp_config(mcp_json))
         if not nodes:
             raise click.ClickException(f"{spec}: nothing hostable found")
         bus, directory = await connect_mesh(mesh_url)
