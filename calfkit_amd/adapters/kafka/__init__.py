@@ -47,9 +47,44 @@ def _aiokafka():
 
 
 class _Lane:
-    def __init__(self) -> None:
-        self.queue: asyncio.Queue = asyncio.Queue()
+    def __init__(self, maxsize: int = 0) -> None:
+        self.queue: asyncio.Queue = asyncio.Queue(maxsize=maxsize)
         self.task: asyncio.Task | None = None
+
+
+class _OffsetLedger:
+    """Per-partition low-watermark of HANDLED offsets.
+
+    Offsets are committed only after the lane handler completes, so a worker
+    crash re-delivers in-flight work (at-least-once) instead of silently
+    dropping it the way auto-commit did.  Handlers run
+    out of order across lanes, so completion is tracked as a contiguity
+    watermark per (topic, partition).
+    """
+
+    def __init__(self) -> None:
+        self._next: dict[tuple[str, int], int] = {}
+        self._done: dict[tuple[str, int], set[int]] = {}
+
+    def expect(self, tp: tuple[str, int], offset: int) -> None:
+        """Called in consume order (= offset order within a partition)."""
+        self._next.setdefault(tp, offset)
+
+    def complete(self, tp: tuple[str, int], offset: int) -> int | None:
+        """Mark handled; returns the new commit watermark when it advanced
+        (the offset AFTER the last contiguously handled record)."""
+        done = self._done.setdefault(tp, set())
+        done.add(offset)
+        nxt = self._next[tp]
+        advanced = False
+        while nxt in done:
+            done.remove(nxt)
+            nxt += 1
+            advanced = True
+        if advanced:
+            self._next[tp] = nxt
+            return nxt
+        return None
 
 
 class KafkaBus:
@@ -65,13 +100,18 @@ class KafkaBus:
                  enable_idempotence: bool = False,
                  max_message_bytes: int = 1024 * 1024,
                  provision_topics: bool = True,
-                 partitions: int = 16):
+                 partitions: int = 16,
+                 lane_queue_max: int = 256):
         self.bootstrap = bootstrap
         self.client_id = client_id
         self.enable_idempotence = enable_idempotence
         self.max_message_bytes = max_message_bytes
         self.provision_topics = provision_topics
         self.partitions = partitions
+        #: bound on queued-but-unhandled deliveries per lane; when a lane
+        #: backs up the consume loop awaits on the full queue, which pauses
+        #: fetching (client-side backpressure instead of unbounded memory)
+        self.lane_queue_max = lane_queue_max
         self._producer: Any = None
         self._consumers: list[Any] = []
         self._tasks: list[asyncio.Task] = []
@@ -103,13 +143,18 @@ class KafkaBus:
             group_id=group or None,
             client_id=self.client_id,
             auto_offset_reset="latest",
-            enable_auto_commit=True)
+            # offsets are committed AFTER the lane handler completes (see
+            # _OffsetLedger); auto-commit would ack queued-but-unhandled
+            # deliveries and drop them on a crash
+            enable_auto_commit=False)
         await consumer.start()
         self._consumers.append(consumer)
         self._tasks.append(asyncio.get_running_loop().create_task(
-            self._consume(consumer, handler)))
+            self._consume(consumer, handler,
+                          _OffsetLedger() if group else None)))
 
-    async def _consume(self, consumer: Any, handler: DeliveryHandler) -> None:
+    async def _consume(self, consumer: Any, handler: DeliveryHandler,
+                       ledger: "_OffsetLedger | None") -> None:
         try:
             async for msg in consumer:
                 raw = RawDelivery(
@@ -119,29 +164,45 @@ class KafkaBus:
                     headers=tuple(msg.headers or ()),
                     offset=msg.offset,
                     partition=msg.partition)
-                self._enqueue(raw, handler)
+                if ledger is not None:
+                    ledger.expect((raw.topic, raw.partition), raw.offset)
+                await self._enqueue(raw, handler, consumer, ledger)
         except asyncio.CancelledError:
             pass
         except Exception:
             _log.exception("kafka consume loop died")
 
-    def _enqueue(self, raw: RawDelivery, handler: DeliveryHandler) -> None:
+    async def _enqueue(self, raw: RawDelivery, handler: DeliveryHandler,
+                       consumer: Any, ledger: "_OffsetLedger | None") -> None:
         lane_key = (raw.topic, raw.key)
         lane = self._lanes.get(lane_key)
         if lane is None:
-            lane = _Lane()
+            lane = _Lane(maxsize=self.lane_queue_max)
             self._lanes[lane_key] = lane
             lane.task = asyncio.get_running_loop().create_task(
                 self._run_lane(lane))
-        lane.queue.put_nowait((raw, handler))
+        # blocks when the lane is full -> the consume loop stops fetching
+        await lane.queue.put((raw, handler, consumer, ledger))
 
     async def _run_lane(self, lane: _Lane) -> None:
         while True:
-            raw, handler = await lane.queue.get()
+            raw, handler, consumer, ledger = await lane.queue.get()
             try:
                 await handler(raw)
             except Exception:
+                # handler faults are the fault rail's business (in-band
+                # ErrorReport); the delivery still counts as handled
                 _log.exception("kafka handler failed on %s", raw.topic)
+            if ledger is not None:
+                watermark = ledger.complete((raw.topic, raw.partition),
+                                            raw.offset)
+                if watermark is not None:
+                    try:
+                        ak = _aiokafka()
+                        tp = ak.TopicPartition(raw.topic, raw.partition)
+                        await consumer.commit({tp: watermark})
+                    except Exception:
+                        _log.exception("offset commit failed on %s", raw.topic)
 
     async def ensure_topics(self, topics: list[str]) -> None:
         """Provision at startup — never hang on no-auto-create brokers."""

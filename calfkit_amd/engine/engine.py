@@ -47,6 +47,15 @@ PREFILL_BUDGET = 8192
 #: enough to LOSE throughput end-to-end (233 -> 210 turns/s at 512)
 RUN_STEPS = (8, 4, 2, 1)
 
+_GOLDEN = 0x9E3779B97F4A7C15
+
+
+def _row_key(seed: int, pos: int) -> int:
+    """Per-(request, token-position) RNG key as a signed int64 (the sample
+    kernel mix64's it with the vocab index)."""
+    k = (seed * _GOLDEN + pos) & 0xFFFFFFFFFFFFFFFF
+    return k - (1 << 64) if k >= (1 << 63) else k
+
 
 @dataclass
 class GenRequest:
@@ -140,7 +149,14 @@ class InferenceEngine:
         self.kv = KVCache(layers=self.cfg.layers, kv_heads=model.hkv,
                           head_dim=model.dh, num_pages=kv_pages,
                           device=self.device)
-        self.max_pages = (self.max_model_len + PAGE_SIZE - 1) // PAGE_SIZE
+        # page-table width carries multi-step overshoot headroom: a sequence
+        # at max_model_len batched with a longer-budget one blind-runs up to
+        # RUN_STEPS[0]-1 extra slots before truncation
+        from ..models.config import MAX_RUN_OVERSHOOT
+
+        assert RUN_STEPS[0] <= MAX_RUN_OVERSHOOT
+        self.max_pages = (self.max_model_len + RUN_STEPS[0]
+                          + PAGE_SIZE - 1) // PAGE_SIZE
 
         # prefill scratch KV (contiguous [rows, hkv, dh] per layer).  The
         # packed varlen prefill with prefix-cache gathering lands the WHOLE
@@ -168,6 +184,9 @@ class InferenceEngine:
         self._temperature = torch.zeros(B, dtype=torch.float32, device=dev)
         self._topk = torch.zeros(B, dtype=torch.int32, device=dev)
         self._seeds = torch.zeros(MS, 2, dtype=torch.int64, device=dev)
+        #: per-(step, row) RNG keys = request seed x token position, so a
+        #: request's sampled tokens do not depend on batch composition
+        self._row_seeds = torch.zeros(MS, B, dtype=torch.int64, device=dev)
         self._sampled = torch.zeros(MS, B, dtype=torch.int32, device=dev)
         self._graphs: dict[tuple[int, int], torch.cuda.CUDAGraph] = {}
         #: one page sacrificed as the write target of padding rows
@@ -581,9 +600,12 @@ class InferenceEngine:
         seed = torch.as_tensor(
             [seqs[0].request.seed or seqs[0].seq_id, self._step_counter],
             dtype=torch.int64).to(logits.device)
+        rks = torch.as_tensor(
+            [_row_key(s.request.seed or s.seq_id, 0) for s in seqs],
+            dtype=torch.int64).to(logits.device)
         ops.sample(out, logits, temp, seed,
                    topk if any(s.request.top_k for s in seqs) else None,
-                   cand=self._prefill_cand)
+                   cand=self._prefill_cand, row_seeds=rks)
         return out
 
     def _sample_host(self, logits: torch.Tensor, seq: _Seq) -> int:
@@ -592,9 +614,12 @@ class InferenceEngine:
         temp = torch.full((1,), r.temperature, device=logits.device)
         seed = torch.as_tensor([r.seed or seq.seq_id, self._step_counter],
                                dtype=torch.int64, device=logits.device)
+        rks = torch.as_tensor([_row_key(r.seed or seq.seq_id, 0)],
+                              dtype=torch.int64, device=logits.device)
         topk = torch.full((1,), r.top_k, dtype=torch.int32, device=logits.device)
         from .. import ops
-        ops.sample(out, logits, temp, seed, topk if r.top_k else None)
+        ops.sample(out, logits, temp, seed, topk if r.top_k else None,
+                   row_seeds=rks)
         return int(out.item())
 
     # ---------------------------------------------------------------- decode
@@ -662,6 +687,15 @@ class InferenceEngine:
         seeds = [[base_seed, self._step_counter + i + 1] for i in range(M)]
         self._seeds[:M].copy_(torch.as_tensor(seeds, dtype=torch.int64),
                               non_blocking=True)
+        # per-row RNG keys: seed x token position (batch-composition-free)
+        bases = np.array([(s.request.seed or s.seq_id) for s in self._running]
+                         + [0] * (pad - B), dtype=np.uint64)
+        toks = np.array([s.new_tokens for s in self._running]
+                        + [0] * (pad - B), dtype=np.uint64)
+        rk = (bases[None, :] * np.uint64(_GOLDEN)
+              + (toks[None, :] + steps.astype(np.uint64)[:, None]))
+        self._row_seeds[:M, :pad].copy_(
+            torch.from_numpy(rk.view(np.int64)), non_blocking=True)
         self._step_counter += M
 
     def _decode_run_gpu(self, pad: int, M: int) -> None:
@@ -678,7 +712,7 @@ class InferenceEngine:
                 seq_lens=self._seq_lens[i, :pad])
             ops.sample(self._sampled[i, :pad], logits,
                        self._temperature[:pad], self._seeds[i],
-                       self._topk[:pad])
+                       self._topk[:pad], row_seeds=self._row_seeds[i, :pad])
             tokens = self._sampled[i, :pad].long()  # on-device feedback
 
     def _capture_graphs(self) -> None:
@@ -771,7 +805,7 @@ class InferenceEngine:
 
     def _check_done(self, seq: _Seq) -> bool:
         r = seq.request
-        eos_hit = (not r.ignore_eos) and seq.last_token == 2
+        eos_hit = (not r.ignore_eos) and seq.last_token == self.cfg.eos_id
         if eos_hit:
             seq.result.finish_reason = "stop"
             return True

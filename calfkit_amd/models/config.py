@@ -10,7 +10,13 @@ from __future__ import annotations
 from dataclasses import dataclass, replace
 
 __all__ = ["ModelConfig", "LLAMA3_8B", "LLAMA3_70B", "MIXTRAL_8X7B", "TINY",
-           "SMOKE", "get_config"]
+           "SMOKE", "get_config", "MAX_RUN_OVERSHOOT"]
+
+#: Multi-step decode runs may blind-run a finishing sequence up to this many
+#: tokens past its budget (engine RUN_STEPS[0] <= this, asserted there); the
+#: RoPE table and the engine's page-table width carry this much headroom so
+#: an overshooting position can never index out of bounds.
+MAX_RUN_OVERSHOOT = 8
 
 
 @dataclass(frozen=True)
@@ -26,6 +32,10 @@ class ModelConfig:
     rope_theta: float = 500000.0
     norm_eps: float = 1e-5
     max_seq: int = 8192
+    #: end-of-sequence token id the engine stops on (SyntheticTokenizer.EOS
+    #: for the synthetic configs; a real llama-3 checkpoint would override
+    #: with 128001/128009)
+    eos_id: int = 2
     # MoE (Mixtral): 0 experts = dense
     num_experts: int = 0
     experts_per_token: int = 0

@@ -119,7 +119,13 @@ class LlamaShard:
         # lm_head replicated (vocab-parallel sharding is a later optimization;
         # 128256 x 8192 bf16 is ~2 GB of the 288 GB rank)
         self.lm_head = self.embed if cfg.tie_embeddings else w(cfg.vocab_size, d, d)
-        self.rope = ops.build_rope_table(cfg.max_seq, self.dh, cfg.rope_theta,
+        # + MAX_RUN_OVERSHOOT: multi-step decode runs blind-run finishing
+        # sequences a few positions past max_seq before truncation; those
+        # positions must still index a real RoPE row.
+        from .config import MAX_RUN_OVERSHOOT
+
+        self.rope = ops.build_rope_table(cfg.max_seq + MAX_RUN_OVERSHOOT,
+                                         self.dh, cfg.rope_theta,
                                          device=self.device)
         self._splitk_cache: dict[int, int] = {}
 

@@ -253,12 +253,16 @@ _SAMPLE_SCRATCH: dict[tuple, torch.Tensor] = {}
 
 def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
            seed_step: torch.Tensor, topk: torch.Tensor | None = None,
-           cand: torch.Tensor | None = None) -> None:
+           cand: torch.Tensor | None = None,
+           row_seeds: torch.Tensor | None = None) -> None:
     """Gumbel-max sampling; temperature 0 = greedy. out: int32 [B].
     Accepts fp32 or bf16 logits (bf16 avoids materializing [B,V] fp32).
     ``cand`` overrides the shared two-stage scratch — callers running on a
     side stream (overlapped prefill) MUST pass their own so a concurrent
-    decode graph replaying the same batch size cannot race on it."""
+    decode graph replaying the same batch size cannot race on it.
+    ``row_seeds`` [B] int64 keys the RNG per row (request seed x token
+    position, built by the engine) so a request's draws are independent of
+    batch composition; without it the legacy (seed, step, row) key applies."""
     if _native(logits):
         if cand is None:
             key = (logits.device.index, int(out.shape[0]))
@@ -271,7 +275,8 @@ def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
             assert cand.numel() >= out.shape[0] * 64 * 2
         _K.sample(out, logits, temperature,
                   seed_step, topk if topk is not None else torch.Tensor(),
-                  cand)
+                  cand,
+                  row_seeds if row_seeds is not None else torch.Tensor())
         return
     B, V = logits.shape
     lg = logits.float().clone()
@@ -286,10 +291,13 @@ def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
         if t <= 0:
             out[b] = int(lg[b].argmax())
         else:
-            # counter-based like the HIP kernel: (seed, step, row) fully
-            # determine the draw — same request seed => same tokens
-            key = (int(seed_step[0]) * 1_000_003
-                   + int(seed_step[1]) * 7919 + b) & 0x7FFFFFFF
+            # counter-based like the HIP kernel: the per-row key (or the
+            # legacy (seed, step, row) triple) fully determines the draw
+            if row_seeds is not None:
+                key = int(row_seeds[b]) & 0x7FFFFFFF
+            else:
+                key = (int(seed_step[0]) * 1_000_003
+                       + int(seed_step[1]) * 7919 + b) & 0x7FFFFFFF
             gen = torch.Generator().manual_seed(key)
             u = torch.rand(V, generator=gen).clamp_min(1e-12)
             g = -torch.log(-torch.log(u))

@@ -34,7 +34,8 @@ void attn_prefill_varlen(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                          torch::Tensor q_starts, torch::Tensor kv_starts,
                          torch::Tensor past_lens, double scale);
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
-            torch::Tensor seed_step, torch::Tensor topk, torch::Tensor cand);
+            torch::Tensor seed_step, torch::Tensor topk, torch::Tensor cand,
+            torch::Tensor row_seeds);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor work, int64_t splitk);
 void skinny_gemm_silu(torch::Tensor out, torch::Tensor x, torch::Tensor w,

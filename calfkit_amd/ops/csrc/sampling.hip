@@ -47,14 +47,21 @@ __global__ void sample_partial_kernel(
     const T* __restrict__ logits,        // [B, V]
     const float* __restrict__ temperature,
     const uint64_t* __restrict__ seed_step,
+    const uint64_t* __restrict__ row_seeds,  // [B] or null: per-request keys
     const int V, const int ns) {
   __shared__ float best_lds[BLOCK / WAVE];
   __shared__ int bestid_lds[BLOCK / WAVE];
   const int b = blockIdx.x;
   const int sl = blockIdx.y;
   const float temp = temperature[b];
-  const uint64_t key = seed_step[0] ^ (seed_step[1] * 0x9e3779b97f4a7c15ULL) ^
-                       ((uint64_t)b << 32);
+  // per-row key (request seed x token position, built host-side) makes the
+  // draw independent of batch composition — same request seed => same
+  // tokens regardless of what else is running.  The legacy
+  // (seed, step, row) key remains for callers without row seeds.
+  const uint64_t key =
+      row_seeds ? mix64(row_seeds[b])
+                : seed_step[0] ^ (seed_step[1] * 0x9e3779b97f4a7c15ULL) ^
+                      ((uint64_t)b << 32);
   const T* row = logits + (int64_t)b * V;
   const int lo = (int)(((int64_t)V * sl) / ns);
   const int hi = (int)(((int64_t)V * (sl + 1)) / ns);
@@ -177,11 +184,16 @@ __global__ void topk_mask_kernel(T* __restrict__ logits, const int V,
 
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor seed_step, torch::Tensor topk,
-            torch::Tensor cand) {
+            torch::Tensor cand, torch::Tensor row_seeds) {
   TORCH_CHECK(logits.is_contiguous());
   TORCH_CHECK(out.scalar_type() == torch::kInt32);
   TORCH_CHECK(seed_step.scalar_type() == torch::kInt64);
   const int B = logits.size(0), V = logits.size(1);
+  const uint64_t* rs = nullptr;
+  if (row_seeds.defined() && row_seeds.numel() >= B) {
+    TORCH_CHECK(row_seeds.scalar_type() == torch::kInt64);
+    rs = (const uint64_t*)row_seeds.data_ptr();
+  }
   // enough slices to fill the chip several blocks deep (the B=256 run at
   // ns=2 measured 100 us for a 65 MB scan — launch-breadth bound)
   int ns = 1;
@@ -207,7 +219,7 @@ void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
                      dim3(BLOCK), 0, s, (float2*)cand.data_ptr(),           \
                      (const T*)logits.data_ptr(),                           \
                      (const float*)temperature.data_ptr(),                  \
-                     (const uint64_t*)seed_step.data_ptr(), V, ns)
+                     (const uint64_t*)seed_step.data_ptr(), rs, V, ns)
   if (bf16)
     P(__hip_bfloat16);
   else

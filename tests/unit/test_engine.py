@@ -243,3 +243,59 @@ def test_large_batch_rounding_to_new_graph_sizes():
         assert len(r.token_ids) == 3
     finally:
         eng.stop()
+
+
+def test_context_cap_sequence_batched_with_longer_budget():
+    """a request whose prompt+max_new reaches
+    max_model_len, batched with a longer-budget request, blind-runs past the
+    cap — the page table and RoPE table must carry the overshoot headroom
+    instead of raising (or silently reading OOB RoPE rows on GPU)."""
+    eng = make_engine(max_model_len=64)
+    try:
+        futs = [
+            # prompt 8 + max_new 56 == max_model_len 64: at the cap
+            eng.submit(GenRequest(prompt_ids=list(range(1, 9)),
+                                  max_new_tokens=56)),
+            # longer budget drives the run length past the first one's end
+            eng.submit(GenRequest(prompt_ids=[1, 2], max_new_tokens=61)),
+        ]
+        r0 = futs[0].result(timeout=120)
+        r1 = futs[1].result(timeout=120)
+        assert len(r0.token_ids) == 56
+        assert len(r1.token_ids) == 61
+    finally:
+        eng.stop()
+
+
+def test_per_request_seed_independent_of_batch_composition():
+    """same request seed => same tokens at temperature>0,
+    whether the request runs alone or batched with others."""
+    def run(extra: int) -> list[int]:
+        eng = make_engine()
+        try:
+            futs = [eng.submit(GenRequest(prompt_ids=[1, 4, 7],
+                                          max_new_tokens=8, temperature=0.8,
+                                          seed=42))]
+            for i in range(extra):
+                futs.append(eng.submit(GenRequest(
+                    prompt_ids=[9, 9, 9], max_new_tokens=8, temperature=0.9,
+                    seed=1000 + i)))
+            return futs[0].result(timeout=120).token_ids
+        finally:
+            eng.stop()
+
+    alone = run(0)
+    assert alone == run(2)
+
+
+def test_eos_id_comes_from_model_config():
+    from calfkit_amd.models.config import get_config
+
+    cfg = get_config("tiny", eos_id=7)
+    from calfkit_amd.models.llama import LlamaShard
+    eng = InferenceEngine(LlamaShard(cfg, device="cpu"), max_batch=4,
+                          use_graphs=False, max_model_len=128)
+    try:
+        assert eng.cfg.eos_id == 7
+    finally:
+        eng.stop()

@@ -71,6 +71,17 @@ class _FakeProducer:
         CLUSTER.produce(topic, key, value, headers)
 
 
+class _TopicPartition:
+    def __init__(self, topic, partition):
+        self.topic, self.partition = topic, partition
+
+    def __hash__(self):
+        return hash((self.topic, self.partition))
+
+    def __eq__(self, other):
+        return (self.topic, self.partition) == (other.topic, other.partition)
+
+
 class _FakeConsumer:
     def __init__(self, *topics, bootstrap_servers=None, group_id=None,
                  auto_offset_reset="latest", **kw):
@@ -79,6 +90,12 @@ class _FakeConsumer:
         self._queue: asyncio.Queue = asyncio.Queue()
         self._started = False
         self._reset = auto_offset_reset
+        self.auto_commit = kw.get("enable_auto_commit", True)
+        self.committed: dict[_TopicPartition, int] = {}
+
+    async def commit(self, offsets):
+        assert self._group, "commit() without a consumer group"
+        self.committed.update(offsets)
 
     async def start(self):
         self._started = True
@@ -134,6 +151,7 @@ def fake_aiokafka(monkeypatch):
     mod = types.ModuleType("aiokafka")
     mod.AIOKafkaProducer = _FakeProducer
     mod.AIOKafkaConsumer = _FakeConsumer
+    mod.TopicPartition = _TopicPartition
     admin = types.ModuleType("aiokafka.admin")
     admin.AIOKafkaAdminClient = _FakeAdmin
     admin.NewTopic = _NewTopic
@@ -186,6 +204,64 @@ async def test_lane_serialization_per_key():
     gate.set()
     await asyncio.sleep(0.05)
     assert [o for o in order if o[0] == "slow"] == [("slow", 0), ("slow", 1)]
+    await bus.close()
+
+
+async def test_offsets_commit_only_after_handler_completes():
+    """at-least-once: a delivery whose handler is still running must not be
+    committed; completion advances a contiguous low-watermark per partition
+    (auto-commit silently dropped in-flight agent calls)."""
+    from calfkit_amd.adapters.kafka import KafkaBus
+
+    bus = KafkaBus("b:9092")
+    gate = asyncio.Event()
+    handled = []
+
+    async def handler(raw):
+        if raw.key == "slow":
+            await gate.wait()
+        handled.append(raw.offset)
+
+    await bus.subscribe("t", handler, group="g")
+    consumer = bus._consumers[0]
+    assert consumer.auto_commit is False
+    await bus.publish("t", "slow", b"0")   # offset 0, blocked in handler
+    await bus.publish("t", "fast", b"1")   # offset 1, completes immediately
+    await asyncio.sleep(0.05)
+    # offset 1 handled, but 0 still in flight -> nothing committable yet
+    assert handled == [1]
+    assert consumer.committed == {}
+    gate.set()
+    await asyncio.sleep(0.05)
+    # both handled -> watermark jumps past the contiguous prefix
+    from calfkit_amd.adapters.kafka import _aiokafka
+    tp = _aiokafka().TopicPartition("t", 0)
+    assert consumer.committed == {tp: 2}
+    await bus.close()
+
+
+async def test_lane_backpressure_bounds_queue_and_pauses_consumption():
+    from calfkit_amd.adapters.kafka import KafkaBus
+
+    bus = KafkaBus("b:9092", lane_queue_max=2)
+    gate = asyncio.Event()
+    handled = []
+
+    async def handler(raw):
+        await gate.wait()
+        handled.append(raw.offset)
+
+    await bus.subscribe("t", handler, group="g")
+    for i in range(8):
+        await bus.publish("t", "k", b"x")
+    await asyncio.sleep(0.05)
+    # one in the handler + at most lane_queue_max queued; the rest stay in
+    # the consumer (fetch paused), not in unbounded lane memory
+    lane = bus._lanes[("t", "k")]
+    assert lane.queue.qsize() <= 2
+    gate.set()
+    await asyncio.sleep(0.1)
+    assert handled == list(range(8))  # order preserved end-to-end
     await bus.close()
 
 
