@@ -103,14 +103,18 @@ class LlamaShard:
                 norm_attn=torch.ones(d, device=self.device, dtype=dtype),
                 norm_mlp=torch.ones(d, device=self.device, dtype=dtype),
             ))
-        # fp32 workspaces for the skinny decode GEMMs' split-K slabs; the
-        # second one carries down-proj slabs across the layer boundary while
-        # the next layer's qkv GEMM already writes the first
-        work_elems = 8 * 64 * max(qkv_out, 2 * self.ffn_s, d, cfg.vocab_size)
+        # fp32 workspaces for the skinny/tiled decode GEMMs' split-K slabs;
+        # the second one carries down-proj slabs across the layer boundary
+        # while the next layer's qkv GEMM already writes the first.  Sized
+        # for the worst routed case: skinny at M<=64 x sk8 over the vocab,
+        # tiled at M<=512 x sk8 over the widest projection.
+        work_elems = max(8 * 64 * max(qkv_out, 2 * self.ffn_s, d,
+                                      cfg.vocab_size),
+                         8 * 512 * max(qkv_out, 2 * self.ffn_s, d))
         if self.device.type == "cuda":
             self._gemm_work = torch.empty(work_elems, dtype=torch.float32,
                                           device=self.device)
-            self._gemm_work2 = torch.empty(8 * 256 * d, dtype=torch.float32,
+            self._gemm_work2 = torch.empty(8 * 512 * d, dtype=torch.float32,
                                            device=self.device)
         else:
             self._gemm_work = torch.empty(0)
@@ -147,6 +151,18 @@ class LlamaShard:
         return (self.device.type == "cuda" and T <= 64 and n_out % 64 == 0
                 and k_in % 128 == 0 and n_out * k_in <= limit)
 
+    def _tile_plan(self, T: int, n_out: int, k_in: int):
+        """(BN, splitk) when the serving-batch tiled GEMM should take this
+        projection (M=65..512 ran hipBLASLt fill-bound at
+        0.65-0.84 PF/s; the 256-wide MFMA tile + split-K beats it — see
+        profiles/r02_tile_gemm.md), else None."""
+        import os
+
+        if (self.device.type != "cuda" or not (64 < T <= 512)
+                or os.environ.get("CALF_TILED", "1") != "1"):
+            return None
+        return ops.tile_plan(T, n_out, k_in)
+
     def _splitk_for(self, weight: torch.Tensor, m: int = 32) -> int:
         key = (weight.shape[0], m > 64)
         sk = self._splitk_cache.get(key)
@@ -163,6 +179,13 @@ class LlamaShard:
             out = torch.empty(x.shape[0], n_out, dtype=x.dtype, device=x.device)
             ops.skinny_gemm(out, x, weight, self._gemm_work, sk)
             return out
+        if decode:
+            plan = self._tile_plan(x.shape[0], n_out, weight.shape[1])
+            if plan is not None:
+                out = torch.empty(x.shape[0], n_out, dtype=x.dtype,
+                                  device=x.device)
+                ops.tile_gemm(out, x, weight, self._gemm_work, *plan)
+                return out
         return torch.nn.functional.linear(x, weight)
 
     # ------------------------------------------------------------------ util
@@ -185,11 +208,22 @@ class LlamaShard:
         T = hidden.shape[0]
         decode = plan is None
         qkv_w = lw.wqkv
+        tplan = (self._tile_plan(T, qkv_w.shape[0], qkv_w.shape[1])
+                 if decode else None)
         if decode and self._use_skinny(T, qkv_w.shape[0], qkv_w.shape[1]):
             # slab-fused path: split-K slabs -> {reduce + RoPE + cache write}
             # in the consumer kernel (no reduce_slabs pass, no re-read)
             sk = self._splitk_for(qkv_w, T)
             Mp = ops.skinny_gemm_slabs(hidden, qkv_w, self._gemm_work, sk)
+            qkv = torch.empty(T, qkv_w.shape[0], dtype=hidden.dtype,
+                              device=hidden.device)
+            ops.rope_qkv_write_slab(qkv, self._gemm_work, kcache, vcache,
+                                    self.rope, positions, slots, self.hq,
+                                    self.hkv, Mp, sk)
+        elif tplan is not None and tplan[1] > 1:
+            # tiled serving-batch path, same slab contract as skinny
+            bn, sk = tplan
+            Mp = ops.tile_gemm_slabs(hidden, qkv_w, self._gemm_work, bn, sk)
             qkv = torch.empty(T, qkv_w.shape[0], dtype=hidden.dtype,
                               device=hidden.device)
             ops.rope_qkv_write_slab(qkv, self._gemm_work, kcache, vcache,
@@ -238,6 +272,12 @@ class LlamaShard:
             sk = self._splitk_for(lw.wo, T)
             Mp = ops.skinny_gemm_slabs(o2, lw.wo, self._gemm_work, sk)
             return ("slabs", Mp, sk)
+        if decode and self.tp_size == 1:
+            oplan = self._tile_plan(T, lw.wo.shape[0], lw.wo.shape[1])
+            if oplan is not None and oplan[1] > 1:
+                bn, sk = oplan
+                Mp = ops.tile_gemm_slabs(o2, lw.wo, self._gemm_work, bn, sk)
+                return ("slabs", Mp, sk)
         out = self._proj(o2, lw.wo, decode=decode)
         return all_reduce_sum(out) if self.tp_size > 1 else out
 
@@ -250,7 +290,9 @@ class LlamaShard:
             sk = ops.choose_splitk(2 * self.ffn_s, self.cfg.hidden, T)
             ops.skinny_gemm_silu(act, hidden, lw.wgateup, self._gemm_work, sk)
         else:
-            gateup = torch.nn.functional.linear(hidden, lw.wgateup)
+            # serving batches route gate-up through the tiled GEMM (via
+            # _proj); prefill stays on the library
+            gateup = self._proj(hidden, lw.wgateup, decode=decode)
             ops.silu_mul(act, gateup)
         if (decode and self.tp_size == 1
                 and self._use_skinny(T, lw.wdown.shape[0], lw.wdown.shape[1])):
@@ -258,6 +300,13 @@ class LlamaShard:
             sk = self._splitk_for(lw.wdown, T)
             Mp = ops.skinny_gemm_slabs(act, lw.wdown, self._gemm_work2, sk)
             return ("slabs2", Mp, sk)
+        if decode and self.tp_size == 1:
+            dplan = self._tile_plan(T, lw.wdown.shape[0], lw.wdown.shape[1])
+            if dplan is not None and dplan[1] > 1:
+                bn, sk = dplan
+                Mp = ops.tile_gemm_slabs(act, lw.wdown, self._gemm_work2,
+                                         bn, sk)
+                return ("slabs2", Mp, sk)
         down = self._proj(act, lw.wdown, decode=decode)
         return all_reduce_sum(down) if self.tp_size > 1 else down
 
@@ -344,4 +393,6 @@ class LlamaShard:
                                   self.cfg.norm_eps)
         if logits_rows is not None:
             hidden = hidden[logits_rows]
-        return torch.nn.functional.linear(hidden, self.lm_head)
+        # decode lm_head routes through the tiled GEMM at serving batches
+        # (501x2 natural tiles at llama vocab: well-filled, no split-K)
+        return self._proj(hidden, self.lm_head, decode=plan is None)

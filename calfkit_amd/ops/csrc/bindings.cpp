@@ -50,6 +50,10 @@ std::vector<std::tuple<int64_t, double>> hipblaslt_algo_bench(
 void hipblaslt_gemm_indexed(torch::Tensor out, torch::Tensor x,
                             torch::Tensor w, int64_t algo_index,
                             torch::Tensor workspace);
+void tile_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+               torch::Tensor work, int64_t bn, int64_t splitk);
+void tile_gemm_slabs(torch::Tensor x, torch::Tensor w, torch::Tensor work,
+                     int64_t bn, int64_t splitk);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16)");
@@ -83,4 +87,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "time every supported hipBLASLt algo for out = x @ w^T");
   m.def("hipblaslt_gemm_indexed", &hipblaslt_gemm_indexed,
         "run out = x @ w^T with a searched hipBLASLt algo index");
+  m.def("tile_gemm", &tile_gemm,
+        "tiled MFMA GEMM for serving batches (M 65..512+), split-K reduce");
+  m.def("tile_gemm_slabs", &tile_gemm_slabs,
+        "tiled MFMA GEMM, fp32 slabs only (consumer fuses the reduce)");
 }

@@ -24,6 +24,7 @@ sources = [
     os.path.join(CSRC, "attn_prefill.hip"),
     os.path.join(CSRC, "sampling.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
+    os.path.join(CSRC, "tile_gemm.hip"),
     os.path.join(CSRC, "moe_gemm.hip"),
     os.path.join(CSRC, "gemm_search.hip"),
 ]

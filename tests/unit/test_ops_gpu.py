@@ -479,3 +479,56 @@ def test_hipblaslt_algo_search_and_indexed_replay():
     torch.cuda.synchronize()
     ref = torch.nn.functional.linear(x, w)
     assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+
+# ------------------------------------------------------------- tiled GEMM
+
+@pytest.mark.parametrize("M,K,N,bn,sk", [
+    (512, 512, 512, 256, 1),    # direct, exact tiles
+    (512, 1024, 512, 256, 4),   # split-K internal reduce
+    (384, 512, 512, 256, 2),    # M tail: second tile row-clamped
+    (512, 512, 384, 128, 2),    # BN=128 wave grid (N%256 != 0)
+    (100, 768, 256, 128, 2),    # single ragged M tile
+])
+def test_tile_gemm_vs_fp32(M, K, N, bn, sk):
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / math.sqrt(K)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    Mp = ((M + 255) // 256) * 256
+    work = torch.empty(sk * Mp * N, dtype=torch.float32, device="cuda")
+    ops.tile_gemm(out, x, w, work, bn, sk)
+    ref = x.float() @ w.float().T
+    assert torch.allclose(out.float(), ref, atol=8e-2, rtol=8e-2), \
+        (out.float() - ref).abs().max()
+
+
+def test_tile_gemm_slabs_match_reduced():
+    """Slab layout contract: work[sk][Mp][N] fp32, summed == product (the
+    fused consumers rope_qkv_write_slab/fused_add_rmsnorm_slab read it)."""
+    M, K, N, bn, sk = 512, 1024, 512, 256, 4
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / math.sqrt(K)
+    Mp = 512
+    work = torch.zeros(sk * Mp * N, dtype=torch.float32, device="cuda")
+    got_mp = ops.tile_gemm_slabs(x, w, work, bn, sk)
+    assert got_mp == Mp
+    summed = work.view(sk, Mp, N).sum(0)[:M]
+    ref = x.float() @ w.float().T
+    assert torch.allclose(summed, ref, atol=8e-2, rtol=8e-2)
+
+
+def test_tile_gemm_matches_linear_serving_shape():
+    """One real llama-8b decode shape (qkv at batch 512) against
+    F.linear/hipBLASLt output in fp32 tolerance."""
+    M, K, N = 512, 4096, 6144
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+         / math.sqrt(K))
+    plan = ops.tile_plan(M, N, K)
+    assert plan is not None
+    bn, sk = plan
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    work = torch.empty(sk * 512 * N, dtype=torch.float32, device="cuda")
+    ops.tile_gemm(out, x, w, work, bn, sk)
+    ref = x.float() @ w.float().T
+    assert torch.allclose(out.float(), ref, atol=8e-2, rtol=8e-2)
