@@ -65,6 +65,13 @@ class GenRequest:
     top_k: int = 0
     seed: int = 0
     ignore_eos: bool = True
+    #: tool-call grammar plan: when tool_names is set the
+    #: decode path runs grammar-constrained sampling — after tool_trigger
+    #: generated tokens the model is steered into the TCALL <name> <args>
+    #: TEND span, with the NAME genuinely sampled among these token ids.
+    tool_names: "list[int] | None" = None
+    tool_trigger: int = 0        # fire after this many generated tokens
+    tool_ncalls: int = 1
     on_token: Callable[[int], None] | None = None
     #: batched delivery — called once per decode RUN with all tokens the
     #: run produced (and once with the prefill's first token).  When set,
@@ -91,6 +98,10 @@ class _Seq:
     result: GenResult = field(default_factory=GenResult)
     last_token: int = 0
     new_tokens: int = 0
+    #: host mirror of the device grammar state (transitions replayed over
+    #: the sampled tokens after each run; gconsumed = tokens replayed)
+    gstate: int = 0
+    gconsumed: int = 0
 
 
 @dataclass
@@ -187,6 +198,14 @@ class InferenceEngine:
         #: per-(step, row) RNG keys = request seed x token position, so a
         #: request's sampled tokens do not depend on batch composition
         self._row_seeds = torch.zeros(MS, B, dtype=torch.int64, device=dev)
+        # tool-call grammar state (static graph I/O; trigger<0 = row off)
+        from .. import ops as _gops
+
+        self._gstate = torch.zeros(B, dtype=torch.int32, device=dev)
+        self._gtrigger = torch.full((B,), -1, dtype=torch.int32, device=dev)
+        self._gnames = torch.full((B, _gops.G_MAX_TOOLS), -1,
+                                  dtype=torch.int32, device=dev)
+        self._gncalls = torch.zeros(B, dtype=torch.int32, device=dev)
         self._sampled = torch.zeros(MS, B, dtype=torch.int32, device=dev)
         self._graphs: dict[tuple[int, int], torch.cuda.CUDAGraph] = {}
         #: one page sacrificed as the write target of padding rows
@@ -299,9 +318,12 @@ class InferenceEngine:
                     p, self._pending = self._pending, None
                     self._prefill_finalize(p)
                 seqs = []
-                for sid, ids, mx, temp, topk, seed in op[1]:
+                for (sid, ids, mx, temp, topk, seed, tnames, ttrig,
+                     tnc) in op[1]:
                     req = GenRequest(prompt_ids=list(ids), max_new_tokens=mx,
-                                     temperature=temp, top_k=topk, seed=seed)
+                                     temperature=temp, top_k=topk, seed=seed,
+                                     tool_names=tnames, tool_trigger=ttrig,
+                                     tool_ncalls=tnc)
                     seqs.append(_Seq(seq_id=sid, request=req, future=Future()))
                 # mirror rank 0's deterministic overlap rule (lockstep state
                 # is identical, so every rank takes the same branch)
@@ -352,7 +374,9 @@ class InferenceEngine:
                         self._tp_bcast(("prefill", [
                             (b.seq_id, b.request.prompt_ids,
                              b.request.max_new_tokens, b.request.temperature,
-                             b.request.top_k, b.request.seed) for b in batch]))
+                             b.request.top_k, b.request.seed,
+                             b.request.tool_names, b.request.tool_trigger,
+                             b.request.tool_ncalls) for b in batch]))
                         if self._overlap_ok(batch):
                             # side-stream launch; the decode run below hides
                             # it — finalize (readback + admit) comes after
@@ -366,7 +390,9 @@ class InferenceEngine:
                             (solo.seq_id, solo.request.prompt_ids,
                              solo.request.max_new_tokens,
                              solo.request.temperature, solo.request.top_k,
-                             solo.request.seed)]))
+                             solo.request.seed, solo.request.tool_names,
+                             solo.request.tool_trigger,
+                             solo.request.tool_ncalls)]))
                         self._prefill_batch([solo])
                 if self._running:
                     self._tp_bcast(("decode",))
@@ -603,10 +629,37 @@ class InferenceEngine:
         rks = torch.as_tensor(
             [_row_key(s.request.seed or s.seq_id, 0) for s in seqs],
             dtype=torch.int64).to(logits.device)
+        if any(s.request.tool_names for s in seqs):
+            self._grammar_prefill(logits, seqs)
         ops.sample(out, logits, temp, seed,
                    topk if any(s.request.top_k for s in seqs) else None,
                    cand=self._prefill_cand, row_seeds=rks)
         return out
+
+    def _grammar_prefill(self, logits: torch.Tensor,
+                         seqs: list["_Seq"]) -> None:
+        """Grammar bias for the first generated token (fresh state; the
+        'previous' token is the last prompt token, never a special)."""
+        from .. import ops
+
+        dev = logits.device
+        B = len(seqs)
+        trig = torch.full((B,), -1, dtype=torch.int32)
+        names = torch.full((B, ops.G_MAX_TOOLS), -1, dtype=torch.int32)
+        ncalls = torch.zeros(B, dtype=torch.int32)
+        pos = torch.zeros(B, dtype=torch.int32)
+        prev = torch.zeros(B, dtype=torch.int64)
+        for b, s in enumerate(seqs):
+            prev[b] = s.request.prompt_ids[-1]
+            pos[b] = len(s.request.prompt_ids) - 1
+            if s.request.tool_names:
+                trig[b] = pos[b] + s.request.tool_trigger
+                ncalls[b] = s.request.tool_ncalls
+                ids = s.request.tool_names[:ops.G_MAX_TOOLS]
+                names[b, :len(ids)] = torch.as_tensor(ids, dtype=torch.int32)
+        gstate = torch.zeros(B, dtype=torch.int32, device=dev)
+        ops.grammar_bias(logits, prev.to(dev), gstate, trig.to(dev),
+                         names.to(dev), pos.to(dev), ncalls.to(dev))
 
     def _sample_host(self, logits: torch.Tensor, seq: _Seq) -> int:
         r = seq.request
@@ -618,6 +671,8 @@ class InferenceEngine:
                               dtype=torch.int64, device=logits.device)
         topk = torch.full((1,), r.top_k, dtype=torch.int32, device=logits.device)
         from .. import ops
+        if r.tool_names:
+            self._grammar_prefill(logits, [seq])
         ops.sample(out, logits, temp, seed, topk if r.top_k else None,
                    row_seeds=rks)
         return int(out.item())
@@ -687,6 +742,34 @@ class InferenceEngine:
         seeds = [[base_seed, self._step_counter + i + 1] for i in range(M)]
         self._seeds[:M].copy_(torch.as_tensor(seeds, dtype=torch.int64),
                               non_blocking=True)
+        # tool-call grammar rows: replay the mirror over tokens the device
+        # has consumed (all but the newest), then load the static buffers
+        from .. import ops as gops
+
+        gtrig = np.full(pad, -1, dtype=np.int32)
+        gnames = np.full((pad, gops.G_MAX_TOOLS), -1, dtype=np.int32)
+        gncalls = np.zeros(pad, dtype=np.int32)
+        gstates = np.zeros(pad, dtype=np.int32)
+        any_tools = False
+        for b, s in enumerate(self._running):
+            if not s.request.tool_names:
+                continue
+            any_tools = True
+            toks = s.result.token_ids
+            while s.gconsumed < len(toks) - 1:
+                s.gstate = gops.grammar_transition(s.gstate, toks[s.gconsumed])
+                s.gconsumed += 1
+            gstates[b] = s.gstate
+            gtrig[b] = len(s.request.prompt_ids) - 1 + s.request.tool_trigger
+            gncalls[b] = s.request.tool_ncalls
+            ids = s.request.tool_names[:gops.G_MAX_TOOLS]
+            gnames[b, :len(ids)] = ids
+        self._any_tools = any_tools
+        self._gstate[:pad].copy_(torch.from_numpy(gstates), non_blocking=True)
+        self._gtrigger[:pad].copy_(torch.from_numpy(gtrig), non_blocking=True)
+        self._gnames[:pad].copy_(torch.from_numpy(gnames), non_blocking=True)
+        self._gncalls[:pad].copy_(torch.from_numpy(gncalls),
+                                  non_blocking=True)
         # per-row RNG keys: seed x token position (batch-composition-free)
         bases = np.array([(s.request.seed or s.seq_id) for s in self._running]
                          + [0] * (pad - B), dtype=np.uint64)
@@ -710,6 +793,11 @@ class InferenceEngine:
                 self.kv.vcache, self._slots[i, :pad],
                 page_table=self._page_table[:pad],
                 seq_lens=self._seq_lens[i, :pad])
+            # tool-call grammar: transition per-row state on the previous
+            # token and bias the logits (no-op for rows without a plan)
+            ops.grammar_bias(logits, tokens, self._gstate[:pad],
+                             self._gtrigger[:pad], self._gnames[:pad],
+                             self._positions[i, :pad], self._gncalls[:pad])
             ops.sample(self._sampled[i, :pad], logits,
                        self._temperature[:pad], self._seeds[i],
                        self._topk[:pad], row_seeds=self._row_seeds[i, :pad])

@@ -134,8 +134,17 @@ class LocalModelClient:
 
     # ------------------------------------------------------------ rendering
     @staticmethod
-    def render_prompt(messages: tuple[Message, ...]) -> str:
+    def render_prompt(messages: tuple[Message, ...],
+                      tools: tuple = ()) -> str:
         lines = []
+        if tools:
+            # the model-facing tool advertisement (reference semantics:
+            # providers render request.tools into the prompt —
+            # /root/reference/CHANGELOG.md:638)
+            decls = " ; ".join(
+                f"{t.name} ( {' '.join(sorted((t.parameters or {}).get('properties', {})))} ) {t.description}"
+                for t in tools)
+            lines.append(f"<|tools|> {decls}")
         for m in messages:
             text = " ".join(p.text for p in m.parts if isinstance(p, TextPart))
             lines.append(f"<|{m.role}|> {text}")
@@ -146,7 +155,7 @@ class LocalModelClient:
     async def complete(self, request: EngineRequest, sink: ChunkSink) -> AssistantMessage:
         engine = self.engine
         tok = self.tokenizer
-        prompt = self.render_prompt(request.messages)
+        prompt = self.render_prompt(request.messages, request.tools)
         ids = tok.encode(prompt)
         s = request.settings
         seq = 0
@@ -159,6 +168,27 @@ class LocalModelClient:
             sink.emit(StreamChunk(event="text_delta", seq=seq,
                                   text=tok.decode(ts) + " "))
 
+        # Tool-call plan: with tools advertised and no tool
+        # results in this turn yet, steer decode through the call grammar —
+        # the NAME is sampled on-device among the advertised tools.  A turn
+        # that already carries tool results produces the closing answer.
+        tool_names = None
+        trigger = 0
+        ncalls = 1
+        if request.tools and not any(m.role == "tool"
+                                     for m in request.messages):
+            import os
+
+            tool_names = [tok.word_id(t.name) for t in request.tools][:8]
+            seedv = (s.seed or 0) or (len(ids) * 2654435761 & 0xFFFF)
+            fan = int(os.environ.get("CALF_TOOLCALL_FANOUT", "1"))
+            if not s.parallel_tool_calls:
+                fan = 1
+            ncalls = max(1, min(fan, len(request.tools)))
+            # fire early enough that all calls fit the token budget
+            span = 7 * ncalls + 2  # TCALL+name+4 args+TEND per call, + EOS
+            trigger = min(seedv % 5 + 1, max(0, s.max_tokens - span))
+
         gen = GenRequest(
             prompt_ids=ids,
             max_new_tokens=max(1, s.max_tokens),
@@ -166,17 +196,25 @@ class LocalModelClient:
             top_k=s.top_k,
             seed=s.seed or 0,
             ignore_eos=False,
+            tool_names=tool_names,
+            tool_trigger=trigger,
+            tool_ncalls=ncalls,
             on_tokens=on_tokens,
         )
         fut = engine.submit(gen)
         result = await asyncio.wrap_future(fut)
-        text = tok.decode(result.token_ids)
-        finish = result.finish_reason
-        for stop in s.stop:  # stop-sequence truncation (client side)
-            idx = text.find(stop)
-            if idx >= 0:
-                text = text[:idx]
-                finish = "stop"
+        parts, calls = _parse_output(result.token_ids, request.tools, tok)
+        finish = "tool_calls" if calls else result.finish_reason
+        if not calls and s.stop:
+            # stop-sequence truncation (client side), text-only turns
+            text = " ".join(p.text for p in parts
+                            if isinstance(p, TextPart))
+            for stop in s.stop:
+                idx = text.find(stop)
+                if idx >= 0:
+                    text = text[:idx]
+                    finish = "stop"
+            parts = (TextPart(text),)
         facts = ProviderFacts(
             model_name=engine.cfg.name,
             usage=Usage(prompt_tokens=result.prompt_tokens,
@@ -185,4 +223,78 @@ class LocalModelClient:
             prefill_ms=result.prefill_ms,
             decode_ms=result.decode_ms,
         )
-        return AssistantMessage((TextPart(text),), facts=facts)
+        return AssistantMessage(parts, facts=facts)
+
+
+def _materialize_args(schema: Any, toks: list[int]) -> dict:
+    """Deterministically fill the tool's JSON schema from sampled arg
+    tokens — the GPU picks the values, the schema dictates the types (so
+    caller-side validation passes and the ToolProcessor can invoke)."""
+    props = (schema.parameters or {}).get("properties", {})
+    args: dict[str, Any] = {}
+    for i, key in enumerate(sorted(props)):
+        tokv = toks[i % len(toks)] if toks else 17
+        ty = (props[key] or {}).get("type", "string")
+        if ty == "integer":
+            args[key] = int(tokv)
+        elif ty == "number":
+            args[key] = round(tokv / 7.0, 3)
+        elif ty == "boolean":
+            args[key] = bool(tokv % 2)
+        elif ty == "array":
+            args[key] = [f"t{tokv}"]
+        elif ty == "object":
+            args[key] = {}
+        else:
+            args[key] = f"t{tokv}"
+    return args
+
+
+def _parse_output(ids: list[int], tools: tuple, tok: SyntheticTokenizer):
+    """Token stream -> (parts, calls): TCALL <name> <args...> TEND spans
+    become ToolCallParts; everything else decodes to text."""
+    import uuid
+
+    from ..messages import ToolCallPart
+
+    by_id: dict[int, Any] = {}
+    for t in tools:
+        by_id.setdefault(tok.word_id(t.name), t)
+    parts: list[Any] = []
+    text_run: list[int] = []
+    calls: list[ToolCallPart] = []
+    i = 0
+    while i < len(ids):
+        t = ids[i]
+        if t == tok.TCALL and i + 1 < len(ids):
+            name_id = ids[i + 1]
+            i += 2
+            arg_toks: list[int] = []
+            while i < len(ids) and ids[i] != tok.TEND:
+                arg_toks.append(ids[i])
+                i += 1
+            if i < len(ids):
+                i += 1  # consume TEND
+            schema = by_id.get(name_id)
+            if schema is None:
+                continue  # name outside the advertised set: drop the span
+            if text_run:
+                txt = tok.decode(text_run)
+                if txt:
+                    parts.append(TextPart(txt))
+                text_run = []
+            call = ToolCallPart(call_id=uuid.uuid4().hex,
+                                tool_name=schema.name,
+                                arguments=_materialize_args(schema, arg_toks))
+            parts.append(call)
+            calls.append(call)
+        else:
+            text_run.append(t)
+            i += 1
+    if text_run:
+        txt = tok.decode(text_run)
+        if txt:
+            parts.append(TextPart(txt))
+    if not parts:
+        parts = [TextPart("")]
+    return tuple(parts), calls

@@ -28,10 +28,19 @@ def _word_id(word: str, span: int) -> int:
 class SyntheticTokenizer:
     BOS = 1
     EOS = 2
+    #: tool-call grammar specials (ops.G_TCALL / G_TEND — the engine's
+    #: constrained sampler steers generation through
+    #: ``TCALL <name> <args> TEND``; see engine/local.py)
+    TCALL = 5
+    TEND = 6
     _RESERVED = 16  # ids 0..15 reserved for specials
 
     def __init__(self, vocab_size: int):
         self.vocab_size = vocab_size
+
+    def word_id(self, word: str) -> int:
+        """Token id a single word encodes to (tool names are single words)."""
+        return self._RESERVED + _word_id(word, self.vocab_size - self._RESERVED)
 
     def encode(self, text: str, *, bos: bool = True) -> list[int]:
         ids = [self.BOS] if bos else []

@@ -33,6 +33,8 @@ __all__ = [
     "tile_gemm",
     "tile_gemm_slabs",
     "tile_plan",
+    "grammar_bias",
+    "grammar_transition",
     "rope_qkv_write_slab",
     "fused_add_rmsnorm_slab",
     "skinny_gemm_silu",
@@ -305,6 +307,69 @@ def sample(out: torch.Tensor, logits: torch.Tensor, temperature: torch.Tensor,
             u = torch.rand(V, generator=gen).clamp_min(1e-12)
             g = -torch.log(-torch.log(u))
             out[b] = int((lg[b] / t + g).argmax())
+
+
+# ----------------------------------------------------- tool-call grammar
+
+#: grammar token ids (SyntheticTokenizer reserved range) + shape constants;
+#: the HIP kernel (sampling.hip grammar_bias_kernel) hard-codes the same
+G_TCALL, G_TEND, G_EOS = 5, 6, 2
+G_ARG_TOKS = 4
+G_MAX_TOOLS = 8
+
+
+def grammar_transition(state: int, prev_token: int) -> int:
+    """One host-side mirror step of the device grammar state machine."""
+    s, calls = state & 0xFF, state >> 8
+    if s == 1:
+        s = 2
+    elif s >= 2:
+        if prev_token == G_TEND:
+            s, calls = 0, calls + 1
+        else:
+            s += 1
+    elif s == 0 and prev_token == G_TCALL:
+        s = 1
+    return s | (calls << 8)
+
+
+def grammar_bias(logits: torch.Tensor, prev: torch.Tensor,
+                 gstate: torch.Tensor, trigger: torch.Tensor,
+                 names: torch.Tensor, pos: torch.Tensor,
+                 ncalls: torch.Tensor) -> None:
+    """Constrained-sampling bias steering rows with a tool plan through the
+    tool-call grammar.  In-place on logits; transitions
+    gstate on the previous token.  Rows with trigger < 0 are untouched."""
+    if _native(logits):
+        _K.grammar_bias(logits, prev, gstate, trigger, names, pos, ncalls)
+        return
+    BIG = 65536.0
+    B = logits.shape[0]
+    for b in range(B):
+        if int(trigger[b]) < 0:
+            continue
+        st = grammar_transition(int(gstate[b]), int(prev[b]))
+        gstate[b] = st
+        s, calls = st & 0xFF, st >> 8
+        row = logits[b]
+        if s == 0:
+            if calls < int(ncalls[b]) and int(pos[b]) >= int(trigger[b]):
+                row[G_TCALL] += BIG
+            elif 0 < calls and calls >= int(ncalls[b]):
+                row[G_EOS] += BIG
+            else:
+                row[G_TCALL] -= BIG
+                row[G_EOS] -= BIG
+        elif s == 1:
+            for t in names[b].tolist():
+                if t >= 0:
+                    row[t] += BIG
+        elif s - 2 >= G_ARG_TOKS:
+            row[G_TEND] += BIG
+        else:
+            row[G_TCALL] -= BIG
+            row[G_TEND] -= BIG
+            row[G_EOS] -= BIG
 
 
 # -------------------------------------------------------------- skinny GEMM

@@ -52,6 +52,10 @@ void hipblaslt_gemm_indexed(torch::Tensor out, torch::Tensor x,
                             torch::Tensor workspace);
 void tile_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                torch::Tensor work, int64_t bn, int64_t splitk);
+void grammar_bias(torch::Tensor logits, torch::Tensor prev,
+                  torch::Tensor gstate, torch::Tensor trigger,
+                  torch::Tensor names, torch::Tensor pos,
+                  torch::Tensor ncalls);
 void tile_gemm_slabs(torch::Tensor x, torch::Tensor w, torch::Tensor work,
                      int64_t bn, int64_t splitk);
 
@@ -91,4 +95,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "tiled MFMA GEMM for serving batches (M 65..512+), split-K reduce");
   m.def("tile_gemm_slabs", &tile_gemm_slabs,
         "tiled MFMA GEMM, fp32 slabs only (consumer fuses the reduce)");
+  m.def("grammar_bias", &grammar_bias,
+        "tool-call grammar constrained-sampling bias (in-place on logits)");
 }

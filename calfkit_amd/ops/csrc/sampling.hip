@@ -179,8 +179,111 @@ __global__ void topk_mask_kernel(T* __restrict__ logits, const int V,
     if (to_f32(row[i]) < hi) row[i] = (T)(-INFINITY);
 }
 
+// ---- tool-call grammar bias -------------------------------
+//
+// Constrained sampling that steers the (random-init) model through the
+// SyntheticTokenizer's tool-call grammar:
+//     ... free text ... TCALL <name> <arg x ARG_TOKS> TEND ... EOS
+// The NAME is genuinely sampled on-device among the advertised tools'
+// name-token ids; args are free samples the parser materializes against
+// the tool's JSON schema.  State is a per-row i32 (low byte = state,
+// high bits = calls made), transitioned on the PREVIOUS sampled token —
+// graph-replay-safe: all inputs live in static device buffers and the
+// host mirrors transitions after each run.
+//
+// States: 0 = free text; 1 = prev was TCALL (force a name);
+//         2+k = name + k arg tokens emitted (k == ARG_TOKS forces TEND).
+
+#define G_TCALL 5
+#define G_TEND 6
+#define G_EOS 2
+#define G_ARG_TOKS 4
+#define G_MAX_TOOLS 8
+#define G_BIG 65536.f
+
+template <typename T>
+__global__ void grammar_bias_kernel(
+    T* __restrict__ logits,              // [B, V]
+    const int64_t* __restrict__ prev,    // [B] previous sampled token
+    int32_t* __restrict__ gstate,        // [B] state | calls<<8
+    const int32_t* __restrict__ trigger, // [B] abs position; <0 = off
+    const int32_t* __restrict__ names,   // [B, G_MAX_TOOLS]; -1 padded
+    const int32_t* __restrict__ pos,     // [B] current position
+    const int32_t* __restrict__ ncalls,  // [B] calls budget
+    const int B, const int V) {
+  const int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  if (trigger[b] < 0) return;            // no tool plan for this row
+  int st = gstate[b];
+  int s = st & 0xff, calls = st >> 8;
+  const int p = (int)prev[b];
+  // transition on the previous token
+  if (s == 1) {
+    s = 2;                               // prev was the name
+  } else if (s >= 2) {
+    if (p == G_TEND) { s = 0; calls += 1; }
+    else s += 1;
+  } else if (s == 0 && p == G_TCALL) {
+    s = 1;
+  }
+  T* row = logits + (int64_t)b * V;
+  const auto bump = [&](int t, float d) {
+    row[t] = (T)(to_f32(row[t]) + d);
+  };
+  if (s == 0) {
+    if (calls < ncalls[b] && pos[b] >= trigger[b]) {
+      bump(G_TCALL, G_BIG);
+    } else if (calls > 0 && calls >= ncalls[b]) {
+      bump(G_EOS, G_BIG);
+    } else {
+      // before the trigger: no call yet, and no premature end of turn
+      bump(G_TCALL, -G_BIG);
+      bump(G_EOS, -G_BIG);
+    }
+  } else if (s == 1) {
+    for (int i = 0; i < G_MAX_TOOLS; ++i) {
+      const int t = names[(int64_t)b * G_MAX_TOOLS + i];
+      if (t >= 0) bump(t, G_BIG);
+    }
+  } else if (s - 2 >= G_ARG_TOKS) {
+    bump(G_TEND, G_BIG);                 // close the call
+  } else {
+    // free arg tokens, but specials stay out of the argument stream
+    bump(G_TCALL, -G_BIG);
+    bump(G_TEND, -G_BIG);
+    bump(G_EOS, -G_BIG);
+  }
+  gstate[b] = s | (calls << 8);
+}
+
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+
+void grammar_bias(torch::Tensor logits, torch::Tensor prev,
+                  torch::Tensor gstate, torch::Tensor trigger,
+                  torch::Tensor names, torch::Tensor pos,
+                  torch::Tensor ncalls) {
+  const int B = logits.size(0), V = logits.size(1);
+  TORCH_CHECK(logits.is_contiguous());
+  TORCH_CHECK(prev.scalar_type() == torch::kInt64);
+  TORCH_CHECK(gstate.scalar_type() == torch::kInt32);
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  const int blocks = (B + 63) / 64;
+#define G(T)                                                                 \
+  hipLaunchKernelGGL((grammar_bias_kernel<T>), dim3(blocks), dim3(64), 0, s, \
+                     (T*)logits.data_ptr(), (const int64_t*)prev.data_ptr(), \
+                     (int32_t*)gstate.data_ptr(),                            \
+                     (const int32_t*)trigger.data_ptr(),                     \
+                     (const int32_t*)names.data_ptr(),                       \
+                     (const int32_t*)pos.data_ptr(),                         \
+                     (const int32_t*)ncalls.data_ptr(), B, V)
+  if (logits.scalar_type() == torch::kBFloat16)
+    G(__hip_bfloat16);
+  else
+    G(float);
+#undef G
+  HIP_CHECK_LAST();
+}
 
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor seed_step, torch::Tensor topk,

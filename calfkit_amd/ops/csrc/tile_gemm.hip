@@ -6,31 +6,29 @@
 // fused_add_rmsnorm_slab in elementwise.hip).
 //
 // Why this kernel exists: at the headline batch (512) the
-// projections ran hipBLASLt at 0.65-0.84 PF/s — fill/latency-bound, far
-// from both the MFMA roof and the HBM roof (profiles/gemm_algo_table.json:
-// qkv 32.8 us, o 26.5, down 72).  The shapes have only 32-96 natural
-// 256-wide tiles on a 256-CU chip, so the lever is per-block LATENCY via
-// split-K (guide: "(N/64)*SPLITK ~ 1/2-1x the CU count") plus a deep
-// glds pipeline for per-CU throughput.
+// projections ran hipBLASLt at 0.65-0.84 PF/s — fill/latency-bound
+// (profiles/gemm_algo_table.json: qkv 32.8 us, o 26.5, down 72).  The
+// shapes have only 32-96 natural 256-wide tiles on a 256-CU chip, so the
+// levers are per-block LATENCY via split-K (guide: "(N/64)*SPLITK ~
+// 1/2-1x the CU count") and per-CU rate via a counted-vmcnt glds pipeline.
 //
-// Structure (CDNA4 guide §5 "256^2 8-phase template", adapted):
+// Structure (CDNA4 guide §5 "256^2 8-phase template" adapted to 256xBN):
 // - 256xBN tile (BN 128/256), BK=64, 8 waves (512 thr), 2 LDS buffers.
 // - Both operands stage global->LDS by __builtin_amdgcn_global_load_lds
-//   (16 B/lane), LDS image lane-linear, XOR swizzle applied to the SOURCE
-//   address and re-applied on the ds_read_b128 side (guide §5.4 rule 21,
-//   T2: byte ^= (row&7)<<4 -> conflict-free column-slice reads).
-// - 4 compute phases per K-tile (one per m-quadrant, 16 MFMAs each for
-//   BN=256) separated by raw s_barriers with s_setprio(1) around the MFMA
-//   cluster (T5); next tile's A/B stages issue in phases 0/1 so their HBM
-//   latency hides under ~2 phases of MFMA before the boundary drain
-//   (T3/T4 in spirit: loads in flight across the inner barriers; the
-//   boundary __syncthreads() emits the vmcnt(0) that publishes the LDS
-//   writes - the only full drain per K-tile).
-// - XCD-bijective block remap (T1) keeps one n-panel's blocks on one XCD
-//   so the W panel is read from that XCD's L2 by its m/sk siblings.
-//
-// Measured on gfx950 (profiles/r02_tile_gemm.md): see routing notes in
-// models/llama.py::_tile_plan.
+//   (16 B/lane), LDS image lane-linear, XOR swizzle on the SOURCE address
+//   re-applied on the ds_read_b128 side (guide §5.4 rule 21 / T2:
+//   byte ^= (row&7)<<4 -> conflict-free column-slice reads).
+// - A K-tile runs as 8 phases (BN=256: m-quadrant x n-half; BN=128: 4
+//   phases) of 8 MFMAs behind raw s_barriers with s_setprio(1) around the
+//   MFMA cluster (T5).  Wave rows/cols interleave at 16-row granularity
+//   (row = mi*16*WR + wr*16) so each phase consumes exactly ONE 128-row
+//   stage unit per operand — that is what makes counted waits possible.
+// - Next tile's 3-4 stage units issue one-per-phase in first-need order
+//   (A0,B0,B1,A1); waits are counted s_waitcnt vmcnt(N) placed before the
+//   barrier that precedes the consuming phase's ds_reads — the pipeline
+//   never drains (T3+T4; v1 of this kernel drained vmcnt(0) per K-tile
+//   via __syncthreads and measured 42% MFU — profiles/r02_tile_gemm.md).
+// - XCD-bijective block remap (T1) keeps one n-panel's blocks on one XCD.
 
 #include "common.h"
 
@@ -38,6 +36,8 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 
 #define MFMA16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
+
+#define VMWAIT(N) asm volatile("s_waitcnt vmcnt(" #N ")" ::: "memory")
 
 template <int BN>
 __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
@@ -49,16 +49,15 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
     const int splitk, const int direct, const int Mp) {
   constexpr int BM = 256, BK = 64;
   constexpr int WC = (BN == 256) ? 4 : 2;  // waves along N
-  constexpr int MI = BM / 16 / (8 / WC);   // m-frags per wave (8 / 4)
+  constexpr int WR = 8 / WC;               // waves along M
+  constexpr int MI = BM / 16 / WR;         // m-frags per wave (8 / 4)
   constexpr int NI = BN / 16 / WC;         // n-frags per wave (4)
-  constexpr int MQ = MI / 4;               // m-frags per phase
   constexpr int A_ELEMS = BM * BK;
   constexpr int B_ELEMS = BN * BK;
   constexpr int BUF = A_ELEMS + B_ELEMS;
   __shared__ __attribute__((aligned(16))) __hip_bfloat16 smem[2 * BUF];
 
-  // ---- block id -> (n_tile, m_tile, sk) with XCD-bijective remap (T1):
-  // consecutive remapped ids share the n panel, and land on one XCD
+  // ---- block id -> (n_tile, m_tile, sk) with XCD-bijective remap (T1)
   int bid = blockIdx.x;
   {
     const int nwg = gridDim.x, q = nwg / 8, r = nwg % 8;
@@ -76,50 +75,54 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
   const int tid = threadIdx.x;
   const int wid = tid / WAVE, lane = tid % WAVE;
   const int l16 = lane & 15, l4 = lane >> 4;
+  const int wr = wid / WC, wc = wid % WC;
   // staging: thread t covers image row sr (8 rows per wave per call),
   // 16B chunk sc of the 128 B row; source chunk is XOR-swizzled (rule 21)
   const int sr = tid / 8;  // == wid*8 + lane/8
   const int sc = tid % 8;
 
-  const auto stage_a = [&](int buf, int ktile) {
+  // one stage unit = 128 image rows = 2 glds per wave (16 KiB total)
+  const auto stage_a = [&](int buf, int ktile, int half) {
     const int kk = k0 + ktile * BK;
 #pragma unroll
-    for (int c = 0; c < BM / 64; ++c) {
-      const int rloc = c * 64 + sr;
+    for (int c = 0; c < 2; ++c) {
+      const int rloc = half * 128 + c * 64 + sr;
       const int grow = min(m0 + rloc, M - 1);  // tail tile row clamp
       const int chunk = sc ^ (rloc & 7);
       __builtin_amdgcn_global_load_lds(
           reinterpret_cast<const unsigned int*>(
               x + (int64_t)grow * K + kk + chunk * 8),
           reinterpret_cast<unsigned int*>(
-              &smem[buf * BUF + (c * 64 + wid * 8) * BK]),
+              &smem[buf * BUF + (half * 128 + c * 64 + wid * 8) * BK]),
           16, 0, 0);
     }
   };
-  const auto stage_b = [&](int buf, int ktile) {
+  const auto stage_b = [&](int buf, int ktile, int half) {
     const int kk = k0 + ktile * BK;
 #pragma unroll
-    for (int c = 0; c < BN / 64; ++c) {
-      const int rloc = c * 64 + sr;
+    for (int c = 0; c < 2; ++c) {  // one unit = 128 rows = 2 x 64-row calls
+      const int rloc = half * 128 + c * 64 + sr;
       const int chunk = sc ^ (rloc & 7);
       __builtin_amdgcn_global_load_lds(
           reinterpret_cast<const unsigned int*>(
               w + (int64_t)(n0 + rloc) * K + kk + chunk * 8),
           reinterpret_cast<unsigned int*>(
-              &smem[buf * BUF + A_ELEMS + (c * 64 + wid * 8) * BK]),
+              &smem[buf * BUF + A_ELEMS + (half * 128 + c * 64 + wid * 8) * BK]),
           16, 0, 0);
     }
   };
-  // fragment reads: row varies over the 16-lane group at one k-chunk ->
-  // the T2 column-slice pattern; same XOR as the staged source
+  // BN=128's B unit is 2 glds over rows 0..127 via c loop; half==0 always.
+  // fragment reads: 16-lane groups span 16 rows at one k-chunk -> the T2
+  // column-slice pattern; same XOR as the staged source.  Wave rows/cols
+  // interleave so mi<MI/2 lies in stage half 0.
   const auto read_a = [&](int buf, int mi, int ks) {
-    const int row = (wid / WC) * (16 * MI) + mi * 16 + l16;
+    const int row = mi * (16 * WR) + wr * 16 + l16;
     const int chunk = (ks * 4 + l4) ^ (row & 7);
     return *reinterpret_cast<const bf16x8v*>(
         &smem[buf * BUF + row * BK + chunk * 8]);
   };
   const auto read_b = [&](int buf, int ni, int ks) {
-    const int row = (wid % WC) * (16 * NI) + ni * 16 + l16;
+    const int row = ni * (16 * WC) + wc * 16 + l16;
     const int chunk = (ks * 4 + l4) ^ (row & 7);
     return *reinterpret_cast<const bf16x8v*>(
         &smem[buf * BUF + A_ELEMS + row * BK + chunk * 8]);
@@ -131,51 +134,136 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  stage_a(0, 0);
-  stage_b(0, 0);
-  __syncthreads();  // drains the glds (vmcnt 0) and publishes LDS
+  // prologue: tile 0 in first-need order; counted waits below cover it
+  if (BN == 256) {
+    stage_b(0, 0, 0);
+    stage_b(0, 0, 1);
+    stage_a(0, 0, 0);
+    stage_a(0, 0, 1);
+  } else {
+    stage_b(0, 0, 0);
+    stage_a(0, 0, 0);
+    stage_a(0, 0, 1);
+  }
+
+  // pre-loop: tile 0's first-phase units landed (B0,B1,A0; A1 still in
+  // flight), all waves synced before any LDS read
+  VMWAIT(2);
+  __builtin_amdgcn_s_barrier();
 
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
-    bf16x8v bfr[NI][2];
+    const bool pf = t + 1 < nt;
+    const int nb = cur ^ 1;
+    // Phase body order matters (PMC r02: stage-issue at phase head cost
+    // ~300 cyc/phase of MFMA-pipe idle): ds_reads FIRST, then the stage
+    // issue, then barrier -> MFMA cluster.  Counted waits sit at a
+    // phase's TAIL (before its end barrier) covering the NEXT phase's
+    // units, so the barrier orders every wave's wait before any read.
+    if (BN == 256) {
+      // 4 phases of 16 MFMA (one m-quadrant x full BN); stage units of
+      // t+1 issue one-per-phase in first-need order B0,B1,A0,A1.
+      bf16x8v bfr[NI][2];
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      // issue next tile's stages early (phases 0/1): ~2 phases of MFMA
-      // cover the HBM latency before the boundary drain below
-      if (q == 0 && t + 1 < nt) stage_a(cur ^ 1, t + 1);
-      if (q == 1 && t + 1 < nt) stage_b(cur ^ 1, t + 1);
-      if (q == 0) {
-#pragma unroll
-        for (int ni = 0; ni < NI; ++ni)
-#pragma unroll
-          for (int ks = 0; ks < 2; ++ks) bfr[ni][ks] = read_b(cur, ni, ks);
-      }
-      bf16x8v afr[MQ][2];
-#pragma unroll
-      for (int mi = 0; mi < MQ; ++mi)
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
-          afr[mi][ks] = read_a(cur, q * MQ + mi, ks);
-      __builtin_amdgcn_s_barrier();  // raw: do NOT drain the in-flight glds
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-        for (int mi = 0; mi < MQ; ++mi)
+      for (int ph = 0; ph < 4; ++ph) {
+        if (ph == 0) {
 #pragma unroll
           for (int ni = 0; ni < NI; ++ni)
-            acc[q * MQ + mi][ni] =
-                MFMA16(afr[mi][ks], bfr[ni][ks], acc[q * MQ + mi][ni]);
-      __builtin_amdgcn_s_setprio(0);
-      __builtin_amdgcn_s_barrier();
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks) bfr[ni][ks] = read_b(cur, ni, ks);
+        }
+        bf16x8v afr[2][2];
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+        if (pf) {
+          if (ph == 0) stage_b(nb, t + 1, 0);
+          else if (ph == 1) stage_b(nb, t + 1, 1);
+          else if (ph == 2) stage_a(nb, t + 1, 0);
+          else stage_a(nb, t + 1, 1);
+        }
+        __builtin_amdgcn_s_barrier();
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < NI; ++ni)
+              acc[ph * 2 + mi][ni] =
+                  MFMA16(afr[mi][ks], bfr[ni][ks], acc[ph * 2 + mi][ni]);
+        __builtin_amdgcn_s_setprio(0);
+        if (ph == 1) {
+          // ph2 reads A1(t): outstanding {A1, B0', B1'} -> vmcnt(4);
+          // final tile issues nothing -> drain what remains
+          if (pf) {
+            VMWAIT(4);
+          } else {
+            VMWAIT(0);
+          }
+        } else if (ph == 3 && pf) {
+          // next tile's ph0 reads {B0',B1',A0'}: outstanding 8 -> vmcnt(2)
+          VMWAIT(2);
+        }
+        __builtin_amdgcn_s_barrier();
+      }
+    } else {
+      // BN=128: 2 phases of 16 MFMA (2 mi x 4 ni); units B,A0,A1.
+      bf16x8v bfr[NI][2];
+#pragma unroll
+      for (int ph = 0; ph < 2; ++ph) {
+        if (ph == 0) {
+#pragma unroll
+          for (int ni = 0; ni < NI; ++ni)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks) bfr[ni][ks] = read_b(cur, ni, ks);
+        }
+        bf16x8v afr[2][2];
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+        if (pf) {
+          if (ph == 0) {
+            stage_b(nb, t + 1, 0);
+            stage_a(nb, t + 1, 0);
+          } else {
+            stage_a(nb, t + 1, 1);
+          }
+        }
+        __builtin_amdgcn_s_barrier();
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < NI; ++ni)
+              acc[ph * 2 + mi][ni] =
+                  MFMA16(afr[mi][ks], bfr[ni][ks], acc[ph * 2 + mi][ni]);
+        __builtin_amdgcn_s_setprio(0);
+        if (ph == 0) {
+          // ph1 reads A1(t): outstanding {A1, B', A0'} -> vmcnt(4)
+          if (pf) {
+            VMWAIT(4);
+          } else {
+            VMWAIT(0);
+          }
+        } else if (pf) {
+          // next ph0 reads {B',A0'}: outstanding {B',A0',A1'} -> vmcnt(2)
+          VMWAIT(2);
+        }
+        __builtin_amdgcn_s_barrier();
+      }
     }
-    __syncthreads();  // K-tile boundary: vmcnt(0) publishes next tile's LDS
     cur ^= 1;
   }
 
   // ---- epilogue.  C layout (mfma_f32_16x16x32): col = l16, row = l4*4+r.
-  const int r0 = (wid / WC) * (16 * MI);
-  const int c0 = (wid % WC) * (16 * NI);
+  // Wave rows interleave: m-frag mi -> tile row mi*16*WR + wr*16.
   if (direct) {
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi)
@@ -183,9 +271,9 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
       for (int ni = 0; ni < NI; ++ni)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int row = m0 + r0 + mi * 16 + l4 * 4 + r;
+          const int row = m0 + mi * (16 * WR) + wr * 16 + l4 * 4 + r;
           if (row < M)
-            out[(int64_t)row * N + n0 + c0 + ni * 16 + l16] =
+            out[(int64_t)row * N + n0 + ni * (16 * WC) + wc * 16 + l16] =
                 f2bf(acc[mi][ni][r]);
         }
   } else {
@@ -196,8 +284,9 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
       for (int ni = 0; ni < NI; ++ni)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int row = m0 + r0 + mi * 16 + l4 * 4 + r;
-          slab[(int64_t)row * N + n0 + c0 + ni * 16 + l16] = acc[mi][ni][r];
+          const int row = m0 + mi * (16 * WR) + wr * 16 + l4 * 4 + r;
+          slab[(int64_t)row * N + n0 + ni * (16 * WC) + wc * 16 + l16] =
+              acc[mi][ni][r];
         }
   }
 }

@@ -299,3 +299,55 @@ def test_eos_id_comes_from_model_config():
         assert eng.cfg.eos_id == 7
     finally:
         eng.stop()
+
+
+def _run_tool_grammar(device="cpu", use_graphs=False):
+    """Engine-level: a request with a tool plan emits a grammar-valid
+    TCALL <name> <args x 4> TEND span with the name sampled among the
+    advertised ids."""
+    from calfkit_amd import ops
+    from calfkit_amd.models.config import TINY
+
+    model = LlamaShard(TINY, device=device)
+    eng = InferenceEngine(model, max_batch=4, use_graphs=use_graphs,
+                          max_model_len=256)
+    try:
+        names = [100, 200, 300]
+        r = eng.generate(GenRequest(
+            prompt_ids=[1, 20, 30, 40], max_new_tokens=24, temperature=0.7,
+            seed=7, ignore_eos=False, tool_names=names, tool_trigger=2,
+            tool_ncalls=2), timeout=300)
+        toks = r.token_ids
+        calls = []
+        i = 0
+        while i < len(toks):
+            if toks[i] == ops.G_TCALL:
+                name = toks[i + 1]
+                j = i + 2
+                args = []
+                while j < len(toks) and toks[j] != ops.G_TEND:
+                    args.append(toks[j])
+                    j += 1
+                calls.append((name, args))
+                i = j + 1
+            else:
+                i += 1
+        assert len(calls) == 2, toks
+        for name, args in calls:
+            assert name in names
+            assert len(args) == ops.G_ARG_TOKS
+            assert all(a >= 16 for a in args)  # no specials in args
+        # the turn ends on EOS right after the final call
+        assert toks[-1] == 2 and r.finish_reason == "stop"
+    finally:
+        eng.stop()
+
+
+def test_tool_grammar_cpu():
+    _run_tool_grammar()
+
+
+@pytest.mark.gpu
+def test_tool_grammar_gpu_graphs():
+    """The same grammar inside the hipGraph-captured multi-step decode."""
+    _run_tool_grammar(device="cuda", use_graphs=True)
