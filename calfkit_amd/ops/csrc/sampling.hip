@@ -248,10 +248,8 @@ __global__ void grammar_bias_kernel(
   } else if (s - 2 >= G_ARG_TOKS) {
     bump(G_TEND, G_BIG);                 // close the call
   } else {
-    // free arg tokens, but specials stay out of the argument stream
-    bump(G_TCALL, -G_BIG);
-    bump(G_TEND, -G_BIG);
-    bump(G_EOS, -G_BIG);
+    // free arg tokens; the whole reserved/special range stays out
+    for (int t = 0; t < 16; ++t) bump(t, -G_BIG);
   }
   gstate[b] = s | (calls << 8);
 }

@@ -39,7 +39,11 @@ using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 
 #define VMWAIT(N) asm volatile("s_waitcnt vmcnt(" #N ")" ::: "memory")
 
-template <int BN>
+// FLAGS bits: 1 = drop the pre-MFMA barrier; 2 = drop setprio around the
+// MFMA cluster (guide worked-example 2: setprio HURT a lockstep GEMM);
+// 4 = issue the phase's stage unit BETWEEN the two ks MFMA half-clusters
+// (glds issue ~60 cyc among bare MFMAs vs 100-185 in a load phase).
+template <int BN, int FLAGS>
 __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
     float* __restrict__ work,             // [splitk, Mp, N] fp32 slabs
     __hip_bfloat16* __restrict__ out,     // [M, N] (direct == 1)
@@ -178,23 +182,27 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
 #pragma unroll
           for (int ks = 0; ks < 2; ++ks)
             afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
-        if (pf) {
+        const auto stage_ph = [&] {
+          if (!pf) return;
           if (ph == 0) stage_b(nb, t + 1, 0);
           else if (ph == 1) stage_b(nb, t + 1, 1);
           else if (ph == 2) stage_a(nb, t + 1, 0);
           else stage_a(nb, t + 1, 1);
-        }
-        __builtin_amdgcn_s_barrier();
-        __builtin_amdgcn_s_setprio(1);
+        };
+        if constexpr (!(FLAGS & 4)) stage_ph();
+        if constexpr (!(FLAGS & 1)) __builtin_amdgcn_s_barrier();
+        if constexpr (!(FLAGS & 2)) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
+        for (int ks = 0; ks < 2; ++ks) {
+          if (ks == 1 && (FLAGS & 4)) stage_ph();
 #pragma unroll
           for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
             for (int ni = 0; ni < NI; ++ni)
               acc[ph * 2 + mi][ni] =
                   MFMA16(afr[mi][ks], bfr[ni][ks], acc[ph * 2 + mi][ni]);
-        __builtin_amdgcn_s_setprio(0);
+        }
+        if constexpr (!(FLAGS & 2)) __builtin_amdgcn_s_setprio(0);
         if (ph == 1) {
           // ph2 reads A1(t): outstanding {A1, B0', B1'} -> vmcnt(4);
           // final tile issues nothing -> drain what remains
@@ -203,11 +211,13 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
           } else {
             VMWAIT(0);
           }
-        } else if (ph == 3 && pf) {
-          // next tile's ph0 reads {B0',B1',A0'}: outstanding 8 -> vmcnt(2)
-          VMWAIT(2);
+          __builtin_amdgcn_s_barrier();
+        } else if (ph == 3) {
+          if (pf) VMWAIT(2);
+          __builtin_amdgcn_s_barrier();
+        } else if constexpr (!(FLAGS & 8)) {
+          __builtin_amdgcn_s_barrier();
         }
-        __builtin_amdgcn_s_barrier();
       }
     } else {
       // BN=128: 2 phases of 16 MFMA (2 mi x 4 ni); units B,A0,A1.
@@ -226,25 +236,29 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
 #pragma unroll
           for (int ks = 0; ks < 2; ++ks)
             afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
-        if (pf) {
+        const auto stage_ph = [&] {
+          if (!pf) return;
           if (ph == 0) {
             stage_b(nb, t + 1, 0);
             stage_a(nb, t + 1, 0);
           } else {
             stage_a(nb, t + 1, 1);
           }
-        }
-        __builtin_amdgcn_s_barrier();
-        __builtin_amdgcn_s_setprio(1);
+        };
+        if constexpr (!(FLAGS & 4)) stage_ph();
+        if constexpr (!(FLAGS & 1)) __builtin_amdgcn_s_barrier();
+        if constexpr (!(FLAGS & 2)) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
+        for (int ks = 0; ks < 2; ++ks) {
+          if (ks == 1 && (FLAGS & 4)) stage_ph();
 #pragma unroll
           for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
             for (int ni = 0; ni < NI; ++ni)
               acc[ph * 2 + mi][ni] =
                   MFMA16(afr[mi][ks], bfr[ni][ks], acc[ph * 2 + mi][ni]);
-        __builtin_amdgcn_s_setprio(0);
+        }
+        if constexpr (!(FLAGS & 2)) __builtin_amdgcn_s_setprio(0);
         if (ph == 0) {
           // ph1 reads A1(t): outstanding {A1, B', A0'} -> vmcnt(4)
           if (pf) {
@@ -293,6 +307,7 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
 
 // fwd decl (skinny_gemm.hip)
 __global__ void reduce_slabs_kernel(__hip_bfloat16* __restrict__ out,
@@ -312,17 +327,33 @@ static void launch_tile(torch::Tensor work, torch::Tensor out,
   const int blocks = (N / (int)bn) * m_tiles * (int)splitk;
   TORCH_CHECK(direct || work.numel() >= (int64_t)splitk * Mp * N,
               "tile_gemm: workspace too small");
-#define LT(BNV)                                                              \
-  hipLaunchKernelGGL((tile_gemm_kernel<BNV>), dim3(blocks), dim3(512), 0, s, \
-                     (float*)work.data_ptr(),                                \
+  const char* fv = getenv("CALF_TILE_FLAGS");
+  const int flags = fv ? atoi(fv) : 7;
+#define LT(BNV, F)                                                           \
+  hipLaunchKernelGGL((tile_gemm_kernel<BNV, F>), dim3(blocks), dim3(512), 0, \
+                     s, (float*)work.data_ptr(),                             \
                      (__hip_bfloat16*)out.data_ptr(),                        \
                      (const __hip_bfloat16*)x.data_ptr(),                    \
                      (const __hip_bfloat16*)w.data_ptr(), M, N, K, m_tiles,  \
                      (int)splitk, direct, Mp)
-  if (bn == 256)
-    LT(256);
-  else
-    LT(128);
+#define LTF(BNV)                                                             \
+  switch (flags) {                                                           \
+    case 1: LT(BNV, 1); break;                                               \
+    case 2: LT(BNV, 2); break;                                               \
+    case 3: LT(BNV, 3); break;                                               \
+    case 4: LT(BNV, 4); break;                                               \
+    case 5: LT(BNV, 5); break;                                               \
+    case 6: LT(BNV, 6); break;                                               \
+    case 7: LT(BNV, 7); break;                                               \
+    case 15: LT(BNV, 15); break;                                             \
+    default: LT(BNV, 0); break;                                              \
+  }
+  if (bn == 256) {
+    LTF(256);
+  } else {
+    LTF(128);
+  }
+#undef LTF
 #undef LT
 }
 
