@@ -158,10 +158,21 @@ class LlamaShard:
         profiles/r02_tile_gemm.md), else None."""
         import os
 
-        if (self.device.type != "cuda" or not (64 < T <= 512)
-                or os.environ.get("CALF_TILED", "1") != "1"):
+        mode = os.environ.get("CALF_TILED", "1")
+        if self.device.type != "cuda" or not (64 < T <= 512) or mode == "0":
             return None
-        return ops.tile_plan(T, n_out, k_in)
+        plan = ops.tile_plan(T, n_out, k_in)
+        if plan is None or mode == "all":
+            return plan
+        # measured-policy dispatch (profiles/r02_tile_gemm.md): route the
+        # shapes where the hand-written kernel beats hipBLASLt in the
+        # serving path today — deep-K down projections (1.03-1.05x raw,
+        # more with the fused slab consumer) and the vocab-wide lm_head
+        # (1.07-1.11x); the rest stay on the library until the kernel
+        # earns them (CALF_TILED=all forces everything tiled).
+        if k_in >= 3 * n_out or (n_out >= 32000 and n_out % 256 == 0):
+            return plan
+        return None
 
     def _splitk_for(self, weight: torch.Tensor, m: int = 32) -> int:
         key = (weight.shape[0], m > 64)
