@@ -67,12 +67,17 @@ __global__ __launch_bounds__(NW * WAVE) void attn_decode4_kernel(
   const int thalf = lane >> 5;        // 0: tokens 0..31, 1: tokens 32..63
 
   const int64_t tbl = (int64_t)seq * max_pages;
+  // software-pipelined page ids: the table read is a dependent scalar load
+  // (L2/HBM latency) feeding every K/V address — prefetch one iteration
+  // ahead so the next page's id is in flight during this page's stream
+  int page_next = (wid < npages) ? page_table[tbl + wid] : 0;
   for (int pi = wid; pi < npages; pi += NW) {
     // bar the compiler from hoisting the per-page q_lds re-reads out of
     // this loop: without it LICM keeps G*128 floats live across pages
     // (251-747 VGPR spills measured; see the r01/r03 kernel profiles)
     asm volatile("" ::: "memory");
-    const int page = page_table[tbl + pi];
+    const int page = page_next;
+    if (pi + NW < npages) page_next = page_table[tbl + pi + NW];
     const int64_t base = (((int64_t)page * hkv + kvh) * KVPAGE) * HD;
     const int token = pi * KVPAGE + lane;
     const bool live = token < slen;
@@ -135,7 +140,10 @@ __global__ __launch_bounds__(NW * WAVE) void attn_decode4_kernel(
     const float2* vbase = reinterpret_cast<const float2*>(vcache + base) + dlane;
     if (nt > 0) {
       // 2x8-row double buffer, macro-expanded so every array index is a
-      // compile-time constant (pointer selection would demote to scratch)
+      // compile-time constant (pointer selection would demote to scratch).
+      // (An early-issue variant that hoisted the first 16 V rows above the
+      // softmax spilled 226 VGPRs and ran 7x slower — r02; kreg + vA/vB
+      // must not be live at once.)
       float2 vA[8], vB[8];
 #define PV_LOAD(BUF, C)                                                     \
   _Pragma("unroll") for (int j = 0; j < 8; ++j)                             \
