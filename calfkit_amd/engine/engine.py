@@ -47,6 +47,49 @@ PREFILL_BUDGET = 8192
 #: enough to LOSE throughput end-to-end (233 -> 210 turns/s at 512)
 RUN_STEPS = (8, 4, 2, 1)
 
+#: lockstep op channel opcodes (tensor-coded — VERDICT r01 #2: the
+#: pickled broadcast_object_list per visit becomes one fixed int64[2]
+#: header broadcast, plus one int64 payload broadcast only for prefills)
+OP_STOP, OP_DECODE, OP_PREFILL = 0, 1, 2
+
+
+def _encode_prefill(seqs) -> list[int]:
+    vals: list[int] = [len(seqs)]
+    for s in seqs:
+        r = s.request
+        names = r.tool_names or []
+        vals += [s.seq_id, r.max_new_tokens, int(r.temperature * 1e6),
+                 r.top_k, r.seed, int(r.ignore_eos),
+                 r.tool_trigger if r.tool_names else -1, r.tool_ncalls,
+                 len(names)]
+        vals += names
+        vals.append(len(r.prompt_ids))
+        vals += r.prompt_ids
+    return vals
+
+
+def _decode_prefill(vals: list[int]) -> list["GenRequest"]:
+    reqs = []
+    i = 1
+    for _ in range(vals[0]):
+        (sid, mx, temp_i, topk, seed, ig_eos, trig, ncalls,
+         n_names) = vals[i:i + 9]
+        i += 9
+        names = vals[i:i + n_names]
+        i += n_names
+        n_ids = vals[i]
+        i += 1
+        ids = vals[i:i + n_ids]
+        i += n_ids
+        req = GenRequest(prompt_ids=list(ids), max_new_tokens=mx,
+                         temperature=temp_i / 1e6, top_k=topk, seed=seed,
+                         ignore_eos=bool(ig_eos),
+                         tool_names=list(names) if trig >= 0 else None,
+                         tool_trigger=max(trig, 0), tool_ncalls=ncalls)
+        reqs.append((sid, req))
+    return reqs
+
+
 _GOLDEN = 0x9E3779B97F4A7C15
 
 
@@ -146,7 +189,14 @@ class InferenceEngine:
             # extension) is not capturable — the capture try/except below
             # would catch it, but don't even try.
             from .. import ops as _ops
-            use_graphs = self.on_gpu and (
+            # DECISION: TP/EP comm stays GRAPH-EXTERNAL.
+            # RCCL-inside-hipGraph is unexercised on this stack, and the
+            # TP-overlap pipeline (llama._forward_tp_overlap) already
+            # multi-streams the collectives with event ordering — capture
+            # would have to freeze that choreography.  tp/ep>1 therefore
+            # decodes eager through the overlap pipeline; use_graphs=True
+            # can still be forced explicitly for experiments.
+            use_graphs = self.on_gpu and not self.tp_lockstep and (
                 not self.cfg.is_moe
                 or (_ops.have_extension()
                     and getattr(model, "ep_size", 1) == 1))
@@ -252,7 +302,7 @@ class InferenceEngine:
             self._started = False
             if self.tp_lockstep:
                 try:
-                    self._tp_bcast(("stop",))
+                    self._op_bcast(OP_STOP)
                 except Exception:
                     _log.exception("tp stop broadcast failed")
 
@@ -286,19 +336,39 @@ class InferenceEngine:
         return self.submit(request).result(timeout)
 
     # ----------------------------------------------------------- TP lockstep
-    def _tp_bcast(self, op) -> None:
-        if self.tp_lockstep:
-            import torch.distributed as td
+    @property
+    def _comm_dev(self) -> torch.device:
+        from ..parallel import get_dist
 
-            objs = [op]
-            td.broadcast_object_list(objs, src=0)
+        return self.device if get_dist().backend == "nccl" \
+            else torch.device("cpu")
 
-    def _tp_recv(self):
-        import torch.distributed as td
+    def _op_bcast(self, opcode: int, payload: "list[int] | None" = None) -> None:
+        if not self.tp_lockstep:
+            return
+        from ..parallel import broadcast_tensor
 
-        objs = [None]
-        td.broadcast_object_list(objs, src=0)
-        return objs[0]
+        dev = self._comm_dev
+        n = len(payload) if payload else 0
+        broadcast_tensor(torch.tensor([opcode, n], dtype=torch.int64,
+                                      device=dev))
+        if n:
+            broadcast_tensor(torch.tensor(payload, dtype=torch.int64,
+                                          device=dev))
+
+    def _op_recv(self) -> "tuple[int, list[int] | None]":
+        from ..parallel import broadcast_tensor
+
+        dev = self._comm_dev
+        hdr = torch.zeros(2, dtype=torch.int64, device=dev)
+        broadcast_tensor(hdr)
+        opcode, n = int(hdr[0]), int(hdr[1])
+        vals = None
+        if n:
+            buf = torch.empty(n, dtype=torch.int64, device=dev)
+            broadcast_tensor(buf)
+            vals = buf.cpu().tolist()
+        return opcode, vals
 
     def follow(self) -> None:
         """Follower loop for TP ranks > 0 (call instead of start())."""
@@ -310,21 +380,15 @@ class InferenceEngine:
                 _log.exception("follower graph capture failed; eager")
                 self.use_graphs = False
         while True:
-            op = self._tp_recv()
-            if op[0] == "stop":
+            opcode, vals = self._op_recv()
+            if opcode == OP_STOP:
                 return
-            if op[0] == "prefill":
+            if opcode == OP_PREFILL:
                 if self._pending is not None:  # safety: rank 0 never does this
                     p, self._pending = self._pending, None
                     self._prefill_finalize(p)
-                seqs = []
-                for (sid, ids, mx, temp, topk, seed, tnames, ttrig,
-                     tnc) in op[1]:
-                    req = GenRequest(prompt_ids=list(ids), max_new_tokens=mx,
-                                     temperature=temp, top_k=topk, seed=seed,
-                                     tool_names=tnames, tool_trigger=ttrig,
-                                     tool_ncalls=tnc)
-                    seqs.append(_Seq(seq_id=sid, request=req, future=Future()))
+                seqs = [_Seq(seq_id=sid, request=req, future=Future())
+                        for sid, req in _decode_prefill(vals)]
                 # mirror rank 0's deterministic overlap rule (lockstep state
                 # is identical, so every rank takes the same branch)
                 if self._overlap_ok(seqs):
@@ -332,7 +396,7 @@ class InferenceEngine:
                         self._pending = self._prefill_launch(seqs)
                 else:
                     self._prefill_batch(seqs)
-            elif op[0] == "decode":
+            elif opcode == OP_DECODE:
                 self._decode_step()
                 if self._pending is not None:
                     p, self._pending = self._pending, None
@@ -371,12 +435,7 @@ class InferenceEngine:
                         budget -= n
                         batch.append(self._waiting.pop(0))
                     if batch:
-                        self._tp_bcast(("prefill", [
-                            (b.seq_id, b.request.prompt_ids,
-                             b.request.max_new_tokens, b.request.temperature,
-                             b.request.top_k, b.request.seed,
-                             b.request.tool_names, b.request.tool_trigger,
-                             b.request.tool_ncalls) for b in batch]))
+                        self._op_bcast(OP_PREFILL, _encode_prefill(batch))
                         if self._overlap_ok(batch):
                             # side-stream launch; the decode run below hides
                             # it — finalize (readback + admit) comes after
@@ -386,16 +445,10 @@ class InferenceEngine:
                         self._prefill_batch(batch)
                     elif self._waiting:
                         solo = self._waiting.pop(0)
-                        self._tp_bcast(("prefill", [
-                            (solo.seq_id, solo.request.prompt_ids,
-                             solo.request.max_new_tokens,
-                             solo.request.temperature, solo.request.top_k,
-                             solo.request.seed, solo.request.tool_names,
-                             solo.request.tool_trigger,
-                             solo.request.tool_ncalls)]))
+                        self._op_bcast(OP_PREFILL, _encode_prefill([solo]))
                         self._prefill_batch([solo])
                 if self._running:
-                    self._tp_bcast(("decode",))
+                    self._op_bcast(OP_DECODE)
                     self._decode_step()
                 if self._pending is not None:
                     p, self._pending = self._pending, None

@@ -78,17 +78,18 @@ def shared_engine(config: Any, device: str | None = None, *, tp: int = 1,
             else:
                 Shard = LlamaShard
             if tp > 1 or ep > 1:
-                from ..parallel import get_dist
+                from ..parallel import get_dist, init_model_parallel
 
                 dist = get_dist()
-                world = max(tp, ep)
+                world = tp * ep  # EP composes with TP (r02)
                 if dist.world_size != world:
                     raise ValueError(
-                        f"tp={tp}/ep={ep} but torch.distributed world is "
+                        f"tp={tp} x ep={ep} but torch.distributed world is "
                         f"{dist.world_size}")
+                init_model_parallel(tp, ep)
                 if config.is_moe:
-                    model = Shard(config, tp_rank=dist.rank if tp > 1 else 0,
-                                  tp_size=tp, ep_rank=dist.rank if ep > 1 else 0,
+                    model = Shard(config, tp_rank=dist.rank % tp,
+                                  tp_size=tp, ep_rank=dist.rank // tp,
                                   ep_size=ep, device=device)
                 else:
                     model = Shard(config, tp_rank=dist.rank, tp_size=tp,

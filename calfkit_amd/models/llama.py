@@ -132,6 +132,11 @@ class LlamaShard:
                                          self.dh, cfg.rope_theta,
                                          device=self.device)
         self._splitk_cache: dict[int, int] = {}
+        #: comm stream for the TP-overlap decode pipeline (north star:
+        #: "RCCL all-reduce over xGMI overlapped with the GEMMs")
+        self._comm_stream = (torch.cuda.Stream(device=self.device)
+                             if tp_size > 1 and self.device.type == "cuda"
+                             else None)
 
     # ------------------------------------------------------- GEMM dispatch
     def _use_skinny(self, T: int, n_out: int, k_in: int) -> bool:
@@ -215,7 +220,8 @@ class LlamaShard:
               prefill_kv: tuple[torch.Tensor, torch.Tensor] | None,
               plan: "PrefillPlan | None",
               page_table: torch.Tensor | None,
-              seq_lens: torch.Tensor | None) -> torch.Tensor:
+              seq_lens: torch.Tensor | None,
+              do_reduce: bool = True) -> torch.Tensor:
         T = hidden.shape[0]
         decode = plan is None
         qkv_w = lw.wqkv
@@ -290,10 +296,10 @@ class LlamaShard:
                 Mp = ops.tile_gemm_slabs(o2, lw.wo, self._gemm_work, bn, sk)
                 return ("slabs", Mp, sk)
         out = self._proj(o2, lw.wo, decode=decode)
-        return all_reduce_sum(out) if self.tp_size > 1 else out
+        return all_reduce_sum(out) if (self.tp_size > 1 and do_reduce) else out
 
     def _mlp(self, lw: LayerWeights, hidden: torch.Tensor, *,
-             decode: bool = False):
+             decode: bool = False, do_reduce: bool = True):
         T = hidden.shape[0]
         act = torch.empty(T, self.ffn_s, dtype=hidden.dtype,
                           device=hidden.device)
@@ -319,7 +325,83 @@ class LlamaShard:
                                          bn, sk)
                 return ("slabs2", Mp, sk)
         down = self._proj(act, lw.wdown, decode=decode)
-        return all_reduce_sum(down) if self.tp_size > 1 else down
+        return all_reduce_sum(down) if (self.tp_size > 1 and do_reduce) \
+            else down
+
+    # ------------------------------------------------- TP-overlap decode
+    def _forward_tp_overlap(self, tokens, positions, kcaches, vcaches,
+                            slots, *, page_table, seq_lens, logits_rows):
+        """Decode forward as a two-chunk software pipeline: chunk c's
+        row-parallel all-reduce runs on the comm stream WHILE chunk 1-c
+        computes, with event ordering both ways (north star: RCCL
+        all-reduce over xGMI overlapped with the GEMMs).
+
+        Every rank issues the same chunk/collective order, so TP lockstep
+        is preserved; on CPU (gloo test lane) the same code runs with the
+        collectives inline — numerics are identical because every op in
+        the chunk pipeline is row-independent."""
+        T = tokens.shape[0]
+        h = T // 2
+        cs = (slice(0, h), slice(h, T))
+        on_gpu = self.device.type == "cuda" and self._comm_stream is not None
+        main = torch.cuda.current_stream() if on_gpu else None
+        comm = self._comm_stream
+
+        emb = self.embed[tokens.long()]
+        residual = torch.zeros_like(emb)
+        hid = [emb[cs[0]], emb[cs[1]]]
+        ev: list = [None, None]
+
+        def reduce_async(c: int, t: torch.Tensor) -> None:
+            if not on_gpu:
+                all_reduce_sum(t)
+                return
+            e1 = torch.cuda.Event()
+            e1.record(main)
+            comm.wait_event(e1)
+            with torch.cuda.stream(comm):
+                t.record_stream(comm)
+                all_reduce_sum(t)
+            e2 = torch.cuda.Event()
+            e2.record(comm)
+            ev[c] = e2
+
+        def wait_reduce(c: int) -> None:
+            if on_gpu and ev[c] is not None:
+                main.wait_event(ev[c])
+                ev[c] = None
+
+        eps = self.cfg.norm_eps
+        for i, lw in enumerate(self.layers):
+            nxt: list = [None, None]
+            for c in (0, 1):
+                wait_reduce(c)
+                ops.fused_add_rmsnorm(hid[c], residual[cs[c]], lw.norm_attn,
+                                      eps)
+                o = self._attn(lw, hid[c], positions[cs[c]], kcaches[i],
+                               vcaches[i], slots[cs[c]], prefill_kv=None,
+                               plan=None, page_table=page_table[cs[c]],
+                               seq_lens=seq_lens[cs[c]], do_reduce=False)
+                reduce_async(c, o)
+                nxt[c] = o
+            for c in (0, 1):
+                wait_reduce(c)
+                hid[c] = nxt[c]
+                ops.fused_add_rmsnorm(hid[c], residual[cs[c]], lw.norm_mlp,
+                                      eps)
+                d = self._mlp(lw, hid[c], decode=True, do_reduce=False)
+                reduce_async(c, d)
+                nxt[c] = d
+            hid = nxt
+        logits = [None, None]
+        for c in (0, 1):
+            wait_reduce(c)
+            ops.fused_add_rmsnorm(hid[c], residual[cs[c]], self.norm_final,
+                                  eps)
+            hc = hid[c] if logits_rows is None else hid[c]
+            logits[c] = self._proj(hc, self.lm_head, decode=True)
+        out = torch.cat(logits)
+        return out if logits_rows is None else out[logits_rows]
 
     # --------------------------------------------------------------- forward
     def make_single_plan(self, scratch, T: int, past_len: int) -> "PrefillPlan":
@@ -365,6 +447,15 @@ class LlamaShard:
                                          past_len)
         elif plan is not None:
             prefill_scratch = plan.scratch
+        if plan is None and prefill_scratch is None and self.tp_size > 1:
+            import os
+
+            if (tokens.shape[0] >= 2
+                    and os.environ.get("CALF_TP_OVERLAP", "1") == "1"):
+                return self._forward_tp_overlap(
+                    tokens, positions, kcaches, vcaches, slots,
+                    page_table=page_table, seq_lens=seq_lens,
+                    logits_rows=logits_rows)
         hidden = self.embed[tokens.long()]  # [T, d]
         residual = torch.zeros_like(hidden)
         for i, lw in enumerate(self.layers):

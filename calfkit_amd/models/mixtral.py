@@ -54,10 +54,10 @@ class MixtralShard(LlamaShard):
         super().__init__(cfg, **kw)
         # Expert parallelism: experts partition over the EP group (attention
         # and router stay replicated; tokens travel to their experts over
-        # RCCL all-to-all on xGMI).  Round-1 scope: EP and TP are alternative
-        # layouts for the MoE MLP — EP>1 requires TP=1.
-        if ep_size > 1 and self.tp_size > 1:
-            raise ValueError("EP>1 with TP>1 is not supported yet")
+        # RCCL all-to-all on xGMI).  EP composes with TP (r02): world =
+        # tp x ep, each expert's ffn TP-shards within the owner's TP group
+        # (parallel.init_model_parallel builds the groups) and the combine
+        # all-reduces over TP after the EP combine.
         if cfg.num_experts % ep_size:
             raise ValueError(f"{cfg.num_experts} experts must divide ep={ep_size}")
         self.ep_rank = ep_rank
@@ -114,7 +114,7 @@ class MixtralShard(LlamaShard):
             self._moe_work = torch.empty(0)
 
     def _mlp(self, lw: MoeLayerWeights, hidden: torch.Tensor, *,
-             decode: bool = False) -> torch.Tensor:
+             decode: bool = False, do_reduce: bool = True) -> torch.Tensor:
         T, d = hidden.shape
         e = self.cfg.num_experts
         k = self.topk
@@ -123,7 +123,9 @@ class MixtralShard(LlamaShard):
         weights = weights / weights.sum(-1, keepdim=True)   # renormalize top-k
 
         if self.ep_size > 1:
-            return self._mlp_ep(lw, hidden, weights, experts)
+            out = self._mlp_ep(lw, hidden, weights, experts,
+                               do_reduce=do_reduce)
+            return out
 
         if decode and self.device.type == "cuda" and ops.have_extension() \
                 and T <= 256:
@@ -140,7 +142,8 @@ class MixtralShard(LlamaShard):
                                 device=hidden.device)
             out32.index_add_(0, tok.long(), y.float())  # k rows per token
             out = out32.to(hidden.dtype)
-            return all_reduce_sum(out) if self.tp_size > 1 else out
+            return all_reduce_sum(out) \
+                if (self.tp_size > 1 and do_reduce) else out
 
         flat_expert = experts.reshape(-1)                   # [T*k]
         flat_token = (torch.arange(T, device=hidden.device)
@@ -166,10 +169,11 @@ class MixtralShard(LlamaShard):
                            (ys.float() * w_sorted[start:start + n, None])
                            .to(out.dtype))
             start += n
-        return all_reduce_sum(out) if self.tp_size > 1 else out
+        return all_reduce_sum(out) if (self.tp_size > 1 and do_reduce) else out
 
     def _mlp_ep(self, lw: MoeLayerWeights, hidden: torch.Tensor,
-                weights: torch.Tensor, experts: torch.Tensor) -> torch.Tensor:
+                weights: torch.Tensor, experts: torch.Tensor,
+                do_reduce: bool = True) -> torch.Tensor:
         """Expert-parallel MLP: dispatch tokens to their experts' owner ranks
         over all-to-all, compute the local expert shard, combine back.
 
@@ -222,4 +226,8 @@ class MixtralShard(LlamaShard):
 
         out32 = torch.zeros(T, d, dtype=torch.float32, device=hidden.device)
         out32.index_add_(0, tok.long(), y_back.float() * scale[:, None])
-        return out32.to(hidden.dtype)
+        out = out32.to(hidden.dtype)
+        # each expert's ffn is TP-sharded: partial sums reduce over the TP
+        # group after the EP combine (one [T,d] all-reduce)
+        return all_reduce_sum(out) if (self.tp_size > 1 and do_reduce) \
+            else out

@@ -13,14 +13,26 @@ from .pg import (
     all_reduce_sum,
     all_to_all_rows,
     barrier,
+    broadcast_tensor,
+    ep_group,
+    ep_rank,
     get_dist,
     init_distributed,
+    init_model_parallel,
     is_initialized,
     shard_dim,
+    tp_group,
+    tp_rank,
 )
 
 __all__ = [
     "DistInfo",
+    "broadcast_tensor",
+    "ep_group",
+    "ep_rank",
+    "init_model_parallel",
+    "tp_group",
+    "tp_rank",
     "all_reduce_sum",
     "all_to_all_rows",
     "barrier",

@@ -20,7 +20,9 @@ import torch
 import torch.distributed as dist
 
 __all__ = ["DistInfo", "init_distributed", "get_dist", "is_initialized",
-           "all_reduce_sum", "all_to_all_rows", "barrier", "shard_dim"]
+           "all_reduce_sum", "all_to_all_rows", "barrier", "shard_dim",
+           "init_model_parallel", "tp_group", "ep_group", "tp_rank",
+           "ep_rank", "broadcast_tensor"]
 
 
 @dataclass(frozen=True)
@@ -77,37 +79,103 @@ def is_initialized() -> bool:
     return _DIST is not None
 
 
-def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
-    """In-place sum all-reduce across the TP group (no-op at world 1)."""
+# ---------------------------------------------------------- TP x EP groups
+#
+# Composition (VERDICT r01 #2 "compose EP x TP"): world = tp * ep; rank r
+# has tp_rank = r % tp (fast axis: TP peers are xGMI neighbours) and
+# ep_rank = r // tp.  Attention/dense collectives run in the TP group,
+# MoE token exchanges in the EP group.  With a single axis (tp == world or
+# ep == world) no subgroups are created and the default WORLD group serves.
+
+_TP_GROUP = None
+_EP_GROUP = None
+_TP = 1
+_EP = 1
+
+
+def init_model_parallel(tp: int = 1, ep: int = 1) -> None:
+    global _TP_GROUP, _EP_GROUP, _TP, _EP
     d = get_dist()
-    if d.world_size > 1:
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    if tp * ep != d.world_size:
+        raise ValueError(f"tp({tp}) x ep({ep}) != world {d.world_size}")
+    _TP, _EP = tp, ep
+    _TP_GROUP = _EP_GROUP = None
+    if tp > 1 and ep > 1:
+        # every rank must create every group, in the same order
+        for e in range(ep):
+            g = dist.new_group(list(range(e * tp, (e + 1) * tp)))
+            if e == d.rank // tp:
+                _TP_GROUP = g
+        for t in range(tp):
+            g = dist.new_group(list(range(t, d.world_size, tp)))
+            if t == d.rank % tp:
+                _EP_GROUP = g
+
+
+def tp_group():
+    return _TP_GROUP
+
+
+def ep_group():
+    return _EP_GROUP
+
+
+def tp_rank() -> int:
+    return get_dist().rank % _TP if _TP > 1 else 0
+
+
+def ep_rank() -> int:
+    return get_dist().rank // _TP if _EP > 1 else 0
+
+
+def _group_size(group) -> int:
+    return dist.get_world_size(group=group) if group is not None \
+        else get_dist().world_size
+
+
+def broadcast_tensor(t: torch.Tensor, src: int = 0) -> torch.Tensor:
+    """World broadcast of a tensor (the lockstep op channel)."""
+    if get_dist().world_size > 1:
+        dist.broadcast(t, src=src)
+    return t
+
+
+def all_reduce_sum(t: torch.Tensor, group=None) -> torch.Tensor:
+    """In-place sum all-reduce across the TP group (no-op at group size 1)."""
+    d = get_dist()
+    if d.world_size == 1:
+        return t
+    g = group if group is not None else _TP_GROUP
+    if g is not None or _EP == 1:
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=g)
     return t
 
 
 def all_to_all_rows(t: torch.Tensor, send_splits: list[int],
-                    recv_splits: list[int] | None = None,
+                    recv_splits: list[int] | None = None, group=None,
                     ) -> tuple[torch.Tensor, list[int]]:
-    """Variable-row all-to-all: rank r receives t's send_splits[r] rows from
-    every peer, concatenated in rank order.
+    """Variable-row all-to-all over the EP group: rank r receives
+    t's send_splits[r] rows from every peer, concatenated in rank order.
 
     The expert-parallel token exchange (MoE dispatch/combine).  When
     recv_splits is unknown it is exchanged first with a counts all-to-all —
-    a host sync, which is why EP decode runs eager (the
-    fixed-capacity capturable variant is the planned next step).  On GPU
-    this rides RCCL's point-to-point xGMI sends; gloo serves the CPU lane.
+    a host sync, which is why EP decode runs eager (capacity
+    padding was analyzed and rejected, STATUS r01).  On GPU this rides
+    RCCL's point-to-point xGMI sends; gloo serves the CPU lane.
     """
     d = get_dist()
-    if d.world_size == 1:
+    g = group if group is not None else _EP_GROUP
+    if d.world_size == 1 or (g is None and _TP > 1 and _EP == 1):
         return t, [t.shape[0]]
     if recv_splits is None:
         counts = torch.tensor(send_splits, dtype=torch.int64, device=t.device)
         rcounts = torch.empty_like(counts)
-        dist.all_to_all_single(rcounts, counts)
+        dist.all_to_all_single(rcounts, counts, group=g)
         recv_splits = [int(c) for c in rcounts]
     out = torch.empty(sum(recv_splits), *t.shape[1:], dtype=t.dtype,
                       device=t.device)
-    dist.all_to_all_single(out, t.contiguous(), recv_splits, send_splits)
+    dist.all_to_all_single(out, t.contiguous(), recv_splits, send_splits,
+                           group=g)
     return out, recv_splits
 
 
