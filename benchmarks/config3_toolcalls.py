@@ -2,9 +2,11 @@
 """BASELINE config 3: concurrent tool-call bench — 1 Agent + 4 @agent_tool
 nodes per worker replica, Llama-3-8B provider, replicas pinned 1/GPU.
 
-Each turn: prefill + decode (scripted into 4 parallel tool calls), 4 tool
-hops over the mesh, then a second full model step for the final answer —
-the whole thing through the wire.  Run under torchrun for N replicas:
+Each turn: prefill + decode steered by the ENGINE's tool-call grammar into
+4 parallel tool calls (zero ToolCallDirector — the sampled tokens ARE the
+calls), 4 tool hops over the mesh, then a second full model
+step for the final answer — the whole thing through the wire.  Run under
+torchrun for N replicas:
 
   torchrun --standalone --nproc-per-node N benchmarks/config3_toolcalls.py
 """
@@ -35,29 +37,32 @@ async def run() -> None:
     from calfkit_amd.models.config import get_config
     from calfkit_amd.parallel import barrier, init_distributed
     from calfkit_amd.schemas import ModelSettings
-    from calfkit_amd.testing import ToolCallDirector
 
+    # 4-way tool fan-out per turn, sampled by the engine's grammar
+    os.environ.setdefault("CALF_TOOLCALL_FANOUT", "4")
     dist = init_distributed()
     on_gpu = torch.cuda.is_available()
     model = "llama-3-8b" if on_gpu else "tiny"
 
+    tool_hits = [0]
     tools = []
     for i in range(4):
         def make(i=i):
             def fn(query: str) -> str:
+                tool_hits[0] += 1
                 return f"tool{i} result for {query}"
             fn.__name__ = f"search_{i}"
             fn.__doc__ = f"Search backend {i}."
             return fn
         tools.append(agent_tool(make()))
 
-    inner = LocalModelClient(get_config(model), max_batch=TURNS_PER_STEP * 2,
-                             max_model_len=1024, use_graphs=on_gpu)
-    provider = ToolCallDirector(
-        inner, calls=[(f"search_{i}", {"query": "benchmark"}) for i in range(4)])
+    provider = LocalModelClient(get_config(model),
+                                max_batch=TURNS_PER_STEP * 2,
+                                max_model_len=1024, use_graphs=on_gpu)
     agent = Agent(name="toolcaller", tools=[t.name for t in tools],
                   model_client=provider,
-                  model_settings=ModelSettings(max_tokens=32, temperature=0.0))
+                  model_settings=ModelSettings(max_tokens=32,
+                                               temperature=0.7, seed=17))
     bus, store, directory = InMemoryBus(), InProcessThreadStore(), InMemoryDirectory()
     lat: list[float] = []
 
@@ -105,7 +110,9 @@ async def run() -> None:
             "unit": "turns/s",
             "n_gpus": world,
             "p50_turn_ms": round(statistics.median(lat), 1),
-            "config": {"model": model, "tools": 4,
+            "tool_calls_per_turn": round(
+                tool_hits[0] / max(1, (STEPS + WARMUP) * TURNS_PER_STEP), 2),
+            "config": {"model": model, "tools": 4, "director": "none",
                         "replicas": f"{world} worker(s), 1/GPU",
                         "turns_per_step": TURNS_PER_STEP},
         }), flush=True)

@@ -122,3 +122,70 @@ def test_ep_lockstep_serving_completes():
         p.join(timeout=120)
         assert p.exitcode == 0
     assert len(out) == 4
+
+
+def _worker_loop_rank(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import asyncio
+
+    import calfkit_amd.parallel.pg as pg
+
+    pg._DIST = None
+    pg.init_distributed(backend="gloo", device_override="cpu")
+
+    if rank != 0:
+        from calfkit_amd.engine.local import shared_engine
+
+        engine = shared_engine(CFG, "cpu", ep=2, max_batch=4,
+                               use_graphs=False, max_model_len=128,
+                               autostart=False)
+        engine.follow()
+    else:
+        async def main():
+            from calfkit_amd import Agent, Client, Worker
+            from calfkit_amd.adapters.memory import (
+                InMemoryBus,
+                InMemoryDirectory,
+                InProcessThreadStore,
+            )
+            from calfkit_amd.engine.local import LocalModelClient
+            from calfkit_amd.schemas import ModelSettings
+
+            provider = LocalModelClient(CFG, device="cpu", ep=2, max_batch=4,
+                                        use_graphs=False, max_model_len=128)
+            agent = Agent(name="moe-server", model_client=provider,
+                          model_settings=ModelSettings(max_tokens=4,
+                                                       temperature=0.0))
+            bus, store, d = (InMemoryBus(), InProcessThreadStore(),
+                             InMemoryDirectory())
+            async with Worker(agent, bus=bus, store=store, directory=d):
+                client = await Client.connect(bus, d)
+                r = await client.execute("moe-server", "hello experts",
+                                         timeout=240)
+                assert r.text.strip()
+                q.put(r.text)
+            provider.engine.stop()
+
+        asyncio.run(main())
+    import torch.distributed as td
+
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_ep_under_the_serving_worker():
+    """the EP engine serves under the REAL Worker loop —
+    agent plane on rank 0, EP follower on rank 1, full mesh round trip."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_loop_rank, args=(r, 2, 29531, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    text = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    assert text.strip()
