@@ -88,6 +88,7 @@ async def run_rank(args, dist) -> dict:
     prompt_words = " ".join(f"w{i}" for i in range(PROMPT_TOKENS - 8))
 
     latencies: list[float] = []
+    failures: list[str] = []
 
     async def one_turn(client, i: int, record: bool) -> None:
         t0 = time.perf_counter()
@@ -111,8 +112,16 @@ async def run_rank(args, dist) -> dict:
             counter += 1
 
             async def turn(i=counter):
-                await one_turn(client, i, recording)
-                done.put_nowait(None)
+                # a turn failure must not starve drain() into a silent
+                # hang: account for it and keep the pool moving
+                try:
+                    await one_turn(client, i, recording)
+                except asyncio.CancelledError:
+                    raise
+                except Exception as e:
+                    failures.append(repr(e))
+                finally:
+                    done.put_nowait(None)
 
             return asyncio.ensure_future(turn())
 
@@ -139,10 +148,28 @@ async def run_rank(args, dist) -> dict:
         recording = False
         for t in tasks:
             t.cancel()
-        await asyncio.gather(*tasks, return_exceptions=True)
+        try:
+            await asyncio.wait_for(
+                asyncio.gather(*tasks, return_exceptions=True), 30)
+        except (asyncio.TimeoutError, Exception):  # noqa: BLE001
+            pass
 
+    if failures:
+        log(f"BENCH INVALID: {len(failures)} turn(s) failed; first: "
+            f"{failures[0]}")
+        sys.exit(3)
     if broker is not None:
-        await broker.stop()
+        # bounded, never-raising teardown: the measurement is already in
+        # hand and broker/connection close races must not lose it
+        for c in (client_bus, bus):
+            try:
+                await asyncio.wait_for(c.close(), 5)
+            except Exception:
+                pass
+        try:
+            await asyncio.wait_for(broker.stop(), 5)
+        except Exception:
+            pass
     return {"elapsed": elapsed, "latencies": latencies,
             "engine_stats": provider.engine.stats, "model": cfg.name}
 

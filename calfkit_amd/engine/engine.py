@@ -803,11 +803,9 @@ class InferenceEngine:
         gnames = np.full((pad, gops.G_MAX_TOOLS), -1, dtype=np.int32)
         gncalls = np.zeros(pad, dtype=np.int32)
         gstates = np.zeros(pad, dtype=np.int32)
-        any_tools = False
         for b, s in enumerate(self._running):
             if not s.request.tool_names:
                 continue
-            any_tools = True
             toks = s.result.token_ids
             while s.gconsumed < len(toks) - 1:
                 s.gstate = gops.grammar_transition(s.gstate, toks[s.gconsumed])
@@ -817,7 +815,6 @@ class InferenceEngine:
             gncalls[b] = s.request.tool_ncalls
             ids = s.request.tool_names[:gops.G_MAX_TOOLS]
             gnames[b, :len(ids)] = ids
-        self._any_tools = any_tools
         self._gstate[:pad].copy_(torch.from_numpy(gstates), non_blocking=True)
         self._gtrigger[:pad].copy_(torch.from_numpy(gtrig), non_blocking=True)
         self._gnames[:pad].copy_(torch.from_numpy(gnames), non_blocking=True)

@@ -59,7 +59,10 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
   constexpr int A_ELEMS = BM * BK;
   constexpr int B_ELEMS = BN * BK;
   constexpr int BUF = A_ELEMS + B_ELEMS;
-  __shared__ __attribute__((aligned(16))) __hip_bfloat16 smem[2 * BUF];
+  // FLAGS bit 16: 3-deep LDS ring (BN=128 fits 3 x 48 KiB in 160 KiB) —
+  // units stage TWO tiles ahead, so every load has ~2 tiles of flight
+  constexpr int BUFS = (BN == 128 && (FLAGS & 16)) ? 3 : 2;
+  __shared__ __attribute__((aligned(16))) __hip_bfloat16 smem[BUFS * BUF];
 
   // ---- block id -> (n_tile, m_tile, sk) with XCD-bijective remap (T1)
   int bid = blockIdx.x;
@@ -150,15 +153,29 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
     stage_a(0, 0, 1);
   }
 
-  // pre-loop: tile 0's first-phase units landed (B0,B1,A0; A1 still in
-  // flight), all waves synced before any LDS read
-  VMWAIT(2);
+  if (BUFS == 3 && nt > 1) {  // deep ring: tile 1 prefetches up front too
+    stage_b(1, 1, 0);
+    stage_a(1, 1, 0);
+    stage_a(1, 1, 1);
+  }
+
+  // pre-loop: tile 0's units landed, all waves synced before any LDS read
+  if (BUFS == 3) {
+    if (nt > 1) {
+      VMWAIT(6);
+    } else {
+      VMWAIT(0);
+    }
+  } else {
+    VMWAIT(2);
+  }
   __builtin_amdgcn_s_barrier();
 
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
     const bool pf = t + 1 < nt;
-    const int nb = cur ^ 1;
+    const int nb = BUFS == 3 ? (t + 2) % 3 : (cur ^ 1);
+    const bool pf2 = BUFS == 3 ? (t + 2 < nt) : pf;
     // Phase body order matters (PMC r02: stage-issue at phase head cost
     // ~300 cyc/phase of MFMA-pipe idle): ds_reads FIRST, then the stage
     // issue, then barrier -> MFMA cluster.  Counted waits sit at a
@@ -236,13 +253,14 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
 #pragma unroll
           for (int ks = 0; ks < 2; ++ks)
             afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+        const int st = BUFS == 3 ? t + 2 : t + 1;
         const auto stage_ph = [&] {
-          if (!pf) return;
+          if (!pf2) return;
           if (ph == 0) {
-            stage_b(nb, t + 1, 0);
-            stage_a(nb, t + 1, 0);
+            stage_b(nb, st, 0);
+            stage_a(nb, st, 0);
           } else {
-            stage_a(nb, t + 1, 1);
+            stage_a(nb, st, 1);
           }
         };
         if constexpr (!(FLAGS & 4)) stage_ph();
@@ -259,6 +277,21 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
                   MFMA16(afr[mi][ks], bfr[ni][ks], acc[ph * 2 + mi][ni]);
         }
         if constexpr (!(FLAGS & 2)) __builtin_amdgcn_s_setprio(0);
+        if (BUFS == 3) {
+          // deep ring: tile t's units landed >= 2 tiles ago; the only
+          // wait is at ph1's tail, covering tile t+1's whole set
+          if (ph == 1 && pf) {
+            if (pf2) {
+              VMWAIT(6);
+            } else {
+              VMWAIT(0);
+            }
+            __builtin_amdgcn_s_barrier();
+            continue;
+          }
+          if (ph == 1) __builtin_amdgcn_s_barrier();
+          continue;
+        }
         if (ph == 0) {
           // ph1 reads A1(t): outstanding {A1, B', A0'} -> vmcnt(4)
           if (pf) {
@@ -273,7 +306,7 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
         __builtin_amdgcn_s_barrier();
       }
     }
-    cur ^= 1;
+    cur = BUFS == 3 ? (t + 1) % 3 : (cur ^ 1);
   }
 
   // ---- epilogue.  C layout (mfma_f32_16x16x32): col = l16, row = l4*4+r.
@@ -346,6 +379,8 @@ static void launch_tile(torch::Tensor work, torch::Tensor out,
     case 6: LT(BNV, 6); break;                                               \
     case 7: LT(BNV, 7); break;                                               \
     case 15: LT(BNV, 15); break;                                             \
+    case 23: LT(BNV, 23); break;                                             \
+    case 31: LT(BNV, 31); break;                                             \
     default: LT(BNV, 0); break;                                              \
   }
   if (bn == 256) {

@@ -115,7 +115,11 @@ class Worker:
 
     async def stop(self) -> None:
         for p in self.processors:
-            await self.directory.withdraw(p.node_name)
+            try:
+                await self.directory.withdraw(p.node_name)
+            except Exception:  # a dead transport must not wedge shutdown
+                _log.warning("withdraw failed for %s", p.node_name,
+                             exc_info=True)
             await p.stop()
         await self.lifecycle.shutdown()
         self._stop.set()
