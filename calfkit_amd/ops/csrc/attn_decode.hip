@@ -24,7 +24,7 @@
 #define KVPAGE 64
 #define HD 128
 
-template <int G, int NW>
+template <int G, int NW, bool PFETCH>
 __global__ __launch_bounds__(NW * WAVE) void attn_decode4_kernel(
     __hip_bfloat16* __restrict__ out,
     const __hip_bfloat16* __restrict__ q,
@@ -70,14 +70,14 @@ __global__ __launch_bounds__(NW * WAVE) void attn_decode4_kernel(
   // software-pipelined page ids: the table read is a dependent scalar load
   // (L2/HBM latency) feeding every K/V address — prefetch one iteration
   // ahead so the next page's id is in flight during this page's stream
-  int page_next = (wid < npages) ? page_table[tbl + wid] : 0;
+  int page_next = PFETCH && (wid < npages) ? page_table[tbl + wid] : 0;
   for (int pi = wid; pi < npages; pi += NW) {
     // bar the compiler from hoisting the per-page q_lds re-reads out of
     // this loop: without it LICM keeps G*128 floats live across pages
     // (251-747 VGPR spills measured; see the r01/r03 kernel profiles)
     asm volatile("" ::: "memory");
-    const int page = page_next;
-    if (pi + NW < npages) page_next = page_table[tbl + pi + NW];
+    const int page = PFETCH ? page_next : page_table[tbl + pi];
+    if (PFETCH && pi + NW < npages) page_next = page_table[tbl + pi + NW];
     const int64_t base = (((int64_t)page * hkv + kvh) * KVPAGE) * HD;
     const int token = pi * KVPAGE + lane;
     const bool live = token < slen;
@@ -247,10 +247,17 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   const int G = hq / hkv;
   const char* nw_env = getenv("CALF_ATTN_NW");
   const int nw = nw_env ? atoi(nw_env) : 2;
+  const char* pf_env = getenv("CALF_ATTN_PFETCH");
+  const int pfe = pf_env ? atoi(pf_env) : 0;  // prefetch LOST the r02 A/B (120.8 vs 168.1 us)
   dim3 grid(B, hkv), block(nw * WAVE);
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
 #define LAUNCH_NW(GV, NWV)                                                \
-  hipLaunchKernelGGL((attn_decode4_kernel<GV, NWV>), grid, block, 0, s,   \
+  do {                                                                    \
+    if (pfe) LAUNCH_NWP(GV, NWV, true);                                   \
+    else LAUNCH_NWP(GV, NWV, false);                                      \
+  } while (0)
+#define LAUNCH_NWP(GV, NWV, PFV)                                          \
+  hipLaunchKernelGGL((attn_decode4_kernel<GV, NWV, PFV>), grid, block, 0, s, \
                      (__hip_bfloat16*)out.data_ptr(),                     \
                      (const __hip_bfloat16*)q.data_ptr(),                 \
                      (const __hip_bfloat16*)kcache.data_ptr(),            \
@@ -272,5 +279,6 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   }
 #undef LAUNCH
 #undef LAUNCH_NW
+#undef LAUNCH_NWP
   HIP_CHECK_LAST();
 }

@@ -361,7 +361,7 @@ static void launch_tile(torch::Tensor work, torch::Tensor out,
   TORCH_CHECK(direct || work.numel() >= (int64_t)splitk * Mp * N,
               "tile_gemm: workspace too small");
   const char* fv = getenv("CALF_TILE_FLAGS");
-  const int flags = fv ? atoi(fv) : 15;
+  const int flags = fv ? atoi(fv) : 31;
 #define LT(BNV, F)                                                           \
   hipLaunchKernelGGL((tile_gemm_kernel<BNV, F>), dim3(blocks), dim3(512), 0, \
                      s, (float*)work.data_ptr(),                             \
