@@ -185,20 +185,52 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
       // 4 phases of 16 MFMA (one m-quadrant x full BN); stage units of
       // t+1 issue one-per-phase in first-need order B0,B1,A0,A1.
       bf16x8v bfr[NI][2];
+      // FLAGS bit 32: A-fragment double-buffer — phase p+1's ds_reads
+      // issue BEFORE phase p's MFMA cluster, so their ~50-100 cyc LDS
+      // latency hides under the MFMAs instead of parking the wave at the
+      // next phase head (r02 PMC: 37.6% of wave time parked).
+      bf16x8v af0[2][2], af1[2][2];
 #pragma unroll
       for (int ph = 0; ph < 4; ++ph) {
+        bf16x8v(&afr)[2][2] = (ph & 1) ? af1 : af0;
+        bf16x8v(&afn)[2][2] = (ph & 1) ? af0 : af1;
         if (ph == 0) {
 #pragma unroll
           for (int ni = 0; ni < NI; ++ni)
 #pragma unroll
             for (int ks = 0; ks < 2; ++ks) bfr[ni][ks] = read_b(cur, ni, ks);
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks) afr[mi][ks] = read_a(cur, mi, ks);
         }
-        bf16x8v afr[2][2];
+        if constexpr ((FLAGS & 32) != 0) {
+          // prefetch only WITHIN an A-half: A(ph+1) may be read at ph's
+          // head only when it shares ph's stage half (phases 0->1, 2->3);
+          // crossing the half boundary would read A1 before its ph1-tail
+          // wait (the r02 f47/f63 numerics failures)
+          if (ph == 2) {
+            // own read first (half1 just landed behind the ph1-tail wait)
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-          for (int ks = 0; ks < 2; ++ks)
-            afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+              for (int ks = 0; ks < 2; ++ks)
+                afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+          }
+          if (ph == 0 || ph == 2) {
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+              for (int ks = 0; ks < 2; ++ks)
+                afn[mi][ks] = read_a(cur, (ph + 1) * 2 + mi, ks);
+          }
+        } else if (ph > 0) {
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks)
+              afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+        }
         const auto stage_ph = [&] {
           if (!pf) return;
           if (ph == 0) stage_b(nb, t + 1, 0);
@@ -239,20 +271,38 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
     } else {
       // BN=128: 2 phases of 16 MFMA (2 mi x 4 ni); units B,A0,A1.
       bf16x8v bfr[NI][2];
+      bf16x8v af0[2][2], af1[2][2];
 #pragma unroll
       for (int ph = 0; ph < 2; ++ph) {
+        bf16x8v(&afr)[2][2] = (ph & 1) ? af1 : af0;
+        bf16x8v(&afn)[2][2] = (ph & 1) ? af0 : af1;
         if (ph == 0) {
 #pragma unroll
           for (int ni = 0; ni < NI; ++ni)
 #pragma unroll
             for (int ks = 0; ks < 2; ++ks) bfr[ni][ks] = read_b(cur, ni, ks);
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks) afr[mi][ks] = read_a(cur, mi, ks);
         }
-        bf16x8v afr[2][2];
+        if constexpr ((FLAGS & 32) != 0 && BUFS == 3) {
+          // safe only with the 3-deep ring: tile t's units landed behind
+          // the PREVIOUS tile's wait, so cross-half prefetch cannot race
+          if (ph < 1) {
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-          for (int ks = 0; ks < 2; ++ks)
-            afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+              for (int ks = 0; ks < 2; ++ks)
+                afn[mi][ks] = read_a(cur, 2 + mi, ks);
+          }
+        } else if (ph > 0) {
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks)
+              afr[mi][ks] = read_a(cur, ph * 2 + mi, ks);
+        }
         const int st = BUFS == 3 ? t + 2 : t + 1;
         const auto stage_ph = [&] {
           if (!pf2) return;
@@ -381,6 +431,8 @@ static void launch_tile(torch::Tensor work, torch::Tensor out,
     case 15: LT(BNV, 15); break;                                             \
     case 23: LT(BNV, 23); break;                                             \
     case 31: LT(BNV, 31); break;                                             \
+    case 47: LT(BNV, 47); break;                                             \
+    case 63: LT(BNV, 63); break;                                             \
     default: LT(BNV, 0); break;                                              \
   }
   if (bn == 256) {
