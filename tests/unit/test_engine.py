@@ -351,3 +351,32 @@ def test_tool_grammar_cpu():
 def test_tool_grammar_gpu_graphs():
     """The same grammar inside the hipGraph-captured multi-step decode."""
     _run_tool_grammar(device="cuda", use_graphs=True)
+
+
+def test_tool_call_parser_edges():
+    """Parser robustness: truncated span (budget ran out mid-call), unknown
+    name (dropped), interleaved text segments."""
+    from calfkit_amd.engine.local import _parse_output
+    from calfkit_amd.engine.tokenizer import SyntheticTokenizer
+    from calfkit_amd.messages import TextPart, ToolCallPart
+    from calfkit_amd.schemas import ToolSchema
+
+    tok = SyntheticTokenizer(512)
+    schema = ToolSchema(name="lookup", parameters={
+        "type": "object", "properties": {"q": {"type": "string"},
+                                         "n": {"type": "integer"}}})
+    nid = tok.word_id("lookup")
+    # text, full call, text, truncated call (no TEND)
+    ids = [30, 31, tok.TCALL, nid, 40, 41, 42, 43, tok.TEND, 50,
+           tok.TCALL, nid, 60, 61]
+    parts, calls = _parse_output(ids, (schema,), tok)
+    assert len(calls) == 2
+    assert all(isinstance(c, ToolCallPart) for c in calls)
+    assert set(calls[0].arguments) == {"q", "n"}
+    assert isinstance(calls[0].arguments["n"], int)
+    texts = [p for p in parts if isinstance(p, TextPart)]
+    assert texts and texts[0].text  # the leading text survived
+    # unknown name: span dropped entirely
+    parts2, calls2 = _parse_output([tok.TCALL, 499, 1, 2, tok.TEND, 70],
+                                   (schema,), tok)
+    assert not calls2

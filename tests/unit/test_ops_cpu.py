@@ -82,3 +82,42 @@ def test_sample_greedy_cpu():
     out = torch.empty(3, dtype=torch.int32)
     ops.sample(out, logits, torch.zeros(3), torch.tensor([1, 2]))
     assert (out == 7).all()
+
+
+def test_tile_plan_matches_measured_policy():
+    """The (BN, splitk) chooser reproduces the r02 sweep winners
+    (profiles/r02_tile_gemm.md) for every llama serving shape."""
+    import calfkit_amd.ops as ops
+
+    expect = {
+        (512, 6144, 4096): (128, 2),     # 8b qkv
+        (512, 4096, 4096): (128, 4),     # 8b o
+        (512, 28672, 4096): (256, 1),    # 8b gateup
+        (512, 4096, 14336): (128, 4),    # 8b down
+        (512, 128256, 4096): (256, 1),   # 8b lm_head
+        (512, 1280, 8192): (128, 8),     # 70b tp8 qkv
+        (512, 8192, 1024): (128, 1),     # 70b tp8 o (K too shallow to split)
+        (512, 7168, 8192): (128, 2),     # 70b tp8 gateup
+        (512, 8192, 3584): (128, 2),     # 70b tp8 down
+    }
+    for (m, n, k), plan in expect.items():
+        assert ops.tile_plan(m, n, k) == plan, (m, n, k)
+    # unroutable: N not divisible by 128
+    assert ops.tile_plan(512, 1000, 4096) is None
+
+
+def test_grammar_transition_state_machine():
+    from calfkit_amd import ops
+
+    s = 0
+    s = ops.grammar_transition(s, 99)            # free text stays free
+    assert s == 0
+    s = ops.grammar_transition(s, ops.G_TCALL)   # -> awaiting name
+    assert s & 0xFF == 1
+    s = ops.grammar_transition(s, 500)           # name consumed -> args
+    assert s & 0xFF == 2
+    for k in range(ops.G_ARG_TOKS):
+        s = ops.grammar_transition(s, 100 + k)
+    assert s & 0xFF == 2 + ops.G_ARG_TOKS
+    s = ops.grammar_transition(s, ops.G_TEND)    # closed: free, calls=1
+    assert s & 0xFF == 0 and s >> 8 == 1
