@@ -159,6 +159,14 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
     stage_a(1, 1, 1);
   }
 
+  // FLAGS bit 64 (T5 static form): the second-dispatched half of the
+  // workgroup loses VALU arbitration on every segment; ONE static
+  // priority raise for it (provably wave-uniform via readfirstlane)
+  if constexpr ((FLAGS & 64) != 0) {
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+      __builtin_amdgcn_s_setprio(1);
+  }
+
   // pre-loop: tile 0's units landed, all waves synced before any LDS read
   if (BUFS == 3) {
     if (nt > 1) {
@@ -411,7 +419,7 @@ static void launch_tile(torch::Tensor work, torch::Tensor out,
   TORCH_CHECK(direct || work.numel() >= (int64_t)splitk * Mp * N,
               "tile_gemm: workspace too small");
   const char* fv = getenv("CALF_TILE_FLAGS");
-  const int flags = fv ? atoi(fv) : 31;
+  const int flags = fv ? atoi(fv) : 95;  // 31 + T5 static young-half priority (r02 sweep)
 #define LT(BNV, F)                                                           \
   hipLaunchKernelGGL((tile_gemm_kernel<BNV, F>), dim3(blocks), dim3(512), 0, \
                      s, (float*)work.data_ptr(),                             \
@@ -433,6 +441,7 @@ static void launch_tile(torch::Tensor work, torch::Tensor out,
     case 31: LT(BNV, 31); break;                                             \
     case 47: LT(BNV, 47); break;                                             \
     case 63: LT(BNV, 63); break;                                             \
+    case 95: LT(BNV, 95); break;                                             \
     default: LT(BNV, 0); break;                                              \
   }
   if (bn == 256) {
