@@ -169,15 +169,25 @@ class LlamaShard:
         plan = ops.tile_plan(T, n_out, k_in)
         if plan is None or mode == "all":
             return plan
-        # measured-policy dispatch (profiles/r02_tile_gemm.md): route the
-        # shapes where the hand-written kernel beats hipBLASLt in the
-        # serving path today — deep-K down projections (1.03-1.05x raw,
-        # more with the fused slab consumer) and the vocab-wide lm_head
-        # (1.07-1.11x); the rest stay on the library until the kernel
-        # earns them (CALF_TILED=all forces everything tiled).
-        if k_in >= 3 * n_out or (n_out >= 32000 and n_out % 256 == 0):
-            return plan
-        return None
+        # measured-policy dispatch (profiles/r02_tile_gemm.md): shape
+        # classes route when the hand-written kernel matches or beats
+        # hipBLASLt in the serving path.  Default: deep-K down
+        # projections and the vocab-wide lm_head (clear wins); qkv is at
+        # fused-parity and opt-in.  CALF_TILED takes "all", "0", or a
+        # comma list of classes (lmhead,down,qkv,o,gateup).
+        if n_out >= 32000 and n_out % 256 == 0:
+            cls = "lmhead"
+        elif k_in >= 3 * n_out and n_out >= 2048:
+            cls = "down"
+        elif 1.2 * k_in <= n_out <= 2 * k_in:
+            cls = "qkv"
+        elif n_out == k_in:
+            cls = "o"
+        else:
+            cls = "gateup"
+        routed = (mode.split(",") if "," in mode or mode.isalpha()
+                  else ["lmhead", "down"])
+        return plan if cls in routed else None
 
     def _splitk_for(self, weight: torch.Tensor, m: int = 32) -> int:
         key = (weight.shape[0], m > 64)
