@@ -136,8 +136,12 @@ class LocalModelClient:
     # ------------------------------------------------------------ rendering
     @staticmethod
     def render_prompt(messages: tuple[Message, ...],
-                      tools: tuple = ()) -> str:
+                      tools: tuple = (), output: Any = None) -> str:
         lines = []
+        if output is not None:
+            keys = " ".join(sorted(
+                (output.json_schema or {}).get("properties", {})))
+            lines.append(f"<|output|> {output.name} ( {keys} )")
         if tools:
             # the model-facing tool advertisement (reference semantics:
             # providers render request.tools into the prompt —
@@ -156,7 +160,8 @@ class LocalModelClient:
     async def complete(self, request: EngineRequest, sink: ChunkSink) -> AssistantMessage:
         engine = self.engine
         tok = self.tokenizer
-        prompt = self.render_prompt(request.messages, request.tools)
+        prompt = self.render_prompt(request.messages, request.tools,
+                                    request.output)
         ids = tok.encode(prompt)
         s = request.settings
         seq = 0
@@ -206,6 +211,17 @@ class LocalModelClient:
         result = await asyncio.wrap_future(fut)
         parts, calls = _parse_output(result.token_ids, request.tools, tok)
         finish = "tool_calls" if calls else result.finish_reason
+        if request.output is not None and not calls:
+            # structured output from the ENGINE's sampled stream: the
+            # schema dictates the shape, the sampled tokens pick the
+            # values (same philosophy as tool-call args), emitted as the
+            # final_result-prefixed union the policy end-shape expects
+            import json as _json
+
+            value = _materialize_value(request.output.json_schema or {},
+                                       result.token_ids or [17], [0])
+            parts = (TextPart(_json.dumps({"final_result": value})),)
+            finish = "stop"
         if not calls and s.stop:
             # stop-sequence truncation (client side), text-only turns
             text = " ".join(p.text for p in parts
@@ -225,6 +241,32 @@ class LocalModelClient:
             decode_ms=result.decode_ms,
         )
         return AssistantMessage(parts, facts=facts)
+
+
+def _materialize_value(schema: dict, toks: list[int], cur: list[int]):
+    """Recursively fill a JSON schema from sampled tokens (engine-driven
+    structured output; cur = mutable token cursor)."""
+    def take() -> int:
+        v = toks[cur[0] % len(toks)]
+        cur[0] += 1
+        return v
+
+    t = (schema or {}).get("type", "object")
+    if t == "object":
+        return {k: _materialize_value(v or {}, toks, cur)
+                for k, v in sorted((schema.get("properties") or {}).items())}
+    if t == "array":
+        item = schema.get("items") or {"type": "string"}
+        return [_materialize_value(item, toks, cur)]
+    if t == "integer":
+        return take()
+    if t == "number":
+        return round(take() / 7.0, 3)
+    if t == "boolean":
+        return bool(take() % 2)
+    if t == "null":
+        return None
+    return f"t{take()}"
 
 
 def _materialize_args(schema: Any, toks: list[int]) -> dict:

@@ -107,3 +107,33 @@ async def test_engine_tool_call_round_trip_cpu():
 @pytest.mark.gpu
 async def test_engine_tool_call_round_trip_gpu():
     await _tool_call_round_trip("cuda", True)
+
+
+async def test_structured_output_from_the_engine():
+    """Engine-driven structured output (final_result union): the policy's
+    end-shape check passes on JSON materialized from the sampled stream,
+    and the client deserializes the typed result."""
+    from dataclasses import dataclass
+
+    from calfkit_amd import Agent, Client, Worker
+
+    @dataclass
+    class Weather:
+        city: str
+        temp_c: int
+
+    bus, store, directory = (InMemoryBus(), InProcessThreadStore(),
+                             InMemoryDirectory())
+    provider = LocalModelClient(TINY, device="cpu", max_batch=4,
+                                max_model_len=256, use_graphs=False)
+    agent = Agent(name="typed-engine", model_client=provider,
+                  output_type=Weather,
+                  model_settings=ModelSettings(max_tokens=12,
+                                               temperature=0.6, seed=9))
+    async with Worker(agent, bus=bus, store=store, directory=directory):
+        client = await Client.connect(bus, directory)
+        result = await client.execute("typed-engine", "weather please",
+                                      timeout=300, output_type=Weather)
+        assert isinstance(result.output, Weather)
+        assert isinstance(result.output.temp_c, int)
+        assert result.output.city.startswith("t")
