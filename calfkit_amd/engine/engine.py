@@ -39,8 +39,11 @@ _log = logging.getLogger("calfkit.engine")
 
 __all__ = ["GenRequest", "GenResult", "InferenceEngine"]
 
-PREFILL_CHUNK = 512
-PREFILL_BUDGET = 8192
+import os as _os
+
+PREFILL_CHUNK = int(_os.environ.get("CALF_PREFILL_CHUNK", "512"))
+#: token budget of one packed varlen prefill forward (the admission batch)
+PREFILL_BUDGET = int(_os.environ.get("CALF_PREFILL_BUDGET", "8192"))
 #: decode steps executed per host visit (graphs exist for each length).
 #: 8 is the measured sweet spot: 16-step runs halve the per-visit host
 #: cost but delay admission by a full run, and the running batch dips

@@ -532,3 +532,32 @@ def test_tile_gemm_matches_linear_serving_shape():
     ops.tile_gemm(out, x, w, work, bn, sk)
     ref = x.float() @ w.float().T
     assert torch.allclose(out.float(), ref, atol=8e-2, rtol=8e-2)
+
+
+def test_attn_decode_gqa8_vs_fp32():
+    """G=8 (the Llama-3-70B head ratio) decode attention vs the torch
+    reference — the single-GPU 70B serving path."""
+    B, hq, hkv, dh, slen = 16, 64, 8, 128, 150
+    pages_per = (slen + 63) // 64
+    npages = B * pages_per + 1
+    kc = torch.randn(npages, hkv, 64, dh, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, hkv, 64, dh, dtype=torch.bfloat16, device="cuda")
+    perm = torch.randperm(B * pages_per, device="cuda").to(torch.int32)
+    pt = perm.reshape(B, pages_per).contiguous()
+    sl = torch.full((B,), slen, dtype=torch.int32, device="cuda")
+    q = torch.randn(B, hq, dh, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    ops.attn_decode(out, q, kc, vc, pt, sl, 0.0883)
+    # torch fp32 reference (same math as the CPU fallback)
+    ref = torch.empty_like(out)
+    G = hq // hkv
+    for b in range(B):
+        pages = pt[b, :pages_per].long()
+        kk = kc[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)[:slen].float()
+        vv = vc[pages].permute(0, 2, 1, 3).reshape(-1, hkv, dh)[:slen].float()
+        for h in range(hq):
+            g = h // G
+            s = kk[:, g] @ (q[b, h].float() * 0.0883)
+            p = torch.softmax(s, -1)
+            ref[b, h] = (p @ vv[:, g]).to(ref.dtype)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
