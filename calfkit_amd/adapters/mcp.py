@@ -1,20 +1,140 @@
-"""Sole importer of the official MCP SDK.
+"""MCP sessions: an SDK-free stdio JSON-RPC client + the SDK adapter.
 
 Capability parity: reference ``adapters/mcp.py``
-(/root/reference/src/calfkit/adapters/mcp.py:1): ``OfficialMCPSession``.
-The MCP SDK is not installed in this image (no network), so opening a real
-session raises a clear error; scripted sessions (``MCPToolbox(session=...)``
-with ``calfkit_amd.testing.FakeMCPSession``) cover the toolbox path.
+(/root/reference/src/calfkit/adapters/mcp.py:1): ``OfficialMCPSession``
+wraps the official SDK when it exists.  This image is offline (no SDK), so
+the stdio transport is implemented DIRECTLY: MCP's stdio framing is
+newline-delimited JSON-RPC 2.0 (initialize -> notifications/initialized ->
+tools/list / tools/call), which needs no SDK at all.  ``open_session``
+dials stdio natively and falls back to the SDK for sse/http transports.
 """
 
 from __future__ import annotations
 
+import asyncio
+import json
+import logging
 from typing import Any
 
 from ..definitions.toolbox import MCPToolbox
 from ..mcp import MCPResult, MCPSession, MCPToolInfo
 
-__all__ = ["OfficialMCPSession", "open_session"]
+__all__ = ["StdioMCPSession", "OfficialMCPSession", "open_session"]
+
+_log = logging.getLogger("calfkit.mcp")
+
+PROTOCOL_VERSION = "2024-11-05"
+
+
+class StdioMCPSession:
+    """A live MCP server over stdio, spoken directly (JSON-RPC 2.0,
+    newline-delimited frames per the MCP stdio transport spec)."""
+
+    def __init__(self, proc: asyncio.subprocess.Process):
+        self._p = proc
+        self._id = 0
+        self._pending: dict[int, asyncio.Future] = {}
+        self._pump = asyncio.get_running_loop().create_task(self._read_loop())
+
+    # ------------------------------------------------------------- transport
+    @classmethod
+    async def open(cls, command: str, args: tuple[str, ...] = (),
+                   timeout: float = 30.0) -> "StdioMCPSession":
+        proc = await asyncio.create_subprocess_exec(
+            command, *args,
+            stdin=asyncio.subprocess.PIPE, stdout=asyncio.subprocess.PIPE)
+        self = cls(proc)
+        await asyncio.wait_for(self._initialize(), timeout)
+        return self
+
+    async def _initialize(self) -> None:
+        result = await self._request("initialize", {
+            "protocolVersion": PROTOCOL_VERSION,
+            "capabilities": {},
+            "clientInfo": {"name": "calfkit-amd", "version": "0.2"},
+        })
+        self.server_info = result.get("serverInfo", {})
+        self._notify("notifications/initialized", {})
+
+    async def _read_loop(self) -> None:
+        assert self._p.stdout is not None
+        try:
+            while True:
+                line = await self._p.stdout.readline()
+                if not line:
+                    break
+                line = line.strip()
+                if not line:
+                    continue
+                try:
+                    msg = json.loads(line)
+                except json.JSONDecodeError:
+                    _log.warning("mcp: undecodable frame %r", line[:200])
+                    continue
+                mid = msg.get("id")
+                fut = self._pending.pop(mid, None) if mid is not None else None
+                if fut is not None and not fut.done():
+                    if "error" in msg:
+                        fut.set_exception(RuntimeError(
+                            f"mcp error {msg['error']}"))
+                    else:
+                        fut.set_result(msg.get("result", {}))
+        except asyncio.CancelledError:
+            pass
+        finally:
+            for fut in self._pending.values():
+                if not fut.done():
+                    fut.set_exception(ConnectionError("mcp server closed"))
+
+    def _send(self, obj: dict[str, Any]) -> None:
+        assert self._p.stdin is not None
+        self._p.stdin.write(json.dumps(obj, separators=(",", ":")).encode()
+                            + b"\n")
+
+    def _notify(self, method: str, params: dict[str, Any]) -> None:
+        self._send({"jsonrpc": "2.0", "method": method, "params": params})
+
+    async def _request(self, method: str, params: dict[str, Any],
+                       timeout: float = 60.0) -> dict[str, Any]:
+        self._id += 1
+        mid = self._id
+        fut: asyncio.Future = asyncio.get_running_loop().create_future()
+        self._pending[mid] = fut
+        self._send({"jsonrpc": "2.0", "id": mid, "method": method,
+                    "params": params})
+        assert self._p.stdin is not None
+        await self._p.stdin.drain()
+        return await asyncio.wait_for(fut, timeout)
+
+    # ------------------------------------------------------------------ port
+    async def list_tools(self) -> list[MCPToolInfo]:
+        result = await self._request("tools/list", {})
+        return [MCPToolInfo(name=t["name"],
+                            description=t.get("description", ""),
+                            input_schema=dict(t.get("inputSchema", {}) or {}))
+                for t in result.get("tools", [])]
+
+    async def call_tool(self, name: str, arguments: dict[str, Any]) -> MCPResult:
+        result = await self._request("tools/call",
+                                     {"name": name, "arguments": arguments})
+        parts = [c.get("text", str(c)) for c in result.get("content", [])]
+        return MCPResult(content="\n".join(parts),
+                         is_error=bool(result.get("isError", False)))
+
+    async def close(self) -> None:
+        self._pump.cancel()
+        if self._p.stdin is not None:
+            try:
+                self._p.stdin.close()
+            except Exception:
+                pass
+        try:
+            await asyncio.wait_for(self._p.wait(), 5)
+        except (asyncio.TimeoutError, ProcessLookupError):
+            try:
+                self._p.kill()
+            except ProcessLookupError:
+                pass
 
 
 class OfficialMCPSession:
@@ -42,12 +162,22 @@ class OfficialMCPSession:
 
 
 async def open_session(toolbox: MCPToolbox) -> MCPSession:
-    """Dial an MCP server per the toolbox config."""
+    """Dial an MCP server per the toolbox config.
+
+    stdio servers are dialed natively (no SDK needed); sse/http need the
+    official SDK's transports and raise a clear error when it is absent.
+    """
+    if toolbox.transport == "stdio":
+        if not toolbox.command:
+            raise ValueError(
+                f"toolbox {toolbox.name!r}: stdio transport needs command=")
+        return await StdioMCPSession.open(toolbox.command, toolbox.args)
     try:
         import mcp  # noqa: F401
     except ImportError as e:
         raise RuntimeError(
-            f"toolbox {toolbox.name!r}: the official MCP SDK is not installed; "
+            f"toolbox {toolbox.name!r}: transport {toolbox.transport!r} needs "
+            "the official MCP SDK (not installed); use a stdio server or "
             "pass a pre-built session via MCPToolbox(session=...)") from e
     raise NotImplementedError(
-        "official MCP transport dialing lands with the SDK available")
+        "official MCP sse/http dialing lands with the SDK available")
