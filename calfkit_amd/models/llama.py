@@ -212,6 +212,8 @@ class LlamaShard:
                                   device=x.device)
                 ops.tile_gemm(out, x, weight, self._gemm_work, *plan)
                 return out
+            # library path with the SEARCHED algo index where one exists
+            return ops.linear_indexed(x, weight)
         return torch.nn.functional.linear(x, weight)
 
     # ------------------------------------------------------------------ util

@@ -33,6 +33,7 @@ __all__ = [
     "tile_gemm",
     "tile_gemm_slabs",
     "tile_plan",
+    "linear_indexed",
     "grammar_bias",
     "grammar_transition",
     "rope_qkv_write_slab",
@@ -368,6 +369,60 @@ def grammar_bias(logits: torch.Tensor, prev: torch.Tensor,
             row[G_TEND] += BIG
         else:
             row[:16] -= BIG
+
+
+# ----------------------------------------------------- searched-algo GEMM
+
+_ALGO_TABLE: "dict | None" = None
+_ALGO_WS: dict = {}
+
+
+def _algo_table() -> dict:
+    global _ALGO_TABLE
+    if _ALGO_TABLE is None:
+        import json
+        import os
+
+        path = os.path.join(os.path.dirname(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__)))), "profiles", "gemm_algo_table.json")
+        table = {}
+        try:
+            with open(path) as f:
+                for v in json.load(f).values():
+                    if v.get("numerics_ok") and v.get("speedup", 0) > 1.0:
+                        table[(v["M"], v["N"], v["K"])] = v["algo_index"]
+        except Exception:
+            pass
+        if os.environ.get("CALF_GEMM_INDEXED", "1") != "1":
+            table = {}
+        _ALGO_TABLE = table
+    return _ALGO_TABLE
+
+
+def linear_indexed(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """hipBLASLt with the SEARCHED algorithm index for known serving
+    shapes (profiles/gemm_algo_table.json: best-vs-heuristic 1.03-1.11x);
+    the heuristic path otherwise.  Wired on the DECODE path only — the
+    per-call setup host cost is paid once at hipGraph capture and never
+    at replay."""
+    if _K is None or not x.is_cuda:
+        return torch.nn.functional.linear(x, w)
+    key = (x.shape[0], w.shape[0], w.shape[1])
+    idx = _algo_table().get(key)
+    if idx is None:
+        return torch.nn.functional.linear(x, w)
+    out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
+    ws = _ALGO_WS.get(x.device.index)
+    if ws is None:
+        ws = torch.empty(64 << 20, dtype=torch.uint8, device=x.device)
+        _ALGO_WS[x.device.index] = ws
+    try:
+        _K.hipblaslt_gemm_indexed(out, x, w, idx, ws)
+    except RuntimeError:
+        # stale index (library version drift): drop it and fall back
+        _algo_table().pop(key, None)
+        return torch.nn.functional.linear(x, w)
+    return out
 
 
 # -------------------------------------------------------------- skinny GEMM
