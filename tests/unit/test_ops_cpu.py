@@ -121,3 +121,17 @@ def test_grammar_transition_state_machine():
     assert s & 0xFF == 2 + ops.G_ARG_TOKS
     s = ops.grammar_transition(s, ops.G_TEND)    # closed: free, calls=1
     assert s & 0xFF == 0 and s >> 8 == 1
+
+
+def test_linear_indexed_falls_back_off_gpu():
+    import torch
+
+    import calfkit_amd.ops as ops
+
+    x = torch.randn(4, 8, dtype=torch.float32)
+    w = torch.randn(6, 8, dtype=torch.float32)
+    out = ops.linear_indexed(x, w)
+    assert torch.allclose(out, x @ w.T, atol=1e-5)
+    # the algo table only admits numerics-ok entries with real speedups
+    table = ops._algo_table()
+    assert all(isinstance(k, tuple) and len(k) == 3 for k in table)
