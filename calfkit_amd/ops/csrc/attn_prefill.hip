@@ -50,7 +50,13 @@ __global__ __launch_bounds__(NW * WAVE) void attn_prefill_kernel(
     const int32_t* __restrict__ past_lens,  // [B] context rows per seq
     const int hq, const int hkv, const float scale,
     const int q_stride, const int kv_stride) {  // row strides in elements
-  const int head = blockIdx.x;
+  // XCD-grouped head order (guide T1): block b lands on XCD b%8 and
+  // blockIdx.x varies fastest, so the plain head=x order puts the G
+  // q-heads of one kv-head on G DIFFERENT XCDs — each re-streams the
+  // same K/V panel from HBM.  This bijection makes the heads of kv-head
+  // g share x%hkv == g, i.e. one XCD (hkv == 8 on llama): 3 of the 4
+  // panel reads become that XCD's L2 hits.
+  const int head = (blockIdx.x % hkv) * (hq / hkv) + blockIdx.x / hkv;
   const int kvh = head / (hq / hkv);
   const int seq = block_map[blockIdx.y * 2];
   const int qb = block_map[blockIdx.y * 2 + 1];
