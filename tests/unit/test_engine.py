@@ -380,3 +380,43 @@ def test_tool_call_parser_edges():
     parts2, calls2 = _parse_output([tok.TCALL, 499, 1, 2, tok.TEND, 70],
                                    (schema,), tok)
     assert not calls2
+
+
+def test_engine_fuzz_random_request_mix_is_deterministic():
+    """Scheduler soak: a randomized mix of requests (lengths, budgets,
+    temperatures, seeds, tool plans, eos behavior) all complete with the
+    right token counts, and the same mix replays identically (the engine
+    is deterministic given request seeds — no hidden RNG, no
+    batch-composition leakage)."""
+    import random
+
+    def run_mix() -> list[list[int]]:
+        rng = random.Random(1234)
+        eng = make_engine(max_batch=8, max_model_len=128)
+        futs = []
+        try:
+            for i in range(24):
+                plan = rng.random() < 0.3
+                req = GenRequest(
+                    prompt_ids=[1] + [rng.randrange(16, 500)
+                                      for _ in range(rng.randrange(2, 20))],
+                    max_new_tokens=rng.randrange(1, 12),
+                    temperature=rng.choice([0.0, 0.7, 1.2]),
+                    seed=100 + i,
+                    ignore_eos=not plan,
+                    tool_names=[100, 200] if plan else None,
+                    tool_trigger=rng.randrange(0, 3),
+                    tool_ncalls=1)
+                futs.append((req, eng.submit(req)))
+            outs = []
+            for req, f in futs:
+                r = f.result(180)
+                assert len(r.token_ids) <= req.max_new_tokens
+                if req.ignore_eos:
+                    assert len(r.token_ids) == req.max_new_tokens
+                outs.append(r.token_ids)
+            return outs
+        finally:
+            eng.stop()
+
+    assert run_mix() == run_mix()
