@@ -18,6 +18,11 @@
 // The kernel is memory-bound by construction (~16 B of KV per lane-op);
 // the design targets the HBM3E stream, not MFMA (decode reads each KV byte
 // exactly once — matrix cores have nothing to multiply twice).
+//
+// r02 occupancy A/B: forcing 5 waves/SIMD via __launch_bounds__(.., 5)
+// squeezes 102 -> 96 VGPR at 28 B/lane scratch spill — same-box bench
+// 241.8 -> 231.8/235.2 and the shuffled-page micro 122.8 -> 177.4 us.
+// The spill costs more than the extra wave hides; stay at 4 waves.
 
 #include "common.h"
 
