@@ -251,6 +251,9 @@ def _materialize_value(schema: dict, toks: list[int], cur: list[int]):
         cur[0] += 1
         return v
 
+    if (schema or {}).get("enum"):
+        e = schema["enum"]
+        return e[take() % len(e)]
     t = (schema or {}).get("type", "object")
     if t == "object":
         return {k: _materialize_value(v or {}, toks, cur)
@@ -277,7 +280,14 @@ def _materialize_args(schema: Any, toks: list[int]) -> dict:
     args: dict[str, Any] = {}
     for i, key in enumerate(sorted(props)):
         tokv = toks[i % len(toks)] if toks else 17
-        ty = (props[key] or {}).get("type", "string")
+        spec = props[key] or {}
+        if spec.get("enum"):
+            # constrained fields (e.g. a capability tool's peer name):
+            # the sampled token picks among the VALID values, so
+            # engine-sampled handoffs / peer messages actually route
+            args[key] = spec["enum"][tokv % len(spec["enum"])]
+            continue
+        ty = spec.get("type", "string")
         if ty == "integer":
             args[key] = int(tokv)
         elif ty == "number":

@@ -137,3 +137,31 @@ async def test_structured_output_from_the_engine():
         assert isinstance(result.output, Weather)
         assert isinstance(result.output.temp_c, int)
         assert result.output.city.startswith("t")
+
+
+async def test_engine_sampled_handoff_routes_to_a_real_peer():
+    """The GPU/engine grammar can drive the HANDOFF capability: 'handoff'
+    is advertised like any tool, the sampled token picks the peer from the
+    schema's enum (so it always names a REAL agent), and the conversation
+    transfers across nodes — no FakeProvider anywhere on the front agent."""
+    from calfkit_amd.definitions.selections import Handoff
+    from calfkit_amd.testing import FakeProvider
+
+    bus, store, directory = (InMemoryBus(), InProcessThreadStore(),
+                             InMemoryDirectory())
+    front = Agent(name="front",
+                  model_client=LocalModelClient(TINY, device="cpu",
+                                                max_batch=4,
+                                                max_model_len=256,
+                                                use_graphs=False),
+                  peers=[Handoff(names=("specialist",))],
+                  model_settings=ModelSettings(max_tokens=16,
+                                               temperature=0.7, seed=21))
+    specialist = Agent(name="specialist",
+                       model_client=FakeProvider(["specialist answer"]))
+    async with Worker(front, specialist, bus=bus, store=store,
+                      directory=directory):
+        client = await Client.connect(bus, directory)
+        result = await client.execute("front", "hard question", timeout=300)
+        # the engine sampled the handoff; the specialist answered the thread
+        assert result.text == "specialist answer"
