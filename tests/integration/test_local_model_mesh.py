@@ -139,7 +139,7 @@ async def test_structured_output_from_the_engine():
         assert result.output.city.startswith("t")
 
 
-async def test_engine_sampled_handoff_routes_to_a_real_peer():
+async def _handoff_round(device: str, use_graphs: bool):
     """The GPU/engine grammar can drive the HANDOFF capability: 'handoff'
     is advertised like any tool, the sampled token picks the peer from the
     schema's enum (so it always names a REAL agent), and the conversation
@@ -150,10 +150,10 @@ async def test_engine_sampled_handoff_routes_to_a_real_peer():
     bus, store, directory = (InMemoryBus(), InProcessThreadStore(),
                              InMemoryDirectory())
     front = Agent(name="front",
-                  model_client=LocalModelClient(TINY, device="cpu",
+                  model_client=LocalModelClient(TINY, device=device,
                                                 max_batch=4,
                                                 max_model_len=256,
-                                                use_graphs=False),
+                                                use_graphs=use_graphs),
                   peers=[Handoff(names=("specialist",))],
                   model_settings=ModelSettings(max_tokens=16,
                                                temperature=0.7, seed=21))
@@ -165,3 +165,12 @@ async def test_engine_sampled_handoff_routes_to_a_real_peer():
         result = await client.execute("front", "hard question", timeout=300)
         # the engine sampled the handoff; the specialist answered the thread
         assert result.text == "specialist answer"
+
+
+async def test_engine_sampled_handoff_routes_to_a_real_peer():
+    await _handoff_round("cpu", False)
+
+
+@pytest.mark.gpu
+async def test_engine_sampled_handoff_routes_gpu():
+    await _handoff_round("cuda", True)
