@@ -161,9 +161,21 @@ class _PendingPrefill:
     t0: float
 
 
+def _graph_batches() -> tuple:
+    base = [1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256, 384, 512, 768, 1024]
+    extra = _os.environ.get("CALF_EXTRA_GRAPH_BATCHES", "")
+    for tok in extra.split(","):
+        tok = tok.strip()
+        if tok:
+            base.append(int(tok))
+    return tuple(sorted(set(base)))
+
+
 class InferenceEngine:
     #: padded decode batch sizes that get their own captured graph
-    GRAPH_BATCHES = (1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256, 384, 512, 768, 1024)
+    #: (CALF_EXTRA_GRAPH_BATCHES adds sizes, e.g. "448,480" for finer
+    #: padding near the serving point)
+    GRAPH_BATCHES = _graph_batches()
 
     def __init__(self, model: LlamaShard, *, max_batch: int = 64,
                  kv_pages: int | None = None, use_graphs: bool | None = None,
