@@ -222,7 +222,7 @@ class LocalModelClient:
                                        result.token_ids or [17], [0])
             parts = (TextPart(_json.dumps({"final_result": value})),)
             finish = "stop"
-        if not calls and s.stop:
+        if not calls and s.stop and request.output is None:
             # stop-sequence truncation (client side), text-only turns
             text = " ".join(p.text for p in parts
                             if isinstance(p, TextPart))
