@@ -31,6 +31,7 @@ __all__ = [
     "SideloadedCapability",
     "sideloads",
     "SIDELOAD_REGISTRY",
+    "HandoffRequest",
     "MessagingCapability",
     "HandoffCapability",
 ]
@@ -70,6 +71,26 @@ class Reject:
 
 
 Effect = Union[Dispatch, Transfer, Reject]
+
+
+@dataclass(frozen=True)
+class HandoffRequest:
+    """Typed argument model of the handoff tool — the field descriptions ARE
+    the model-facing schema (reference /root/reference/CHANGELOG.md:157,
+    "HandoffRequest field descriptions")."""
+
+    to: str
+    reason: str = ""
+
+    FIELD_DESCRIPTIONS = {
+        "to": "agent to hand off to",
+        "reason": "why the conversation is being handed off",
+    }
+
+    @classmethod
+    def from_arguments(cls, arguments: dict) -> "HandoffRequest":
+        return cls(to=str(arguments.get("to", "")),
+                   reason=str(arguments.get("reason", "")))
 
 
 # ----------------------------------------------------------- capabilities
@@ -178,9 +199,9 @@ class HandoffCapability(SideloadedCapability):
                 "type": "object",
                 "properties": {
                     "to": {"type": "string", "enum": sorted(peers),
-                            "description": "agent to hand off to"},
+                            "description": HandoffRequest.FIELD_DESCRIPTIONS["to"]},
                     "reason": {"type": "string",
-                                "description": "why the conversation is being handed off"},
+                                "description": HandoffRequest.FIELD_DESCRIPTIONS["reason"]},
                 },
                 "required": ["to"],
             })
@@ -190,11 +211,10 @@ class HandoffCapability(SideloadedCapability):
                 source_agent: str) -> Effect | None:
         if call.tool_name != self.TOOL:
             return None
-        target = str(call.arguments.get("to", ""))
-        advert = peers.get(target)
+        req = HandoffRequest.from_arguments(call.arguments)
+        advert = peers.get(req.to)
         if advert is None:
             return Reject(call_id=call.call_id, tool_name=call.tool_name,
-                          reason=f"unknown agent {target!r}; known: {sorted(peers)}")
-        return Transfer(target=target, target_inbox=advert.inbox,
-                        reason=str(call.arguments.get("reason", "")),
-                        call_id=call.call_id)
+                          reason=f"unknown agent {req.to!r}; known: {sorted(peers)}")
+        return Transfer(target=req.to, target_inbox=advert.inbox,
+                        reason=req.reason, call_id=call.call_id)

@@ -109,3 +109,23 @@ async def test_unknown_peer_rejected_surfaced():
     assert res.finished
     tool_msgs = [m for m in res.new_messages if m.role == "tool"]
     assert "ghost" in str(tool_msgs[0].parts[0].content)
+
+
+def test_handoff_request_is_the_typed_argument_model():
+    """HandoffRequest's field descriptions ARE the handoff tool schema
+    (reference CHANGELOG.md:157)."""
+    from calfkit_amd._capabilities import HandoffCapability, HandoffRequest, Transfer
+    from calfkit_amd.messages import ToolCallPart
+    from calfkit_amd.ports import Advert
+
+    peers = {"expert": Advert(node="expert", kind="agent", inbox="in.expert",
+                              description="knows things")}
+    cap = HandoffCapability()
+    schema = cap.contribute(peers)[0].schema
+    props = schema.parameters["properties"]
+    assert props["to"]["description"] == HandoffRequest.FIELD_DESCRIPTIONS["to"]
+    assert props["reason"]["description"] == HandoffRequest.FIELD_DESCRIPTIONS["reason"]
+    eff = cap.realize(ToolCallPart(call_id="c1", tool_name="handoff",
+                                   arguments={"to": "expert", "reason": "r"}),
+                      peers, "me")
+    assert isinstance(eff, Transfer) and eff.target == "expert" and eff.reason == "r"
