@@ -118,9 +118,20 @@ def main() -> None:
                 continue
             if best is None or t < best[2]:
                 best = (bn, sk, t)
+        t_small = None
+        if N % 128 == 0 and K % 64 == 0:
+            ops.tile_gemm_small(out, x, w)
+            err_s = (out.float() - x.float() @ w.float().T).abs().max().item()
+            if err_s <= 0.1:
+                t_small = bench(lambda: ops.tile_gemm_small(out, x, w),
+                                args.iters)
+            else:
+                results["small_ERR"] = err_s
         line = {
             "shape": name, "M": M, "K": K, "N": N,
             "lib_us": round(t_lib, 1),
+            "small_us": round(t_small, 1) if t_small else None,
+            "small_vs_lib": round(t_lib / t_small, 3) if t_small else None,
             "lib_indexed_us": round(t_idx, 1) if t_idx else None,
             "tile_us": round(best[2], 1) if best else None,
             "tile_plan": f"bn{best[0]}_sk{best[1]}" if best else None,

@@ -488,6 +488,16 @@ def tile_gemm(out: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     out.copy_((x.float() @ w.float().T).to(out.dtype))
 
 
+def tile_gemm_small(out: torch.Tensor, x: torch.Tensor,
+                    w: torch.Tensor) -> None:
+    """out[M,N] = x[M,K] @ w[N,K]^T via 64x128 tiles (natural fill, no
+    split-K) — the small-tile variant for the M=384/512 projections."""
+    if _native(x):
+        _K.tile_gemm_small(out, x, w)
+        return
+    out.copy_((x.float() @ w.float().T).to(out.dtype))
+
+
 def tile_gemm_slabs(x: torch.Tensor, w: torch.Tensor, work: torch.Tensor,
                     bn: int, splitk: int) -> int:
     """Slab-only tiled GEMM; returns Mp. A fused consumer reduces."""

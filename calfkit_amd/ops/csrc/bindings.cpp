@@ -56,6 +56,7 @@ void grammar_bias(torch::Tensor logits, torch::Tensor prev,
                   torch::Tensor gstate, torch::Tensor trigger,
                   torch::Tensor names, torch::Tensor pos,
                   torch::Tensor ncalls);
+void tile_gemm_small(torch::Tensor out, torch::Tensor x, torch::Tensor w);
 void tile_gemm_slabs(torch::Tensor x, torch::Tensor w, torch::Tensor work,
                      int64_t bn, int64_t splitk);
 
@@ -93,6 +94,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "run out = x @ w^T with a searched hipBLASLt algo index");
   m.def("tile_gemm", &tile_gemm,
         "tiled MFMA GEMM for serving batches (M 65..512+), split-K reduce");
+  m.def("tile_gemm_small", &tile_gemm_small,
+        "64x128-tile MFMA GEMM, natural fill for M=384/512 projections");
   m.def("tile_gemm_slabs", &tile_gemm_slabs,
         "tiled MFMA GEMM, fp32 slabs only (consumer fuses the reduce)");
   m.def("grammar_bias", &grammar_bias,

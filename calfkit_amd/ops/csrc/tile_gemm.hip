@@ -396,6 +396,155 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Small-tile variant: 64xBN tiles, 4 waves, 3 blocks/CU.
+//
+// Why: at M=512 the 256-wide kernel has only 2 m-tiles, so qkv/o land 48-96
+// natural tiles on a 256-CU chip and need split-K (and its fp32 slab tax) to
+// fill it; the library's winning macro-tiles on exactly these shapes are
+// MT96x128/MT128x64 (profiles/r02_bench_kernelstats.md).  BM=64 gives 8
+// m-tiles -> 256-384 natural tiles with NO split-K, and the latency-hiding
+// burden moves from intra-block pipelining (8-phase counted-wait ring) to
+// occupancy: 3 blocks/CU x 4 waves = 12 waves/CU, so one block's post-stage
+// drain parks under another block's MFMAs.  B panels are re-read M/64 times,
+// but consecutive logical blocks share an n-panel and the XCD-bijective remap
+// keeps them on one XCD, so re-reads hit that XCD's 4 MB L2 (panel = 1 MB at
+// BN=128, K-slice resident) instead of HBM.
+template <int BN>
+__global__ __launch_bounds__(256, 3) void tile_gemm_small_kernel(
+    __hip_bfloat16* __restrict__ out,     // [M, N]
+    const __hip_bfloat16* __restrict__ x, // [M, K]
+    const __hip_bfloat16* __restrict__ w, // [N, K]
+    const int M, const int N, const int K, const int m_tiles) {
+  constexpr int BM = 64, BK = 64;
+  constexpr int WC = 4;             // waves along N
+  constexpr int NI = BN / 16 / WC;  // n-frags per wave (2 at BN=128)
+  constexpr int MI = BM / 16;       // m-frags per wave (4); WR == 1
+  constexpr int A_ELEMS = BM * BK;
+  constexpr int B_ELEMS = BN * BK;
+  constexpr int BUF = A_ELEMS + B_ELEMS;
+  __shared__ __attribute__((aligned(16))) __hip_bfloat16 smem[2 * BUF];
+
+  int bid = blockIdx.x;
+  {  // XCD-bijective remap (T1): contiguous logical range per XCD
+    const int nwg = gridDim.x, q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, i = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  // m inner: consecutive logical blocks share the n-panel (L2 reuse)
+  const int n_t = bid / m_tiles, m_t = bid % m_tiles;
+  const int n0 = n_t * BN, m0 = m_t * BM;
+  const int nt = K / BK;
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE, lane = tid % WAVE;
+  const int l16 = lane & 15, l4 = lane >> 4;
+  const int wc = wid;  // WR == 1: every wave owns all 64 rows, 32 cols
+  const int sr = tid / 8;  // staging: 32 image rows per glds call
+  const int sc = tid % 8;
+
+  const auto stage_a = [&](int buf, int ktile) {
+    const int kk = ktile * BK;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      const int rloc = c * 32 + sr;
+      const int grow = min(m0 + rloc, M - 1);
+      const int chunk = sc ^ (rloc & 7);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int*>(
+              x + (int64_t)grow * K + kk + chunk * 8),
+          reinterpret_cast<unsigned int*>(
+              &smem[buf * BUF + (c * 32 + wid * 8) * BK]),
+          16, 0, 0);
+    }
+  };
+  const auto stage_b = [&](int buf, int ktile) {
+    const int kk = ktile * BK;
+#pragma unroll
+    for (int c = 0; c < BN / 32; ++c) {
+      const int rloc = c * 32 + sr;
+      const int chunk = sc ^ (rloc & 7);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int*>(
+              w + (int64_t)(n0 + rloc) * K + kk + chunk * 8),
+          reinterpret_cast<unsigned int*>(
+              &smem[buf * BUF + A_ELEMS + (c * 32 + wid * 8) * BK]),
+          16, 0, 0);
+    }
+  };
+  const auto read_a = [&](int buf, int mi, int ks) {
+    const int row = mi * 16 + l16;
+    const int chunk = (ks * 4 + l4) ^ (row & 7);
+    return *reinterpret_cast<const bf16x8v*>(
+        &smem[buf * BUF + row * BK + chunk * 8]);
+  };
+  const auto read_b = [&](int buf, int ni, int ks) {
+    const int row = ni * (16 * WC) + wc * 16 + l16;
+    const int chunk = (ks * 4 + l4) ^ (row & 7);
+    return *reinterpret_cast<const bf16x8v*>(
+        &smem[buf * BUF + A_ELEMS + row * BK + chunk * 8]);
+  };
+
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  stage_a(0, 0);
+  stage_b(0, 0);
+  VMWAIT(0);
+  __builtin_amdgcn_s_barrier();
+
+  int cur = 0;
+  for (int t = 0; t < nt; ++t) {
+    const bool pf = t + 1 < nt;
+    const int nb = cur ^ 1;
+    bf16x8v afr[MI][2], bfr[NI][2];
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) afr[mi][ks] = read_a(cur, mi, ks);
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) bfr[ni][ks] = read_b(cur, ni, ks);
+    // next tile's stages issue behind this tile's MFMAs; the tail
+    // VMWAIT(0)+barrier land them before the next head's ds_reads.
+    // nb was last READ at t-1, and every wave passed the t-1 tail
+    // barrier after those reads, so the overwrite is ordered.
+    if (pf) {
+      stage_a(nb, t + 1);
+      stage_b(nb, t + 1);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          acc[mi][ni] = MFMA16(afr[mi][ks], bfr[ni][ks], acc[mi][ni]);
+    if (pf) {
+      VMWAIT(0);
+      __builtin_amdgcn_s_barrier();
+    }
+    cur ^= 1;
+  }
+
+  // epilogue (mfma_f32_16x16x32 C layout: col = l16, row = l4*4 + r)
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + mi * 16 + l4 * 4 + r;
+        if (row < M)
+          out[(int64_t)row * N + n0 + ni * (16 * WC) + wc * 16 + l16] =
+              f2bf(acc[mi][ni][r]);
+      }
+}
+
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include <cstdlib>
@@ -469,6 +618,24 @@ void tile_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                        (__hip_bfloat16*)out.data_ptr(),
                        (const float*)work.data_ptr(), M, N, Mp, (int)splitk);
   }
+  HIP_CHECK_LAST();
+}
+
+void tile_gemm_small(torch::Tensor out, torch::Tensor x, torch::Tensor w) {
+  // out[M,N] = x[M,K] @ w[N,K]^T via 64x128 tiles — the natural-fill
+  // variant for the M=384/512 projections (no split-K, direct bf16 out)
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(N % 128 == 0, "tile_gemm_small: N % 128 != 0");
+  TORCH_CHECK(K % 64 == 0, "tile_gemm_small: K % 64 != 0");
+  const int m_tiles = (M + 63) / 64;
+  const int blocks = (N / 128) * m_tiles;
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL((tile_gemm_small_kernel<128>), dim3(blocks), dim3(256),
+                     0, s, (__hip_bfloat16*)out.data_ptr(),
+                     (const __hip_bfloat16*)x.data_ptr(),
+                     (const __hip_bfloat16*)w.data_ptr(), M, N, K, m_tiles);
   HIP_CHECK_LAST();
 }
 

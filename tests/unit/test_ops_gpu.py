@@ -502,6 +502,22 @@ def test_tile_gemm_vs_fp32(M, K, N, bn, sk):
         (out.float() - ref).abs().max()
 
 
+@pytest.mark.parametrize("M,K,N", [
+    (512, 4096, 6144),   # qkv serving shape (the kernel's reason to exist)
+    (384, 512, 128),     # single n-tile, M multiple of 64
+    (100, 768, 256),     # ragged M tail (row clamp + store guard)
+    (64, 64, 128),       # minimum: one tile, one K-iteration
+])
+def test_tile_gemm_small_vs_fp32(M, K, N):
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / math.sqrt(K)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    ops.tile_gemm_small(out, x, w)
+    ref = x.float() @ w.float().T
+    assert torch.allclose(out.float(), ref, atol=8e-2, rtol=8e-2), \
+        (out.float() - ref).abs().max()
+
+
 def test_tile_gemm_slabs_match_reduced():
     """Slab layout contract: work[sk][Mp][N] fp32, summed == product (the
     fused consumers rope_qkv_write_slab/fused_add_rmsnorm_slab read it)."""
