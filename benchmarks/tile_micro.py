@@ -31,6 +31,8 @@ SHAPES = [  # (name, M, K, N)
     ("o_8b_m384", 384, 4096, 4096),
     ("gateup_8b_m384", 384, 4096, 28672),
     ("down_8b_m384", 384, 14336, 4096),
+    ("o_8b_tp8_m512", 512, 512, 4096),
+    ("down_8b_tp8_m512", 512, 1792, 4096),
     ("qkv_70b_tp8_m512", 512, 8192, 1280),
     ("o_70b_tp8_m512", 512, 1024, 8192),
     ("gateup_70b_tp8_m512", 512, 8192, 7168),

@@ -189,6 +189,18 @@ class LlamaShard:
                   else ["lmhead", "down"])
         return plan if cls in routed else None
 
+    def _small_ok(self, T: int, n_out: int, k_in: int) -> bool:
+        """Route the 64x128 small-tile kernel: the shallow-K wide-N class
+        (TP-sharded o projections, K = d/tp <= 1024) where it measured 1.58x vs
+        hipBLASLt (ops.small_plan / profiles/r02_tile_gemm.md)."""
+        import os
+
+        if self.device.type != "cuda" or not (64 < T <= 512):
+            return False
+        if os.environ.get("CALF_TILED", "1") == "0":
+            return False
+        return ops.small_plan(T, n_out, k_in)
+
     def _splitk_for(self, weight: torch.Tensor, m: int = 32) -> int:
         key = (weight.shape[0], m > 64)
         sk = self._splitk_cache.get(key)
@@ -206,6 +218,11 @@ class LlamaShard:
             ops.skinny_gemm(out, x, weight, self._gemm_work, sk)
             return out
         if decode:
+            if self._small_ok(x.shape[0], n_out, weight.shape[1]):
+                out = torch.empty(x.shape[0], n_out, dtype=x.dtype,
+                                  device=x.device)
+                ops.tile_gemm_small(out, x, weight)
+                return out
             plan = self._tile_plan(x.shape[0], n_out, weight.shape[1])
             if plan is not None:
                 out = torch.empty(x.shape[0], n_out, dtype=x.dtype,

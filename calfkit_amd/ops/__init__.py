@@ -33,6 +33,7 @@ __all__ = [
     "tile_gemm",
     "tile_gemm_slabs",
     "tile_plan",
+    "small_plan",
     "linear_indexed",
     "grammar_bias",
     "grammar_transition",
@@ -444,6 +445,20 @@ def choose_splitk(N: int, K: int, m: int = 32) -> int:
         if (N // 64) * sk <= target:
             best = sk
     return best
+
+
+def small_plan(M: int, N: int, K: int) -> bool:
+    """True when the 64x128 small-tile kernel is the measured winner.
+
+    r02 tile_micro: shallow-K wide-N shapes (the TP-sharded o/down
+    projections, K = d/tp <= 1024) land 256+ natural 64x128 tiles and
+    only ~16-32 K-iterations, where the small kernel beat hipBLASLt
+    1.58x (o_70b_tp8: 12.2 vs 19.3 us); deep-K shapes (K >= 4096) lose
+    to the per-tile drain and stay on the big kernel / library.
+    """
+    if N % 128 or K % 64 or K > 1024 or N < 4096:
+        return False
+    return (N // 128) * ((M + 63) // 64) >= 256
 
 
 def tile_plan(M: int, N: int, K: int) -> "tuple[int, int] | None":

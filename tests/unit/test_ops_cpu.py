@@ -135,3 +135,15 @@ def test_linear_indexed_falls_back_off_gpu():
     # the algo table only admits numerics-ok entries with real speedups
     table = ops._algo_table()
     assert all(isinstance(k, tuple) and len(k) == 3 for k in table)
+
+
+def test_small_plan_routes_only_the_measured_class():
+    """Shallow-K wide-N (TP-sharded o projections) routes to the 64x128
+    kernel; deep-K, narrow-N and skinny batches stay off it
+    (profiles/r02_tile_gemm.md small-kernel sweep)."""
+    assert ops.small_plan(512, 8192, 1024)       # o_70b_tp8: 1.58x vs lib
+    assert ops.small_plan(512, 4096, 512)        # o_8b_tp8
+    assert not ops.small_plan(512, 4096, 1792)   # mid-K: unmeasured
+    assert not ops.small_plan(512, 6144, 4096)   # qkv deep-K: big kernel's
+    assert not ops.small_plan(64, 8192, 1024)    # skinny: skinny_gemm's
+    assert not ops.small_plan(512, 1000, 512)    # N % 128
