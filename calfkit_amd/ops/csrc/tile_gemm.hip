@@ -410,7 +410,10 @@ __global__ __launch_bounds__(512, 2) void tile_gemm_kernel(
 // but consecutive logical blocks share an n-panel and the XCD-bijective remap
 // keeps them on one XCD, so re-reads hit that XCD's 4 MB L2 (panel = 1 MB at
 // BN=128, K-slice resident) instead of HBM.
-template <int BN>
+// BUFS=2: per-tile drain (VMWAIT 0), 3 blocks/CU hide it across blocks.
+// BUFS=3: ring stages t+2, counted vmcnt(6) tail wait — no drain, but
+// 72 KiB LDS caps occupancy at 2 blocks/CU.  CALF_SMALL_BUFS A/Bs them.
+template <int BN, int BUFS>
 __global__ __launch_bounds__(256, 3) void tile_gemm_small_kernel(
     __hip_bfloat16* __restrict__ out,     // [M, N]
     const __hip_bfloat16* __restrict__ x, // [M, K]
@@ -423,7 +426,7 @@ __global__ __launch_bounds__(256, 3) void tile_gemm_small_kernel(
   constexpr int A_ELEMS = BM * BK;
   constexpr int B_ELEMS = BN * BK;
   constexpr int BUF = A_ELEMS + B_ELEMS;
-  __shared__ __attribute__((aligned(16))) __hip_bfloat16 smem[2 * BUF];
+  __shared__ __attribute__((aligned(16))) __hip_bfloat16 smem[BUFS * BUF];
 
   int bid = blockIdx.x;
   {  // XCD-bijective remap (T1): contiguous logical range per XCD
@@ -493,13 +496,21 @@ __global__ __launch_bounds__(256, 3) void tile_gemm_small_kernel(
 
   stage_a(0, 0);
   stage_b(0, 0);
-  VMWAIT(0);
+  if (BUFS == 3 && nt > 1) {
+    stage_a(1, 1);
+    stage_b(1, 1);
+    VMWAIT(6);
+  } else {
+    VMWAIT(0);
+  }
   __builtin_amdgcn_s_barrier();
 
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
     const bool pf = t + 1 < nt;
-    const int nb = cur ^ 1;
+    const int st = BUFS == 3 ? t + 2 : t + 1;
+    const bool pf2 = st < nt;
+    const int nb = BUFS == 3 ? (t + 2) % 3 : (cur ^ 1);
     bf16x8v afr[MI][2], bfr[NI][2];
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi)
@@ -509,13 +520,13 @@ __global__ __launch_bounds__(256, 3) void tile_gemm_small_kernel(
     for (int ni = 0; ni < NI; ++ni)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) bfr[ni][ks] = read_b(cur, ni, ks);
-    // next tile's stages issue behind this tile's MFMAs; the tail
-    // VMWAIT(0)+barrier land them before the next head's ds_reads.
-    // nb was last READ at t-1, and every wave passed the t-1 tail
-    // barrier after those reads, so the overwrite is ordered.
-    if (pf) {
-      stage_a(nb, t + 1);
-      stage_b(nb, t + 1);
+    // next stages issue behind this tile's MFMAs; the tail wait +
+    // barrier land tile t+1 before the next head's ds_reads.  nb was
+    // last READ at t-1 (BUFS=2) / t-1's ring slot (BUFS=3), and every
+    // wave passed the t-1 tail barrier after those reads.
+    if (pf2) {
+      stage_a(nb, st);
+      stage_b(nb, st);
     }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks)
@@ -525,10 +536,15 @@ __global__ __launch_bounds__(256, 3) void tile_gemm_small_kernel(
         for (int ni = 0; ni < NI; ++ni)
           acc[mi][ni] = MFMA16(afr[mi][ks], bfr[ni][ks], acc[mi][ni]);
     if (pf) {
-      VMWAIT(0);
+      // BUFS=3: t+1's 6 units landed when t+2's 6 are still in flight
+      if (BUFS == 3 && pf2) {
+        VMWAIT(6);
+      } else {
+        VMWAIT(0);
+      }
       __builtin_amdgcn_s_barrier();
     }
-    cur ^= 1;
+    cur = BUFS == 3 ? (t + 1) % 3 : (cur ^ 1);
   }
 
   // epilogue (mfma_f32_16x16x32 C layout: col = l16, row = l4*4 + r)
@@ -632,10 +648,19 @@ void tile_gemm_small(torch::Tensor out, torch::Tensor x, torch::Tensor w) {
   const int m_tiles = (M + 63) / 64;
   const int blocks = (N / 128) * m_tiles;
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL((tile_gemm_small_kernel<128>), dim3(blocks), dim3(256),
-                     0, s, (__hip_bfloat16*)out.data_ptr(),
-                     (const __hip_bfloat16*)x.data_ptr(),
-                     (const __hip_bfloat16*)w.data_ptr(), M, N, K, m_tiles);
+  const char* bv = getenv("CALF_SMALL_BUFS");
+  const int bufs = bv ? atoi(bv) : 2;  // r02 A/B: 2-buf + occupancy won
+  if (bufs == 3) {
+    hipLaunchKernelGGL((tile_gemm_small_kernel<128, 3>), dim3(blocks),
+                       dim3(256), 0, s, (__hip_bfloat16*)out.data_ptr(),
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (const __hip_bfloat16*)w.data_ptr(), M, N, K, m_tiles);
+  } else {
+    hipLaunchKernelGGL((tile_gemm_small_kernel<128, 2>), dim3(blocks),
+                       dim3(256), 0, s, (__hip_bfloat16*)out.data_ptr(),
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (const __hip_bfloat16*)w.data_ptr(), M, N, K, m_tiles);
+  }
   HIP_CHECK_LAST();
 }
 
