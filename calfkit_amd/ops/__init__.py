@@ -451,12 +451,14 @@ def small_plan(M: int, N: int, K: int) -> bool:
     """True when the 64x128 small-tile kernel is the measured winner.
 
     r02 tile_micro: shallow-K wide-N shapes (the TP-sharded o/down
-    projections, K = d/tp <= 1024) land 256+ natural 64x128 tiles and
-    only ~16-32 K-iterations, where the small kernel beat hipBLASLt
-    1.58x (o_70b_tp8: 12.2 vs 19.3 us); deep-K shapes (K >= 4096) lose
-    to the per-tile drain and stay on the big kernel / library.
+    projections, K = d/tp <= 2048) land 256+ natural 64x128 tiles with
+    few K-iterations, where the kernel beat hipBLASLt on every measured
+    shape — o_8b_tp8 2.77x (7.0 vs 19.4 us), o_70b_tp8 1.51x,
+    down_8b_tp8 1.32x (3-buf ring).  Deep-K shapes (K >= 4096) lose to
+    stage latency even with the ring (qkv 36.2 vs lib 32.5 us) and stay
+    on the big kernel / library.
     """
-    if N % 128 or K % 64 or K > 1024 or N < 4096:
+    if N % 128 or K % 64 or K > 2048 or N < 4096:
         return False
     return (N // 128) * ((M + 63) // 64) >= 256
 

@@ -649,7 +649,11 @@ void tile_gemm_small(torch::Tensor out, torch::Tensor x, torch::Tensor w) {
   const int blocks = (N / 128) * m_tiles;
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
   const char* bv = getenv("CALF_SMALL_BUFS");
-  const int bufs = bv ? atoi(bv) : 2;  // r02 A/B: 2-buf + occupancy won
+  // r02 A/B: shallow K (<=1024, 8-16 K-iters) wins with 2-buf + 3-block
+  // occupancy (o70b 12.8 vs 13.7 us); deeper K amortizes the ring's
+  // occupancy loss over more drain-free iterations (down8b K=1792:
+  // 14.5 vs 15.6; qkv K=4096: 36.2 vs 44.2)
+  const int bufs = bv ? atoi(bv) : (K > 1024 ? 3 : 2);
   if (bufs == 3) {
     hipLaunchKernelGGL((tile_gemm_small_kernel<128, 3>), dim3(blocks),
                        dim3(256), 0, s, (__hip_bfloat16*)out.data_ptr(),

@@ -141,9 +141,9 @@ def test_small_plan_routes_only_the_measured_class():
     """Shallow-K wide-N (TP-sharded o projections) routes to the 64x128
     kernel; deep-K, narrow-N and skinny batches stay off it
     (profiles/r02_tile_gemm.md small-kernel sweep)."""
-    assert ops.small_plan(512, 8192, 1024)       # o_70b_tp8: 1.58x vs lib
-    assert ops.small_plan(512, 4096, 512)        # o_8b_tp8
-    assert not ops.small_plan(512, 4096, 1792)   # mid-K: unmeasured
+    assert ops.small_plan(512, 8192, 1024)       # o_70b_tp8: 1.51x vs lib
+    assert ops.small_plan(512, 4096, 512)        # o_8b_tp8: 2.77x
+    assert ops.small_plan(512, 4096, 1792)       # down_8b_tp8: 1.32x (ring)
     assert not ops.small_plan(512, 6144, 4096)   # qkv deep-K: big kernel's
     assert not ops.small_plan(64, 8192, 1024)    # skinny: skinny_gemm's
     assert not ops.small_plan(512, 1000, 512)    # N % 128
