@@ -191,7 +191,7 @@ class LlamaShard:
 
     def _small_ok(self, T: int, n_out: int, k_in: int) -> bool:
         """Route the 64x128 small-tile kernel: the shallow-K wide-N class
-        (TP-sharded o/down projections, K = d/tp <= 2048) where it measured
+        (TP-sharded o/down projections, K = d/tp <= 1792) where it measured
         1.3-2.8x vs hipBLASLt (ops.small_plan / profiles/r02_tile_gemm.md)."""
         import os
 

@@ -451,14 +451,14 @@ def small_plan(M: int, N: int, K: int) -> bool:
     """True when the 64x128 small-tile kernel is the measured winner.
 
     r02 tile_micro: shallow-K wide-N shapes (the TP-sharded o/down
-    projections, K = d/tp <= 2048) land 256+ natural 64x128 tiles with
-    few K-iterations, where the kernel beat hipBLASLt on every measured
-    shape — o_8b_tp8 2.77x (7.0 vs 19.4 us), o_70b_tp8 1.51x,
-    down_8b_tp8 1.32x (3-buf ring).  Deep-K shapes (K >= 4096) lose to
-    stage latency even with the ring (qkv 36.2 vs lib 32.5 us) and stay
-    on the big kernel / library.
+    projections, K = d/tp <= 1792, i.e. <= 28 K-iterations) land 256+
+    natural 64x128 tiles, where the kernel beat hipBLASLt on every
+    measured shape — o_8b_tp8 2.77x (7.0 vs 19.4 us), o_70b_tp8 1.51x,
+    down_8b_tp8 1.32x (3-buf ring).  From K=2048 up the library edges
+    ahead (o_70b_tp4 0.97x, down70b_tp8 0.85x, qkv_8b 0.90x even with
+    the ring) and those shapes stay on the big kernel / library.
     """
-    if N % 128 or K % 64 or K > 2048 or N < 4096:
+    if N % 128 or K % 64 or K > 1792 or N < 4096:
         return False
     return (N // 128) * ((M + 63) // 64) >= 256
 
