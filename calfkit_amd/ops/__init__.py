@@ -454,13 +454,17 @@ def small_plan(M: int, N: int, K: int) -> bool:
     projections, K = d/tp <= 1792, i.e. <= 28 K-iterations) land 256+
     natural 64x128 tiles, where the kernel beat hipBLASLt on every
     measured shape — o_8b_tp8 2.77x (7.0 vs 19.4 us), o_70b_tp8 1.51x,
-    down_8b_tp8 1.32x (3-buf ring).  From K=2048 up the library edges
-    ahead (o_70b_tp4 0.97x, down70b_tp8 0.85x, qkv_8b 0.90x even with
-    the ring) and those shapes stay on the big kernel / library.
+    down_8b_tp8 1.32x (3-buf ring).  The win holds down to M=192 / 96
+    blocks (o_8b_tp8 M=192: 6.3 vs 18.0 us — the library has an ~18 us
+    floor on this class at every M), which covers the TP-overlap path's
+    half-chunks of the 384/512 graph batches.  From K=2048 up the
+    library edges ahead (o_70b_tp4 0.97x, down70b_tp8 0.85x, qkv_8b
+    0.90x even with the ring) and those shapes stay on the big
+    kernel / library.
     """
     if N % 128 or K % 64 or K > 1792 or N < 4096:
         return False
-    return (N // 128) * ((M + 63) // 64) >= 256
+    return (N // 128) * ((M + 63) // 64) >= 96
 
 
 def tile_plan(M: int, N: int, K: int) -> "tuple[int, int] | None":

@@ -145,5 +145,7 @@ def test_small_plan_routes_only_the_measured_class():
     assert ops.small_plan(512, 4096, 512)        # o_8b_tp8: 2.77x
     assert ops.small_plan(512, 4096, 1792)       # down_8b_tp8: 1.32x (ring)
     assert not ops.small_plan(512, 6144, 4096)   # qkv deep-K: big kernel's
+    assert ops.small_plan(256, 4096, 512)        # TP-overlap half-chunk
+    assert ops.small_plan(192, 4096, 1792)       # 96 blocks: measured 1.3x+
     assert not ops.small_plan(64, 8192, 1024)    # skinny: skinny_gemm's
     assert not ops.small_plan(512, 1000, 512)    # N % 128
