@@ -577,3 +577,21 @@ def test_attn_decode_gqa8_vs_fp32():
             p = torch.softmax(s, -1)
             ref[b, h] = (p @ vv[:, g]).to(ref.dtype)
     assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_tp_sharded_proj_routes_small_kernel():
+    """A TP=8 rank's o-projection at the serving batch goes through the
+    64x128 small-tile kernel (ops.small_plan) and matches fp32."""
+    from calfkit_amd.models.config import get_config
+    from calfkit_amd.models.llama import LlamaShard
+
+    cfg = get_config("llama-3-8b")
+    shard = LlamaShard(cfg, tp_rank=0, tp_size=8, device="cuda")
+    T = 512
+    k_in = shard.hq * shard.dh  # 4 heads * 128 = 512 at TP=8
+    assert ops.small_plan(T, cfg.hidden, k_in)
+    x = torch.randn(T, k_in, dtype=torch.bfloat16, device="cuda")
+    w = shard.layers[0].wo
+    got = shard._proj(x, w, decode=True)
+    ref = x.float() @ w.float().T
+    assert torch.allclose(got.float(), ref, atol=8e-2, rtol=8e-2)
