@@ -32,6 +32,7 @@ async def test_flush_publishes_one_wire_record():
     led = StepLedger(node="analyst", thread_id="t9", hop_kind="tool")
     led.note(outcome="ok")
     rec = await led.flush(bus)
+    await bus.quiesce()
     assert rec["kind"] == "tool"
     assert len(seen) == 1
     wire = json.loads(seen[0].value)
