@@ -104,3 +104,30 @@ def test_stream_chunk_wire_and_sink():
     sink.emit(c)
     sink.emit(StreamChunk(event="step_finished"))
     assert sink.text == "hi"
+
+
+def test_accumulator_builds_final_message_from_deltas():
+    """Accumulator (providers.py): deltas stream to the sink in real time
+    and fold into ONE AssistantMessage — 'a real-time reflection of facts,
+    never the facts' (reference streaming.py:1-2)."""
+    from calfkit_amd.messages import ProviderFacts, TextPart, ToolCallPart
+    from calfkit_amd.providers import Accumulator
+
+    sink = CollectingSink()
+    acc = Accumulator(sink, node="analyst", thread_id="t1")
+    acc.text_delta("Hel")
+    acc.text_delta("lo")
+    acc.tool_call(ToolCallPart(call_id="c1", tool_name="search",
+                               arguments={"q": "x"}))
+    msg = acc.finish(ProviderFacts(model_name="m", finish_reason="tool_calls"),
+                     author="analyst")
+    # the reflection: ordered chunks with increasing seq
+    events = [(c.event, c.seq) for c in sink.chunks]
+    assert events == [("text_delta", 1), ("text_delta", 2), ("tool_call", 3)]
+    assert all(c.node == "analyst" and c.thread_id == "t1"
+               for c in sink.chunks)
+    # the facts: one message, text folded, calls attached
+    assert isinstance(msg.parts[0], TextPart) and msg.parts[0].text == "Hello"
+    assert msg.tool_calls[0].tool_name == "search"
+    assert msg.facts.finish_reason == "tool_calls"
+    assert msg.author == "analyst"
