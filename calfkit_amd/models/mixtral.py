@@ -119,18 +119,20 @@ class MixtralShard(LlamaShard):
         e = self.cfg.num_experts
         k = self.topk
         logits = torch.nn.functional.linear(hidden, lw.router).float()  # [T,E]
-        weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
-        weights = weights / weights.sum(-1, keepdim=True)   # renormalize top-k
 
         if self.ep_size > 1:
+            weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
+            weights = weights / weights.sum(-1, keepdim=True)
             out = self._mlp_ep(lw, hidden, weights, experts,
                                do_reduce=do_reduce)
             return out
 
         if decode and self.device.type == "cuda" and ops.have_extension() \
                 and T <= 256:
-            # grouped CDNA4 path: no host sync, hipGraph-capturable
-            tok, off, scale = ops.moe_buckets(experts, weights, e)
+            # grouped CDNA4 path: no host sync, hipGraph-capturable; the
+            # FUSED router replaces softmax/topk/argsort/cumsum/gather
+            # (~8-10 launches -> 1, profiles/r02_moe_kernelstats.md)
+            tok, off, scale = ops.moe_route(logits, e, k)
             R = T * k
             act = torch.empty(R, self.ffn_s, dtype=hidden.dtype,
                               device=hidden.device)
@@ -145,6 +147,8 @@ class MixtralShard(LlamaShard):
             return all_reduce_sum(out) \
                 if (self.tp_size > 1 and do_reduce) else out
 
+        weights, experts = torch.topk(torch.softmax(logits, -1), k, dim=-1)
+        weights = weights / weights.sum(-1, keepdim=True)   # renormalize top-k
         flat_expert = experts.reshape(-1)                   # [T*k]
         flat_token = (torch.arange(T, device=hidden.device)
                       .repeat_interleave(k))                # [T*k]

@@ -43,6 +43,7 @@ __all__ = [
     "moe_gateup_silu",
     "moe_down",
     "moe_buckets",
+    "moe_route",
     "choose_splitk",
     "build_rope_table",
 ]
@@ -583,6 +584,26 @@ def moe_buckets(experts: torch.Tensor, weights: torch.Tensor,
     tok = flat_token[order].contiguous()
     scale = weights.reshape(-1)[order].float().contiguous()
     return tok, off, scale
+
+
+def moe_route(logits: torch.Tensor, num_experts: int,
+              topk: int) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Fused router: softmax -> top-K -> renormalize -> stable expert
+    buckets in one launch (T <= 256).  Same (tok, off, scale) contract as
+    ``moe_buckets``; replaces the torch softmax/topk/argsort pipeline on
+    the grouped decode path (profiles/r02_moe_kernelstats.md)."""
+    T = logits.shape[0]
+    if _native(logits):
+        dev = logits.device
+        tok = torch.empty(T * topk, dtype=torch.int32, device=dev)
+        off = torch.empty(num_experts + 1, dtype=torch.int32, device=dev)
+        scale = torch.empty(T * topk, dtype=torch.float32, device=dev)
+        _K.moe_route(tok, off, scale, logits.contiguous(), topk)
+        return tok, off, scale
+    weights, experts = torch.topk(torch.softmax(logits.float(), -1), topk,
+                                  dim=-1)
+    weights = weights / weights.sum(-1, keepdim=True)
+    return moe_buckets(experts, weights, num_experts)
 
 
 def moe_gateup_silu(act: torch.Tensor, x: torch.Tensor, w: torch.Tensor,

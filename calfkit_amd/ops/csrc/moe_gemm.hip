@@ -249,3 +249,100 @@ void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w,
   launch_moe(y /*unused in direct mode: any fp32-typed arg*/, y, act, w,
              nullptr, off, (const float*)scale.data_ptr(), /*direct=*/1, R, s);
 }
+
+// ---------------------------------------------------------------------------
+// Fused router: softmax -> top-K -> renormalize -> stable counting sort by
+// expert, in ONE launch.  Replaces the torch pipeline (softmax, topk,
+// argsort, scatter_add, cumsum, two gathers — ~8-10 kernels, ~50 us/layer
+// of the MoE decode step in profiles/r02_moe_kernelstats.md) with a ~5 us
+// kernel.  Deterministic by construction: per-expert placement scans flat
+// entries in ascending index order, exactly the stable argsort the torch
+// path produced, so downstream index_add sees the same row order.
+// One block; T <= 256 (the grouped decode gate), E <= 16, K <= 4.
+__global__ __launch_bounds__(256, 1) void moe_route_kernel(
+    int32_t* __restrict__ tok,     // [T*K] source token per bucket row
+    int32_t* __restrict__ off,     // [E+1] bucket offsets
+    float* __restrict__ scale,     // [T*K] renormalized router weight
+    const float* __restrict__ logits,  // [T, E] fp32
+    const int T, const int E, const int K) {
+  __shared__ int16_t eids[256 * 4];
+  __shared__ float ws[256 * 4];
+  __shared__ int counts[17];
+  __shared__ int offs[17];
+  const int t = threadIdx.x;
+
+  if (t < T) {
+    float v[16];
+    float mx = -INFINITY;
+    for (int e = 0; e < E; ++e) {
+      v[e] = logits[t * E + e];
+      mx = fmaxf(mx, v[e]);
+    }
+    float denom = 0.f;
+    for (int e = 0; e < E; ++e) {
+      v[e] = __expf(v[e] - mx);
+      denom += v[e];
+    }
+    // top-K selection (descending, matching torch.topk's order) + renorm
+    float wsum = 0.f;
+    for (int j = 0; j < K; ++j) {
+      int best = -1;
+      float bw = -1.f;
+      for (int e = 0; e < E; ++e) {
+        if (v[e] > bw) {
+          bw = v[e];
+          best = e;
+        }
+      }
+      eids[t * K + j] = (int16_t)best;
+      ws[t * K + j] = bw / denom;
+      wsum += bw / denom;
+      v[best] = -1.f;  // remove from later picks
+    }
+    for (int j = 0; j < K; ++j) ws[t * K + j] /= wsum;
+  }
+  if (t <= E) counts[t] = 0;
+  __syncthreads();
+
+  const int R = T * K;
+  if (t < E) {  // per-expert count (serial: R <= 1024, E-way parallel)
+    int c = 0;
+    for (int i = 0; i < R; ++i) c += (eids[i] == t);
+    counts[t] = c;
+  }
+  __syncthreads();
+  if (t == 0) {
+    int acc = 0;
+    for (int e = 0; e <= E; ++e) {
+      offs[e] = acc;
+      off[e] = acc;
+      if (e < E) acc += counts[e];
+    }
+  }
+  __syncthreads();
+  if (t < E) {  // stable placement in ascending flat-index order
+    int pos = offs[t];
+    for (int i = 0; i < R; ++i) {
+      if (eids[i] == t) {
+        tok[pos] = i / K;
+        scale[pos] = ws[i];
+        ++pos;
+      }
+    }
+  }
+}
+
+void moe_route(torch::Tensor tok, torch::Tensor off, torch::Tensor scale,
+               torch::Tensor logits, int64_t topk) {
+  TORCH_CHECK(logits.is_contiguous() && logits.scalar_type() == torch::kFloat32);
+  const int T = logits.size(0), E = logits.size(1);
+  TORCH_CHECK(T <= 256 && E <= 16 && topk <= 4, "moe_route: T<=256, E<=16, K<=4");
+  TORCH_CHECK(tok.numel() >= T * topk && scale.numel() >= T * topk);
+  TORCH_CHECK(off.numel() >= E + 1);
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(moe_route_kernel, dim3(1), dim3(256), 0, s,
+                     (int32_t*)tok.data_ptr(), (int32_t*)off.data_ptr(),
+                     (float*)scale.data_ptr(),
+                     (const float*)logits.data_ptr(), T, E, (int)topk);
+  HIP_CHECK_LAST();
+}

@@ -595,3 +595,19 @@ def test_tp_sharded_proj_routes_small_kernel():
     got = shard._proj(x, w, decode=True)
     ref = x.float() @ w.float().T
     assert torch.allclose(got.float(), ref, atol=8e-2, rtol=8e-2)
+
+
+@pytest.mark.parametrize("T,E,K", [(64, 8, 2), (256, 8, 2), (1, 8, 2),
+                                   (100, 16, 4)])
+def test_moe_route_matches_torch_pipeline(T, E, K):
+    """Fused router == softmax/topk/renorm/stable-argsort torch pipeline
+    (same tok order, offsets and scales — downstream index_add must see
+    identical row order for determinism)."""
+    logits = torch.randn(T, E, dtype=torch.float32, device="cuda")
+    tok, off, scale = ops.moe_route(logits, E, K)
+    w, e = torch.topk(torch.softmax(logits, -1), K, dim=-1)
+    w = w / w.sum(-1, keepdim=True)
+    tok2, off2, scale2 = ops.moe_buckets(e, w, E)
+    assert torch.equal(off.cpu(), off2.cpu())
+    assert torch.equal(tok.cpu(), tok2.cpu())
+    assert torch.allclose(scale.cpu(), scale2.cpu(), atol=1e-5)
