@@ -129,10 +129,21 @@ class MixtralShard(LlamaShard):
 
         if decode and self.device.type == "cuda" and ops.have_extension() \
                 and T <= 256:
+            import os
+
             # grouped CDNA4 path: no host sync, hipGraph-capturable; the
-            # FUSED router replaces softmax/topk/argsort/cumsum/gather
-            # (~8-10 launches -> 1, profiles/r02_moe_kernelstats.md)
-            tok, off, scale = ops.moe_route(logits, e, k)
+            # FUSED router + combine replace softmax/topk/argsort/cumsum/
+            # gather and zeros/index_add/converts (~12-14 launches -> 2,
+            # profiles/r02_moe_kernelstats.md); CALF_MOE_ROUTE=0 A/Bs the
+            # torch pipeline
+            fused = os.environ.get("CALF_MOE_ROUTE", "1") != "0"
+            if fused:
+                tok, off, scale, inv = ops.moe_route(logits, e, k)
+            else:
+                weights, experts = torch.topk(torch.softmax(logits, -1), k,
+                                              dim=-1)
+                weights = weights / weights.sum(-1, keepdim=True)
+                tok, off, scale = ops.moe_buckets(experts, weights, e)
             R = T * k
             act = torch.empty(R, self.ffn_s, dtype=hidden.dtype,
                               device=hidden.device)
@@ -140,10 +151,14 @@ class MixtralShard(LlamaShard):
                                 self._moe_work)
             y = torch.empty(R, d, dtype=hidden.dtype, device=hidden.device)
             ops.moe_down(y, act, lw.wdown, off, scale)
-            out32 = torch.zeros(T, d, dtype=torch.float32,
-                                device=hidden.device)
-            out32.index_add_(0, tok.long(), y.float())  # k rows per token
-            out = out32.to(hidden.dtype)
+            if fused:
+                out = torch.empty_like(hidden)
+                ops.moe_combine(out, y, inv, k)  # fp32-accum fold of k rows
+            else:
+                out32 = torch.zeros(T, d, dtype=torch.float32,
+                                    device=hidden.device)
+                out32.index_add_(0, tok.long(), y.float())
+                out = out32.to(hidden.dtype)
             return all_reduce_sum(out) \
                 if (self.tp_size > 1 and do_reduce) else out
 

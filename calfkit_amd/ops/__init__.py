@@ -44,6 +44,7 @@ __all__ = [
     "moe_down",
     "moe_buckets",
     "moe_route",
+    "moe_combine",
     "choose_splitk",
     "build_rope_table",
 ]
@@ -587,23 +588,43 @@ def moe_buckets(experts: torch.Tensor, weights: torch.Tensor,
 
 
 def moe_route(logits: torch.Tensor, num_experts: int,
-              topk: int) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+              topk: int) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor,
+                                  torch.Tensor]:
     """Fused router: softmax -> top-K -> renormalize -> stable expert
-    buckets in one launch (T <= 256).  Same (tok, off, scale) contract as
-    ``moe_buckets``; replaces the torch softmax/topk/argsort pipeline on
-    the grouped decode path (profiles/r02_moe_kernelstats.md)."""
+    buckets in one launch (T <= 256).  Returns (tok, off, scale, inv) —
+    the ``moe_buckets`` contract plus the inverse map inv[t*K+j] = bucket
+    row, which ``moe_combine`` uses to fold the k expert outputs per
+    token deterministically (profiles/r02_moe_kernelstats.md)."""
     T = logits.shape[0]
     if _native(logits):
         dev = logits.device
         tok = torch.empty(T * topk, dtype=torch.int32, device=dev)
         off = torch.empty(num_experts + 1, dtype=torch.int32, device=dev)
         scale = torch.empty(T * topk, dtype=torch.float32, device=dev)
-        _K.moe_route(tok, off, scale, logits.contiguous(), topk)
-        return tok, off, scale
+        inv = torch.empty(T * topk, dtype=torch.int32, device=dev)
+        _K.moe_route(tok, off, scale, inv, logits.contiguous(), topk)
+        return tok, off, scale, inv
     weights, experts = torch.topk(torch.softmax(logits.float(), -1), topk,
                                   dim=-1)
     weights = weights / weights.sum(-1, keepdim=True)
-    return moe_buckets(experts, weights, num_experts)
+    tok, off, scale = moe_buckets(experts, weights, num_experts)
+    flat_expert = experts.reshape(-1)
+    order = torch.argsort(flat_expert, stable=True)
+    inv = torch.empty_like(order, dtype=torch.int32)
+    inv[order] = torch.arange(order.numel(), dtype=torch.int32)
+    return tok, off, scale, inv
+
+
+def moe_combine(out: torch.Tensor, y: torch.Tensor, inv: torch.Tensor,
+                topk: int) -> None:
+    """out[t] = sum_j y[inv[t*K+j]] with fp32 accumulation — replaces
+    zeros + index_add_ + two dtype converts (4 launches -> 1)."""
+    if _native(y):
+        _K.moe_combine(out, y, inv, topk)
+        return
+    T, d = out.shape
+    idx = inv.long().view(T, topk)
+    out.copy_(y.float()[idx].sum(1).to(out.dtype))
 
 
 def moe_gateup_silu(act: torch.Tensor, x: torch.Tensor, w: torch.Tensor,

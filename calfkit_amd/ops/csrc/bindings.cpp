@@ -45,7 +45,9 @@ void moe_gateup_silu(torch::Tensor act, torch::Tensor x, torch::Tensor w,
 void moe_down(torch::Tensor y, torch::Tensor act, torch::Tensor w,
               torch::Tensor off, torch::Tensor scale);
 void moe_route(torch::Tensor tok, torch::Tensor off, torch::Tensor scale,
-               torch::Tensor logits, int64_t topk);
+               torch::Tensor inv, torch::Tensor logits, int64_t topk);
+void moe_combine(torch::Tensor out, torch::Tensor y, torch::Tensor inv,
+                 int64_t topk);
 std::vector<std::tuple<int64_t, double>> hipblaslt_algo_bench(
     torch::Tensor out, torch::Tensor x, torch::Tensor w, int64_t iters,
     int64_t max_algos);
@@ -92,6 +94,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped MoE down: y[r] = scale[r] * act[r] @ W[e(r)]^T");
   m.def("moe_route", &moe_route,
         "fused router: softmax -> top-K -> renorm -> stable expert buckets");
+  m.def("moe_combine", &moe_combine,
+        "out[t] = sum_j y[inv[t*K+j]] (deterministic scatter-combine)");
   m.def("hipblaslt_algo_bench", &hipblaslt_algo_bench,
         "time every supported hipBLASLt algo for out = x @ w^T");
   m.def("hipblaslt_gemm_indexed", &hipblaslt_gemm_indexed,

@@ -263,6 +263,7 @@ __global__ __launch_bounds__(256, 1) void moe_route_kernel(
     int32_t* __restrict__ tok,     // [T*K] source token per bucket row
     int32_t* __restrict__ off,     // [E+1] bucket offsets
     float* __restrict__ scale,     // [T*K] renormalized router weight
+    int32_t* __restrict__ inv,     // [T*K] bucket row of flat entry t*K+j
     const float* __restrict__ logits,  // [T, E] fp32
     const int T, const int E, const int K) {
   __shared__ int16_t eids[256 * 4];
@@ -326,14 +327,32 @@ __global__ __launch_bounds__(256, 1) void moe_route_kernel(
       if (eids[i] == t) {
         tok[pos] = i / K;
         scale[pos] = ws[i];
+        inv[i] = pos;
         ++pos;
       }
     }
   }
 }
 
+// out[t] = sum_j y[inv[t*K+j]] in fp32 — the deterministic replacement
+// for zeros + index_add_ + two dtype converts (4 launches -> 1).
+__global__ void moe_combine_kernel(__hip_bfloat16* __restrict__ out,  // [T,d]
+                                   const __hip_bfloat16* __restrict__ y,  // [R,d]
+                                   const int32_t* __restrict__ inv,   // [T*K]
+                                   const int d, const int K) {
+  const int t = blockIdx.x;
+  int rows[4];
+  for (int j = 0; j < K; ++j) rows[j] = inv[t * K + j];
+  for (int c = threadIdx.x; c < d; c += blockDim.x) {
+    float acc = 0.f;
+    for (int j = 0; j < K; ++j)
+      acc += bf2f(y[(int64_t)rows[j] * d + c]);
+    out[(int64_t)t * d + c] = f2bf(acc);
+  }
+}
+
 void moe_route(torch::Tensor tok, torch::Tensor off, torch::Tensor scale,
-               torch::Tensor logits, int64_t topk) {
+               torch::Tensor inv, torch::Tensor logits, int64_t topk) {
   TORCH_CHECK(logits.is_contiguous() && logits.scalar_type() == torch::kFloat32);
   const int T = logits.size(0), E = logits.size(1);
   TORCH_CHECK(T <= 256 && E <= 16 && topk <= 4, "moe_route: T<=256, E<=16, K<=4");
@@ -342,7 +361,22 @@ void moe_route(torch::Tensor tok, torch::Tensor off, torch::Tensor scale,
   hipStream_t s = at::hip::getCurrentHIPStream().stream();
   hipLaunchKernelGGL(moe_route_kernel, dim3(1), dim3(256), 0, s,
                      (int32_t*)tok.data_ptr(), (int32_t*)off.data_ptr(),
-                     (float*)scale.data_ptr(),
+                     (float*)scale.data_ptr(), (int32_t*)inv.data_ptr(),
                      (const float*)logits.data_ptr(), T, E, (int)topk);
+  HIP_CHECK_LAST();
+}
+
+void moe_combine(torch::Tensor out, torch::Tensor y, torch::Tensor inv,
+                 int64_t topk) {
+  TORCH_CHECK(out.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(out.scalar_type() == torch::kBFloat16 &&
+              y.scalar_type() == torch::kBFloat16);
+  const int T = out.size(0), d = out.size(1);
+  TORCH_CHECK(inv.numel() >= T * topk && topk <= 4);
+  hipStream_t s = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(moe_combine_kernel, dim3(T), dim3(256), 0, s,
+                     (__hip_bfloat16*)out.data_ptr(),
+                     (const __hip_bfloat16*)y.data_ptr(),
+                     (const int32_t*)inv.data_ptr(), d, (int)topk);
   HIP_CHECK_LAST();
 }

@@ -604,10 +604,22 @@ def test_moe_route_matches_torch_pipeline(T, E, K):
     (same tok order, offsets and scales — downstream index_add must see
     identical row order for determinism)."""
     logits = torch.randn(T, E, dtype=torch.float32, device="cuda")
-    tok, off, scale = ops.moe_route(logits, E, K)
+    tok, off, scale, inv = ops.moe_route(logits, E, K)
     w, e = torch.topk(torch.softmax(logits, -1), K, dim=-1)
     w = w / w.sum(-1, keepdim=True)
     tok2, off2, scale2 = ops.moe_buckets(e, w, E)
     assert torch.equal(off.cpu(), off2.cpu())
     assert torch.equal(tok.cpu(), tok2.cpu())
     assert torch.allclose(scale.cpu(), scale2.cpu(), atol=1e-5)
+    # inv inverts the stable bucket order, and moe_combine folds the K
+    # expert rows exactly like zeros + index_add_ did
+    assert torch.equal(torch.sort(inv).values.cpu(),
+                       torch.arange(T * K, dtype=torch.int32))
+    d = 512
+    y = torch.randn(T * K, d, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty(T, d, dtype=torch.bfloat16, device="cuda")
+    ops.moe_combine(out, y, inv, K)
+    ref = torch.zeros(T, d, dtype=torch.float32, device="cuda")
+    ref.index_add_(0, tok.long(), y.float())
+    assert torch.allclose(out.float(), ref.to(torch.bfloat16).float(),
+                          atol=1e-2, rtol=1e-2)
