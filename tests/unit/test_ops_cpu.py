@@ -149,3 +149,25 @@ def test_small_plan_routes_only_the_measured_class():
     assert ops.small_plan(192, 4096, 1792)       # 96 blocks: measured 1.3x+
     assert not ops.small_plan(64, 8192, 1024)    # skinny: skinny_gemm's
     assert not ops.small_plan(512, 1000, 512)    # N % 128
+
+
+def test_moe_route_cpu_matches_bucket_pipeline():
+    """CPU fallback of the fused router == softmax/topk/renorm +
+    moe_buckets, and inv inverts the stable bucket order so moe_combine
+    reproduces index_add exactly."""
+    torch.manual_seed(5)
+    T, E, K = 16, 8, 2
+    logits = torch.randn(T, E)
+    tok, off, scale, inv = ops.moe_route(logits, E, K)
+    w, e = torch.topk(torch.softmax(logits.float(), -1), K, dim=-1)
+    w = w / w.sum(-1, keepdim=True)
+    tok2, off2, scale2 = ops.moe_buckets(e, w, E)
+    assert torch.equal(tok, tok2) and torch.equal(off, off2)
+    assert torch.allclose(scale, scale2)
+    assert torch.equal(torch.sort(inv).values.long(), torch.arange(T * K))
+    y = torch.randn(T * K, 8, dtype=torch.bfloat16)
+    out = torch.empty(T, 8, dtype=torch.bfloat16)
+    ops.moe_combine(out, y, inv, K)
+    ref = torch.zeros(T, 8)
+    ref.index_add_(0, tok.long(), y.float())
+    assert torch.allclose(out.float(), ref.to(torch.bfloat16).float())
