@@ -273,6 +273,7 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
 #define LAUNCH(GV)                                                        \
   do {                                                                    \
     if (nw == 4) LAUNCH_NW(GV, 4);                                        \
+    else if (nw == 1) LAUNCH_NW(GV, 1);                                   \
     else LAUNCH_NW(GV, 2);                                                \
   } while (0)
   switch (G) {
