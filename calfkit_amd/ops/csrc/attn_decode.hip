@@ -251,6 +251,10 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   TORCH_CHECK(page == KVPAGE, "decode kernel assumes page_size == 64");
   const int G = hq / hkv;
   const char* nw_env = getenv("CALF_ATTN_NW");
+  // r02 late A/B at the bench shape (3 pages/seq, shuffled): NW=1 (single
+  // wave, no cross-wave combine or syncs) wins the micro 108.0 vs 122.2 us
+  // but the end-to-end bench is a wash (243.8 vs 243.5); longer contexts
+  // serialize more pages per wave, so the default stays 2.
   const int nw = nw_env ? atoi(nw_env) : 2;
   const char* pf_env = getenv("CALF_ATTN_PFETCH");
   const int pfe = pf_env ? atoi(pf_env) : 0;  // prefetch LOST the r02 A/B (120.8 vs 168.1 us)
