@@ -29,7 +29,7 @@ from ..providers import Accumulator, HttpProvider
 from ..schemas import EngineRequest
 from ..streaming import ChunkSink
 
-__all__ = ["OpenAIChatProvider"]
+__all__ = ["OpenAIChatProvider", "OpenAIResponsesProvider"]
 
 
 class OpenAIChatProvider(HttpProvider):
@@ -170,3 +170,104 @@ def _finish(reason: str | None, has_calls: bool) -> str:
     if has_calls or reason == "tool_calls":
         return "tool_calls"
     return {"length": "length", "stop": "stop"}.get(reason or "stop", "stop")
+
+
+class OpenAIResponsesProvider(HttpProvider):
+    """OpenAI Responses API client — `POST {base_url}/responses`.
+
+    Capability parity: the reference names ``OpenAIResponsesModelClient``
+    (/root/reference/README.md:49; Responses API support
+    /root/reference/CHANGELOG.md:448).  Non-streaming: the Responses API
+    returns typed output ITEMS (message / function_call) rather than
+    chat-completions choices, so the mapping differs from
+    OpenAIChatProvider on both edges.
+    """
+
+    def __init__(self, base_url: str = "https://api.openai.com/v1",
+                 api_key: str = "", model: str = "gpt-4o-mini",
+                 transport: Any = None):
+        headers = {"Authorization": f"Bearer {api_key}"} if api_key else {}
+        super().__init__(base_url=base_url, headers=headers)
+        self.model = model
+        self._transport = transport
+
+    def build_payload(self, request: EngineRequest) -> dict[str, Any]:
+        items: list[dict[str, Any]] = []
+        for m in request.messages:
+            from ..messages import ToolReturnPart
+
+            returns = [p for p in m.parts if isinstance(p, ToolReturnPart)]
+            calls = [p for p in m.parts if isinstance(p, ToolCallPart)]
+            text = " ".join(p.text for p in m.parts if isinstance(p, TextPart))
+            if m.role == "tool" or returns:
+                for r in returns:
+                    items.append({"type": "function_call_output",
+                                  "call_id": r.call_id,
+                                  "output": str(r.content)})
+                if m.role == "tool" and not returns and text:
+                    items.append({"type": "function_call_output",
+                                  "call_id": "", "output": text})
+                continue
+            for c in calls:
+                items.append({"type": "function_call", "call_id": c.call_id,
+                              "name": c.tool_name,
+                              "arguments": json.dumps(c.arguments)})
+            if text or not calls:
+                ctype = ("output_text" if m.role == "assistant"
+                         else "input_text")
+                items.append({"role": m.role,
+                              "content": [{"type": ctype, "text": text}]})
+        s = request.settings
+        payload: dict[str, Any] = {
+            "model": request.model_name or self.model,
+            "input": items,
+            "max_output_tokens": s.max_tokens,
+            "temperature": s.temperature,
+            "top_p": s.top_p,
+        }
+        if request.tools:
+            # Responses API tools are FLAT (no nested "function" object)
+            payload["tools"] = [{"type": "function", **t.to_wire()}
+                                for t in request.tools]
+        return payload
+
+    def parse_response(self, data: dict[str, Any],
+                       sink: ChunkSink) -> AssistantMessage:
+        parts: list[Any] = []
+        has_calls = False
+        for item in data.get("output", ()):
+            if item.get("type") == "message":
+                for c in item.get("content", ()):
+                    if c.get("type") == "output_text" and c.get("text"):
+                        parts.append(TextPart(c["text"]))
+            elif item.get("type") == "function_call":
+                has_calls = True
+                try:
+                    args = json.loads(item.get("arguments") or "{}")
+                except json.JSONDecodeError:
+                    args = {"_raw": item.get("arguments")}
+                parts.append(ToolCallPart(call_id=item.get("call_id", ""),
+                                          tool_name=item.get("name", ""),
+                                          arguments=args))
+        u = data.get("usage") or {}
+        facts = ProviderFacts(
+            model_name=data.get("model", self.model),
+            usage=Usage(prompt_tokens=u.get("input_tokens", 0),
+                        completion_tokens=u.get("output_tokens", 0)),
+            finish_reason="tool_calls" if has_calls else (
+                "length" if data.get("incomplete_details") else "stop"))
+        return AssistantMessage(tuple(parts) or (TextPart(""),), facts=facts)
+
+    async def complete(self, request: EngineRequest,
+                       sink: ChunkSink) -> AssistantMessage:
+        import httpx
+
+        url = f"{self.base_url.rstrip('/')}/responses"
+        kw: dict[str, Any] = {"timeout": self.timeout_s}
+        if self._transport is not None:
+            kw["transport"] = self._transport
+        async with httpx.AsyncClient(**kw) as client:
+            resp = await client.post(url, json=self.build_payload(request),
+                                     headers=self.headers)
+            resp.raise_for_status()
+            return self.parse_response(resp.json(), sink)

@@ -122,7 +122,8 @@ def test_adapters_are_sole_importers_of_their_sdk():
     # sole-importer family (/root/reference/pyproject.toml:130-155:
     # "httpx -> providers.py + adapters")
     sdk_owners = {"aiokafka": ("adapters/kafka",),
-                  "httpx": ("providers.py", "adapters/openai.py"),
+                  "httpx": ("providers.py", "adapters/openai.py",
+                            "adapters/anthropic.py"),
                   "mcp": ("adapters/mcp.py",)}
     for f in PKG.rglob("*.py"):
         text = f.read_text()
