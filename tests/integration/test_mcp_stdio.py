@@ -103,3 +103,86 @@ async def test_toolbox_node_dials_stdio_server_over_the_mesh(server_script):
         req = provider.requests[-1]
         tool_msgs = [m for m in req.messages if m.role == "tool"]
         assert tool_msgs and "42" in str(tool_msgs[-1].parts[0])
+
+
+async def test_sse_session_speaks_the_protocol():
+    """Native SSE MCP dialing: endpoint event -> JSON-RPC POSTs ->
+    message events on the live stream (no SDK; a real uvicorn server on
+    a loopback port, because buffered test transports cannot stream)."""
+    import asyncio
+    import json as _json
+
+    import pytest as _pytest
+    uvicorn = _pytest.importorskip("uvicorn")
+    from calfkit_amd.adapters.mcp import SseMCPSession
+
+    queue: asyncio.Queue = asyncio.Queue()
+
+    def sse(event, obj):
+        payload = _json.dumps(obj) if not isinstance(obj, str) else obj
+        return f"event: {event}\ndata: {payload}\n\n".encode()
+
+    async def app(scope, receive, send):
+        if scope["type"] != "http":
+            return
+        if scope["method"] == "GET":
+            await send({"type": "http.response.start", "status": 200,
+                        "headers": [(b"content-type", b"text/event-stream")]})
+            await send({"type": "http.response.body",
+                        "body": sse("endpoint", "/messages"),
+                        "more_body": True})
+            while True:
+                item = await queue.get()
+                if item is None:
+                    break
+                await send({"type": "http.response.body",
+                            "body": sse("message", item),
+                            "more_body": True})
+            await send({"type": "http.response.body", "body": b""})
+            return
+        body = b""
+        while True:
+            ev = await receive()
+            body += ev.get("body", b"")
+            if not ev.get("more_body"):
+                break
+        msg = _json.loads(body)
+        m, mid = msg.get("method"), msg.get("id")
+        if m == "initialize":
+            await queue.put({"jsonrpc": "2.0", "id": mid, "result": {
+                "protocolVersion": msg["params"]["protocolVersion"],
+                "capabilities": {"tools": {}},
+                "serverInfo": {"name": "sse-mcp", "version": "1.0"}}})
+        elif m == "tools/list":
+            await queue.put({"jsonrpc": "2.0", "id": mid, "result": {
+                "tools": [{"name": "mul", "description": "Multiply.",
+                           "inputSchema": {"type": "object"}}]}})
+        elif m == "tools/call":
+            a = msg["params"]["arguments"]
+            await queue.put({"jsonrpc": "2.0", "id": mid, "result": {
+                "content": [{"type": "text",
+                             "text": str(a["a"] * a["b"])}]}})
+        await send({"type": "http.response.start", "status": 202,
+                    "headers": []})
+        await send({"type": "http.response.body", "body": b""})
+
+    config = uvicorn.Config(app, host="127.0.0.1", port=0, log_level="error",
+                            lifespan="off")
+    server = uvicorn.Server(config)
+    task = asyncio.ensure_future(server.serve())
+    while not server.started:
+        await asyncio.sleep(0.02)
+    port = server.servers[0].sockets[0].getsockname()[1]
+
+    s = await SseMCPSession.open(f"http://127.0.0.1:{port}/sse")
+    try:
+        assert s.server_info["name"] == "sse-mcp"
+        tools = await s.list_tools()
+        assert [t.name for t in tools] == ["mul"]
+        r = await s.call_tool("mul", {"a": 6, "b": 7})
+        assert r.content == "42" and not r.is_error
+    finally:
+        await queue.put(None)
+        await s.close()
+        server.should_exit = True
+        await asyncio.wait_for(task, 5)

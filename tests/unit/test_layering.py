@@ -123,7 +123,7 @@ def test_adapters_are_sole_importers_of_their_sdk():
     # "httpx -> providers.py + adapters")
     sdk_owners = {"aiokafka": ("adapters/kafka",),
                   "httpx": ("providers.py", "adapters/openai.py",
-                            "adapters/anthropic.py"),
+                            "adapters/anthropic.py", "adapters/mcp.py"),
                   "mcp": ("adapters/mcp.py",)}
     for f in PKG.rglob("*.py"):
         text = f.read_text()
