@@ -40,10 +40,14 @@ class StdioMCPSession:
     # ------------------------------------------------------------- transport
     @classmethod
     async def open(cls, command: str, args: tuple[str, ...] = (),
-                   timeout: float = 30.0) -> "StdioMCPSession":
+                   timeout: float = 30.0,
+                   env: dict[str, str] | None = None) -> "StdioMCPSession":
+        import os
+
         proc = await asyncio.create_subprocess_exec(
             command, *args,
-            stdin=asyncio.subprocess.PIPE, stdout=asyncio.subprocess.PIPE)
+            stdin=asyncio.subprocess.PIPE, stdout=asyncio.subprocess.PIPE,
+            env={**os.environ, **env} if env else None)
         self = cls(proc)
         await asyncio.wait_for(self._initialize(), timeout)
         return self
@@ -296,7 +300,8 @@ async def open_session(toolbox: MCPToolbox) -> MCPSession:
         if not toolbox.command:
             raise ValueError(
                 f"toolbox {toolbox.name!r}: stdio transport needs command=")
-        return await StdioMCPSession.open(toolbox.command, toolbox.args)
+        return await StdioMCPSession.open(toolbox.command, toolbox.args,
+                                          env=dict(toolbox.env) or None)
     if toolbox.transport == "sse":
         if not toolbox.url:
             raise ValueError(

@@ -186,3 +186,28 @@ async def test_sse_session_speaks_the_protocol():
         await s.close()
         server.should_exit = True
         await asyncio.wait_for(task, 5)
+
+
+async def test_stdio_session_passes_toolbox_env(tmp_path):
+    """mcp.json `env` entries reach the stdio server subprocess."""
+    import sys as _sys
+
+    from calfkit_amd.adapters.mcp import StdioMCPSession
+
+    script = tmp_path / "envserver.py"
+    script.write_text(
+        "import json, os, sys\n"
+        "for line in sys.stdin:\n"
+        "    msg = json.loads(line)\n"
+        "    if msg.get('method') == 'initialize':\n"
+        "        sys.stdout.write(json.dumps({'jsonrpc': '2.0',\n"
+        "            'id': msg['id'], 'result': {'serverInfo':\n"
+        "            {'name': os.environ.get('CALF_MCP_TOKEN', '?')}}})\n"
+        "            + '\\n')\n"
+        "        sys.stdout.flush()\n")
+    s = await StdioMCPSession.open(_sys.executable, (str(script),),
+                                  env={"CALF_MCP_TOKEN": "sesame"})
+    try:
+        assert s.server_info["name"] == "sesame"
+    finally:
+        await s.close()
