@@ -94,8 +94,13 @@ async def drive_step(engine: Engine, surface: StepSurface, policy: TurnPolicy,
 
     while True:
         # D1 — step the engine
+        # ToolMode rides the request: required forces some call, choice
+        # forces ONE named tool (reference selections v10.1 ToolMode)
+        mode = surface.mode
+        tool_choice = mode.choice or ("required" if mode.required else "")
         message = await engine.step(work, surface.tool_schemas(), sink, ctx,
-                                    settings_override=settings_override)
+                                    settings_override=settings_override,
+                                    tool_choice=tool_choice)
         resolution.new_messages.append(message)
         work = work + (message,)
 

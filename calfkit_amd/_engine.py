@@ -30,13 +30,15 @@ class Engine:
 
     async def step(self, history: tuple[Message, ...], tools: tuple[ToolSchema, ...],
                    sink: ChunkSink, ctx: TurnContext, *,
-                   settings_override: ModelSettings | None = None) -> AssistantMessage:
+                   settings_override: ModelSettings | None = None,
+                   tool_choice: str = "") -> AssistantMessage:
         request = EngineRequest(
             messages=history,
             tools=tools,
             settings=self._settings.merged(settings_override),
             output=self.agent.output_spec,
             model_name=self.agent.model.name if self.agent.model else "",
+            tool_choice=tool_choice,
         )
         message = await self.provider.complete(request, sink)
         if not isinstance(message, AssistantMessage):

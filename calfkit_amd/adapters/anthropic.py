@@ -97,6 +97,11 @@ class AnthropicMessagesProvider(HttpProvider):
             payload["tools"] = [{"name": t.name, "description": t.description,
                                  "input_schema": t.parameters}
                                 for t in request.tools]
+            if request.tool_choice == "required":
+                payload["tool_choice"] = {"type": "any"}
+            elif request.tool_choice:
+                payload["tool_choice"] = {"type": "tool",
+                                          "name": request.tool_choice}
         return payload
 
     # ------------------------------------------------------------ response

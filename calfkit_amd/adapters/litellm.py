@@ -85,6 +85,12 @@ class LiteLLMProvider:
         if request.tools:
             payload["tools"] = [{"type": "function", "function": t.to_wire()}
                                 for t in request.tools]
+            if request.tool_choice == "required":
+                payload["tool_choice"] = "required"
+            elif request.tool_choice:
+                payload["tool_choice"] = {
+                    "type": "function",
+                    "function": {"name": request.tool_choice}}
         return payload
 
     @staticmethod

@@ -78,6 +78,12 @@ class OpenAIChatProvider(HttpProvider):
                 {"type": "function", "function": t.to_wire()}
                 for t in request.tools]
             payload["parallel_tool_calls"] = s.parallel_tool_calls
+            if request.tool_choice == "required":
+                payload["tool_choice"] = "required"
+            elif request.tool_choice:
+                payload["tool_choice"] = {
+                    "type": "function",
+                    "function": {"name": request.tool_choice}}
         return payload
 
     # ------------------------------------------------------------ non-stream
@@ -229,6 +235,11 @@ class OpenAIResponsesProvider(HttpProvider):
             # Responses API tools are FLAT (no nested "function" object)
             payload["tools"] = [{"type": "function", **t.to_wire()}
                                 for t in request.tools]
+            if request.tool_choice == "required":
+                payload["tool_choice"] = "required"
+            elif request.tool_choice:
+                payload["tool_choice"] = {"type": "function",
+                                          "name": request.tool_choice}
         return payload
 
     def parse_response(self, data: dict[str, Any],

@@ -185,15 +185,25 @@ class LocalModelClient:
                                      for m in request.messages):
             import os
 
-            tool_names = [tok.word_id(t.name) for t in request.tools][:8]
+            advertised = request.tools
+            if request.tool_choice and request.tool_choice != "required":
+                # ToolMode.choice: the grammar may only sample THIS tool
+                chosen = tuple(t for t in request.tools
+                               if t.name == request.tool_choice)
+                advertised = chosen or request.tools
+            tool_names = [tok.word_id(t.name) for t in advertised][:8]
             seedv = (s.seed or 0) or (len(ids) * 2654435761 & 0xFFFF)
             fan = int(os.environ.get("CALF_TOOLCALL_FANOUT", "1"))
             if not s.parallel_tool_calls:
                 fan = 1
-            ncalls = max(1, min(fan, len(request.tools)))
+            ncalls = max(1, min(fan, len(advertised)))
             # fire early enough that all calls fit the token budget
             span = 7 * ncalls + 2  # TCALL+name+4 args+TEND per call, + EOS
             trigger = min(seedv % 5 + 1, max(0, s.max_tokens - span))
+            if request.tool_choice:
+                # ToolMode.required: the call fires immediately — no
+                # free-text preamble before the grammar trigger
+                trigger = 1
 
         gen = GenRequest(
             prompt_ids=ids,

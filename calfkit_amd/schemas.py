@@ -98,3 +98,7 @@ class EngineRequest:
     output: OutputSpec | None = None
     #: routing hint for multi-model providers; engine ignores unknown names
     model_name: str = ""
+    #: tool choice presented to the model: "" = auto, "required" = must
+    #: call some tool, or one advertised tool's name (ToolMode, reference
+    #: selections v10.1)
+    tool_choice: str = ""

@@ -128,3 +128,35 @@ async def test_openai_responses_round_trip():
     assert out.parts[0].text == "done"
     assert out.tool_calls[0].tool_name == "add"
     assert out.facts.finish_reason == "tool_calls"
+
+
+def test_tool_choice_maps_per_provider_dialect():
+    """tool_choice='required' and a named tool map to each API's shape."""
+    from calfkit_amd.adapters.litellm import LiteLLMProvider
+    from calfkit_amd.adapters.openai import OpenAIChatProvider
+
+    msgs = (UserMessage((TextPart("q"),)),)
+
+    def req(choice):
+        return EngineRequest(messages=msgs, tools=(TOOL,),
+                             tool_choice=choice)
+
+    chat = OpenAIChatProvider()
+    assert chat.build_payload(req("required"))["tool_choice"] == "required"
+    assert chat.build_payload(req("add"))["tool_choice"] == {
+        "type": "function", "function": {"name": "add"}}
+    assert "tool_choice" not in chat.build_payload(req(""))
+
+    responses = OpenAIResponsesProvider()
+    assert responses.build_payload(req("required"))["tool_choice"] == "required"
+    assert responses.build_payload(req("add"))["tool_choice"] == {
+        "type": "function", "name": "add"}
+
+    anthropic = AnthropicMessagesProvider()
+    assert anthropic.build_payload(req("required"))["tool_choice"] == {
+        "type": "any"}
+    assert anthropic.build_payload(req("add"))["tool_choice"] == {
+        "type": "tool", "name": "add"}
+
+    lite = LiteLLMProvider("m")
+    assert lite.build_payload(req("required"))["tool_choice"] == "required"

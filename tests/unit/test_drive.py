@@ -129,3 +129,35 @@ def test_handoff_request_is_the_typed_argument_model():
                                    arguments={"to": "expert", "reason": "r"}),
                       peers, "me")
     assert isinstance(eff, Transfer) and eff.target == "expert" and eff.reason == "r"
+
+
+async def test_tool_mode_threads_to_the_request():
+    """ToolMode(required/choice) rides StepSurface -> drive -> the
+    EngineRequest the provider sees (reference selections v10.1)."""
+    from calfkit_amd._drive import drive_step
+    from calfkit_amd._engine import Engine
+    from calfkit_amd._policy import TurnPolicy
+    from calfkit_amd._surface import StepSurface
+    from calfkit_amd.context import TurnContext
+    from calfkit_amd.definitions.agent import Agent
+    from calfkit_amd.definitions.selections import ToolMode
+    from calfkit_amd.schemas import ToolBinding, ToolSchema
+    from calfkit_amd.streaming import CollectingSink
+    from calfkit_amd.testing import FakeProvider, call_tool
+
+    schema = ToolSchema(name="add", description="Add.", parameters={
+        "type": "object", "properties": {}})
+    provider = FakeProvider([call_tool("add"), call_tool("add")])
+    agent = Agent(name="a", model_client=provider)
+    engine = Engine(agent, provider)
+    ctx = TurnContext(agent_name="a", thread_id="t")
+
+    surface = StepSurface(tools=[ToolBinding(schema=schema,
+                                             target_inbox="in.add")],
+                          mode=ToolMode(required=True))
+    await drive_step(engine, surface, TurnPolicy(), (), CollectingSink(), ctx)
+    assert provider.requests[-1].tool_choice == "required"
+
+    surface.mode = ToolMode(choice="add")
+    await drive_step(engine, surface, TurnPolicy(), (), CollectingSink(), ctx)
+    assert provider.requests[-1].tool_choice == "add"
