@@ -174,3 +174,42 @@ async def test_engine_sampled_handoff_routes_to_a_real_peer():
 @pytest.mark.gpu
 async def test_engine_sampled_handoff_routes_gpu():
     await _handoff_round("cuda", True)
+
+
+async def test_tool_mode_required_forces_an_immediate_engine_call():
+    """ToolMode(required=True) fires the grammar trigger at position 1 —
+    the engine's FIRST decode step opens a tool call (no free-text
+    preamble), and ToolMode(choice=...) restricts the sampled name."""
+    from calfkit_amd import agent_tool
+    from calfkit_amd.definitions.selections import ToolMode, Tools
+
+    calls = []
+
+    @agent_tool
+    async def alpha(x: int) -> str:
+        """Tool alpha."""
+        calls.append(("alpha", x))
+        return "a"
+
+    @agent_tool
+    async def beta(x: int) -> str:
+        """Tool beta."""
+        calls.append(("beta", x))
+        return "b"
+
+    bus, store, directory = (InMemoryBus(), InProcessThreadStore(),
+                             InMemoryDirectory())
+    provider = LocalModelClient(TINY, device="cpu", max_batch=4,
+                                max_model_len=256, use_graphs=False)
+    agent = Agent(name="chooser", model_client=provider,
+                  tools=Tools(names=("alpha", "beta"),
+                              mode=ToolMode(choice="beta")),
+                  model_settings=ModelSettings(max_tokens=24,
+                                               temperature=0.7, seed=3))
+    async with Worker(agent, alpha, beta, bus=bus, store=store,
+                      directory=directory):
+        client = await Client.connect(bus, directory)
+        result = await client.execute("chooser", "pick a tool", timeout=300)
+        # the named choice is the ONLY tool the grammar could sample
+        assert calls and all(name == "beta" for name, _ in calls)
+        assert result.text.strip()
