@@ -37,7 +37,11 @@ __global__ void rmsnorm_kernel(__hip_bfloat16* __restrict__ out,
       ss += f * f;
     }
   }
-  ss = block_sum<BLOCK>(ss, red);
+  if constexpr (BLOCK == WAVE) {
+    ss = wave_sum(ss);  // single-wave block: no LDS, no barrier
+  } else {
+    ss = block_sum<BLOCK>(ss, red);
+  }
   const float inv = rsqrtf(ss / d + eps);
 
   const bf16x8* wv = reinterpret_cast<const bf16x8*>(weight);
@@ -61,7 +65,11 @@ __global__ void fused_add_rmsnorm_kernel(__hip_bfloat16* __restrict__ hidden,
                                          __hip_bfloat16* __restrict__ residual,
                                          const __hip_bfloat16* __restrict__ weight,
                                          const int d, const float eps) {
-  constexpr int VPT = 4;  // up to d = BLOCK*8*VPT
+  // VPT scales so every BLOCK size covers d <= 8192 (70B hidden);
+  // smaller blocks = more independent blocks per CU = rows park on
+  // their own HBM round trips instead of a shared barrier (the r02 PMC
+  // read 79.6% SQ_WAIT_ANY at BLOCK=256 — latency parking, not issue).
+  constexpr int VPT = (256 * 4) / BLOCK;
   __shared__ float red[BLOCK / WAVE];
   const int64_t row = blockIdx.x;
   __hip_bfloat16* h = hidden + row * d;
@@ -86,7 +94,11 @@ __global__ void fused_add_rmsnorm_kernel(__hip_bfloat16* __restrict__ hidden,
     keep[c] = b;
     rv[i] = b;  // residual holds the sum
   }
-  ss = block_sum<BLOCK>(ss, red);
+  if constexpr (BLOCK == WAVE) {
+    ss = wave_sum(ss);  // single-wave block: no LDS, no barrier
+  } else {
+    ss = block_sum<BLOCK>(ss, red);
+  }
   const float inv = rsqrtf(ss / d + eps);
 
   const bf16x8* wv = reinterpret_cast<const bf16x8*>(weight);
@@ -360,6 +372,7 @@ __global__ void fused_add_rmsnorm_slab_kernel(
 // ------------------------------------------------------------- host glue
 
 #include <torch/extension.h>
+#include <cstdlib>
 #include <ATen/hip/HIPContext.h>
 
 static inline hipStream_t cur_stream() {
@@ -388,13 +401,17 @@ void fused_add_rmsnorm(torch::Tensor hidden, torch::Tensor residual,
   const int d = hidden.size(-1);
   TORCH_CHECK(d % 8 == 0);
   const int64_t rows = hidden.numel() / d;
-  constexpr int BLOCK = 256;
-  TORCH_CHECK(d <= BLOCK * 8 * 4, "fused_add_rmsnorm: d too large");
-  hipLaunchKernelGGL(fused_add_rmsnorm_kernel<BLOCK>, dim3(rows), dim3(BLOCK), 0,
-                     cur_stream(),
-                     (__hip_bfloat16*)hidden.data_ptr(),
-                     (__hip_bfloat16*)residual.data_ptr(),
-                     (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps);
+  TORCH_CHECK(d <= 8192, "fused_add_rmsnorm: d too large");
+  const char* bv = getenv("CALF_RMSNORM_BLOCK");
+  const int blk = bv ? atoi(bv) : 256;
+#define LFAR(B)                                                               hipLaunchKernelGGL(fused_add_rmsnorm_kernel<B>, dim3(rows), dim3(B), 0,                        cur_stream(), (__hip_bfloat16*)hidden.data_ptr(),                           (__hip_bfloat16*)residual.data_ptr(),                                       (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps)
+  switch (blk) {
+    case 64: LFAR(64); break;
+    case 128: LFAR(128); break;
+    case 512: LFAR(512); break;
+    default: LFAR(256); break;
+  }
+#undef LFAR
   HIP_CHECK_LAST();
 }
 
