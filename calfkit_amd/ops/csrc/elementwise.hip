@@ -65,10 +65,10 @@ __global__ void fused_add_rmsnorm_kernel(__hip_bfloat16* __restrict__ hidden,
                                          __hip_bfloat16* __restrict__ residual,
                                          const __hip_bfloat16* __restrict__ weight,
                                          const int d, const float eps) {
-  // VPT scales so every BLOCK size covers d <= 8192 (70B hidden);
-  // smaller blocks = more independent blocks per CU = rows park on
-  // their own HBM round trips instead of a shared barrier (the r02 PMC
-  // read 79.6% SQ_WAIT_ANY at BLOCK=256 — latency parking, not issue).
+  // VPT scales so every BLOCK size covers d <= 8192 (70B hidden).
+  // r02 PMC at BLOCK=256: 79.6% SQ_WAIT_ANY, 4.5% issue-stall — pure
+  // HBM-latency parking; the A/B showed bigger blocks (fewer, wider
+  // phases) beat more-independent smaller blocks.
   constexpr int VPT = (256 * 4) / BLOCK;
   __shared__ float red[BLOCK / WAVE];
   const int64_t row = blockIdx.x;
@@ -403,7 +403,11 @@ void fused_add_rmsnorm(torch::Tensor hidden, torch::Tensor residual,
   const int64_t rows = hidden.numel() / d;
   TORCH_CHECK(d <= 8192, "fused_add_rmsnorm: d too large");
   const char* bv = getenv("CALF_RMSNORM_BLOCK");
-  const int blk = bv ? atoi(bv) : 256;
+  // r02 late A/B (us, 100 iters): 512x4096 -> b512 4.18 / b256 4.29 /
+  // b64 6.07; 128x8192 -> 3.61 / 4.49 / 7.86.  Fewer-wider phases win;
+  // the wave-per-row barrier-free variant LOSES (the parking is HBM
+  // round trips, not the block barrier).  Stream floor ~2.5 us.
+  const int blk = bv ? atoi(bv) : 512;
 #define LFAR(B)                                                               hipLaunchKernelGGL(fused_add_rmsnorm_kernel<B>, dim3(rows), dim3(B), 0,                        cur_stream(), (__hip_bfloat16*)hidden.data_ptr(),                           (__hip_bfloat16*)residual.data_ptr(),                                       (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps)
   switch (blk) {
     case 64: LFAR(64); break;
