@@ -251,11 +251,12 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   TORCH_CHECK(page == KVPAGE, "decode kernel assumes page_size == 64");
   const int G = hq / hkv;
   const char* nw_env = getenv("CALF_ATTN_NW");
-  // r02 late A/B at the bench shape (3 pages/seq, shuffled): NW=1 (single
-  // wave, no cross-wave combine or syncs) wins the micro 108.0 vs 122.2 us
-  // but the end-to-end bench is a wash (243.8 vs 243.5); longer contexts
-  // serialize more pages per wave, so the default stays 2.
-  const int nw = nw_env ? atoi(nw_env) : 2;
+  // r02 closing A/Bs: NW=1 (single wave — no cross-wave combine, no
+  // barriers) wins the shuffled-page micro 108.0 vs 122.2 us AND the
+  // interleaved same-box bench 1,2,2,1 ladder (244.6/254.6 vs
+  // 239.4/240.0, ~+4%).  Long contexts serialize npages per wave —
+  // CALF_ATTN_NW=2/4 restores the split if a workload needs it.
+  const int nw = nw_env ? atoi(nw_env) : 1;
   const char* pf_env = getenv("CALF_ATTN_PFETCH");
   const int pfe = pf_env ? atoi(pf_env) : 0;  // prefetch LOST the r02 A/B (120.8 vs 168.1 us)
   dim3 grid(B, hkv), block(nw * WAVE);
@@ -277,8 +278,8 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
 #define LAUNCH(GV)                                                        \
   do {                                                                    \
     if (nw == 4) LAUNCH_NW(GV, 4);                                        \
-    else if (nw == 1) LAUNCH_NW(GV, 1);                                   \
-    else LAUNCH_NW(GV, 2);                                                \
+    else if (nw == 2) LAUNCH_NW(GV, 2);                                   \
+    else LAUNCH_NW(GV, 1);                                                \
   } while (0)
   switch (G) {
     case 1: LAUNCH(1); break;
