@@ -254,8 +254,9 @@ void attn_decode(torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
   // r02 closing A/Bs: NW=1 (single wave — no cross-wave combine, no
   // barriers) wins the shuffled-page micro 108.0 vs 122.2 us AND the
   // interleaved same-box bench 1,2,2,1 ladder (244.6/254.6 vs
-  // 239.4/240.0, ~+4%).  Long contexts serialize npages per wave —
-  // CALF_ATTN_NW=2/4 restores the split if a workload needs it.
+  // 239.4/240.0, ~+4%), and the 512p/128d long-context bench too
+  // (71.0 vs 69.1 turns/s at ~10 pages/seq) — block-level parallelism
+  // covers the serial pages.  CALF_ATTN_NW=2/4 keeps the split form.
   const int nw = nw_env ? atoi(nw_env) : 1;
   const char* pf_env = getenv("CALF_ATTN_PFETCH");
   const int pfe = pf_env ? atoi(pf_env) : 0;  // prefetch LOST the r02 A/B (120.8 vs 168.1 us)
