@@ -383,6 +383,12 @@ static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
+static int norm_block() {
+  // shared with fused_add_rmsnorm: the r02 closing A/B picked 512
+  const char* bv = getenv("CALF_RMSNORM_BLOCK");
+  return bv ? atoi(bv) : 512;
+}
+
 void rmsnorm(torch::Tensor out, torch::Tensor inp, torch::Tensor weight,
              double eps) {
   TORCH_CHECK(inp.is_contiguous() && out.is_contiguous());
@@ -390,12 +396,14 @@ void rmsnorm(torch::Tensor out, torch::Tensor inp, torch::Tensor weight,
   const int d = inp.size(-1);
   TORCH_CHECK(d % 8 == 0, "hidden dim must be a multiple of 8");
   const int64_t rows = inp.numel() / d;
-  constexpr int BLOCK = 256;
-  hipLaunchKernelGGL(rmsnorm_kernel<BLOCK>, dim3(rows), dim3(BLOCK), 0,
-                     cur_stream(),
-                     (__hip_bfloat16*)out.data_ptr(),
-                     (const __hip_bfloat16*)inp.data_ptr(),
-                     (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps);
+#define LRN(B)                                                                hipLaunchKernelGGL(rmsnorm_kernel<B>, dim3(rows), dim3(B), 0,                                  cur_stream(), (__hip_bfloat16*)out.data_ptr(),                              (const __hip_bfloat16*)inp.data_ptr(),                                      (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps)
+  switch (norm_block()) {
+    case 64: LRN(64); break;
+    case 128: LRN(128); break;
+    case 256: LRN(256); break;
+    default: LRN(512); break;
+  }
+#undef LRN
   HIP_CHECK_LAST();
 }
 
@@ -406,12 +414,11 @@ void fused_add_rmsnorm(torch::Tensor hidden, torch::Tensor residual,
   TORCH_CHECK(d % 8 == 0);
   const int64_t rows = hidden.numel() / d;
   TORCH_CHECK(d <= 8192, "fused_add_rmsnorm: d too large");
-  const char* bv = getenv("CALF_RMSNORM_BLOCK");
   // r02 late A/B (us, 100 iters): 512x4096 -> b512 4.18 / b256 4.29 /
   // b64 6.07; 128x8192 -> 3.61 / 4.49 / 7.86.  Fewer-wider phases win;
   // the wave-per-row barrier-free variant LOSES (the parking is HBM
   // round trips, not the block barrier).  Stream floor ~2.5 us.
-  const int blk = bv ? atoi(bv) : 512;
+  const int blk = norm_block();
 #define LFAR(B)                                                               hipLaunchKernelGGL(fused_add_rmsnorm_kernel<B>, dim3(rows), dim3(B), 0,                        cur_stream(), (__hip_bfloat16*)hidden.data_ptr(),                           (__hip_bfloat16*)residual.data_ptr(),                                       (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps)
   switch (blk) {
     case 64: LFAR(64); break;
@@ -512,14 +519,14 @@ void fused_add_rmsnorm_slab(torch::Tensor hidden, torch::Tensor residual,
   const int d = hidden.size(-1);
   TORCH_CHECK(d % 8 == 0);
   const int64_t rows = hidden.numel() / d;
-  constexpr int BLOCK = 256;
-  hipLaunchKernelGGL(fused_add_rmsnorm_slab_kernel<BLOCK>, dim3(rows),
-                     dim3(BLOCK), 0, cur_stream(),
-                     (__hip_bfloat16*)hidden.data_ptr(),
-                     (__hip_bfloat16*)residual.data_ptr(),
-                     (const float*)work.data_ptr(),
-                     (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps,
-                     (int)Mp, (int)splitk);
+#define LFS(B)                                                                hipLaunchKernelGGL(fused_add_rmsnorm_slab_kernel<B>, dim3(rows), dim3(B),                      0, cur_stream(), (__hip_bfloat16*)hidden.data_ptr(),                        (__hip_bfloat16*)residual.data_ptr(),                                       (const float*)work.data_ptr(),                                              (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps,                     (int)Mp, (int)splitk)
+  switch (norm_block()) {
+    case 64: LFS(64); break;
+    case 128: LFS(128); break;
+    case 256: LFS(256); break;
+    default: LFS(512); break;
+  }
+#undef LFS
   HIP_CHECK_LAST();
 }
 
