@@ -383,10 +383,13 @@ static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
-static int norm_block() {
-  // shared with fused_add_rmsnorm: the r02 closing A/B picked 512
+static int norm_block(int dflt) {
+  // per-kernel measured defaults (r02 closing A/Bs): fused_add 512
+  // (4.18 vs 4.29 us); the slab twin 256 (8.34 vs 8.88 — its fp32 slab
+  // reads prefer more blocks in flight); plain rmsnorm a wash (3.72 vs
+  // 3.76), kept at its winner 256.  Env overrides all three for A/B.
   const char* bv = getenv("CALF_RMSNORM_BLOCK");
-  return bv ? atoi(bv) : 512;
+  return bv ? atoi(bv) : dflt;
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor inp, torch::Tensor weight,
@@ -397,11 +400,11 @@ void rmsnorm(torch::Tensor out, torch::Tensor inp, torch::Tensor weight,
   TORCH_CHECK(d % 8 == 0, "hidden dim must be a multiple of 8");
   const int64_t rows = inp.numel() / d;
 #define LRN(B)                                                                hipLaunchKernelGGL(rmsnorm_kernel<B>, dim3(rows), dim3(B), 0,                                  cur_stream(), (__hip_bfloat16*)out.data_ptr(),                              (const __hip_bfloat16*)inp.data_ptr(),                                      (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps)
-  switch (norm_block()) {
+  switch (norm_block(256)) {
     case 64: LRN(64); break;
     case 128: LRN(128); break;
-    case 256: LRN(256); break;
-    default: LRN(512); break;
+    case 512: LRN(512); break;
+    default: LRN(256); break;
   }
 #undef LRN
   HIP_CHECK_LAST();
@@ -418,7 +421,7 @@ void fused_add_rmsnorm(torch::Tensor hidden, torch::Tensor residual,
   // b64 6.07; 128x8192 -> 3.61 / 4.49 / 7.86.  Fewer-wider phases win;
   // the wave-per-row barrier-free variant LOSES (the parking is HBM
   // round trips, not the block barrier).  Stream floor ~2.5 us.
-  const int blk = norm_block();
+  const int blk = norm_block(512);
 #define LFAR(B)                                                               hipLaunchKernelGGL(fused_add_rmsnorm_kernel<B>, dim3(rows), dim3(B), 0,                        cur_stream(), (__hip_bfloat16*)hidden.data_ptr(),                           (__hip_bfloat16*)residual.data_ptr(),                                       (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps)
   switch (blk) {
     case 64: LFAR(64); break;
@@ -520,11 +523,11 @@ void fused_add_rmsnorm_slab(torch::Tensor hidden, torch::Tensor residual,
   TORCH_CHECK(d % 8 == 0);
   const int64_t rows = hidden.numel() / d;
 #define LFS(B)                                                                hipLaunchKernelGGL(fused_add_rmsnorm_slab_kernel<B>, dim3(rows), dim3(B),                      0, cur_stream(), (__hip_bfloat16*)hidden.data_ptr(),                        (__hip_bfloat16*)residual.data_ptr(),                                       (const float*)work.data_ptr(),                                              (const __hip_bfloat16*)weight.data_ptr(), d, (float)eps,                     (int)Mp, (int)splitk)
-  switch (norm_block()) {
+  switch (norm_block(256)) {
     case 64: LFS(64); break;
     case 128: LFS(128); break;
-    case 256: LFS(256); break;
-    default: LFS(512); break;
+    case 512: LFS(512); break;
+    default: LFS(256); break;
   }
 #undef LFS
   HIP_CHECK_LAST();
